@@ -1,0 +1,60 @@
+// Python bindings for the gfx950 kernel set.
+#include <torch/extension.h>
+
+// elementwise.hip
+at::Tensor relu_fwd(at::Tensor x);
+at::Tensor relu_bwd(at::Tensor gy, at::Tensor y);
+at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b);
+// bn.hip
+std::vector<at::Tensor> bn_sums(at::Tensor x);
+at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
+                        int64_t act, c10::optional<at::Tensor> res);
+std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
+                               c10::optional<at::Tensor> res, at::Tensor mean,
+                               at::Tensor rstd, at::Tensor gamma,
+                               at::Tensor beta, int64_t act, bool training,
+                               bool need_gres);
+// pool.hip
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
+                                    int64_t P);
+at::Tensor maxpool_bwd(at::Tensor gy, at::Tensor idx, int64_t H, int64_t W,
+                       int64_t K, int64_t S, int64_t P);
+at::Tensor gap_fwd(at::Tensor x);
+at::Tensor gap_bwd(at::Tensor gy, int64_t H, int64_t W);
+at::Tensor avgpool_fwd(at::Tensor x, int64_t K, int64_t S);
+at::Tensor avgpool_bwd(at::Tensor gy, int64_t K, int64_t S, int64_t H,
+                       int64_t W);
+// loss.hip
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                  at::Tensor gl);
+std::vector<at::Tensor> topk_acc(at::Tensor logits, at::Tensor target,
+                                 int64_t topk);
+// sgd.hip
+void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+              std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
+              double lr, double momentum, double dampening,
+              double weight_decay, bool nesterov);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("relu_fwd", &relu_fwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("bn_sums", &bn_sums);
+  m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
+        py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
+  m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"), py::arg("y"),
+        py::arg("res"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
+        py::arg("beta"), py::arg("act"), py::arg("training"),
+        py::arg("need_gres"));
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("gap_fwd", &gap_fwd);
+  m.def("gap_bwd", &gap_bwd);
+  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("avgpool_bwd", &avgpool_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("topk_acc", &topk_acc);
+  m.def("sgd_step", &sgd_step);
+}

@@ -1,0 +1,254 @@
+#include "hip/hip_runtime.h"
+// Pooling kernels: NHWC max-pool with argmax indices,
+// global average pool (one-workgroup-per-(n,c-slab) reduction), avg-pool 2x2.
+#include "common_hip.h"
+
+namespace {
+
+// ---- max pool fwd: thread per (n,ho,wo,c), coalesced over c ---------------
+template <typename T>
+__global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int* __restrict__ idx, int N, int H, int W,
+                                   int C, int Ho, int Wo, int K, int S, int P) {
+  const int64_t total = (int64_t)N * Ho * Wo * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int wo = t % Wo;
+    t /= Wo;
+    const int ho = t % Ho;
+    const int n = t / Ho;
+    const int h0 = ho * S - P, w0 = wo * S - P;
+    float best = -INFINITY;
+    int best_idx = 0;
+    for (int kh = 0; kh < K; ++kh) {
+      const int h = h0 + kh;
+      if (h < 0 || h >= H) continue;
+      for (int kw = 0; kw < K; ++kw) {
+        const int w = w0 + kw;
+        if (w < 0 || w >= W) continue;
+        float v = to_f32(x[(((int64_t)n * H + h) * W + w) * C + c]);
+        if (v > best) {
+          best = v;
+          best_idx = h * W + w;
+        }
+      }
+    }
+    y[i] = from_f32<T>(best);
+    idx[i] = best_idx;
+  }
+}
+
+// ---- max pool bwd: gather per input element (no atomics) ------------------
+template <typename T>
+__global__ void maxpool_bwd_kernel(const T* __restrict__ gy,
+                                   const int* __restrict__ idx,
+                                   T* __restrict__ gx, int N, int H, int W,
+                                   int C, int Ho, int Wo, int K, int S, int P) {
+  const int64_t total = (int64_t)N * H * W * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int w = t % W;
+    t /= W;
+    const int h = t % H;
+    const int n = t / H;
+    const int flat = h * W + w;
+    float acc = 0.f;
+    // output windows that can cover (h, w)
+    const int ho_lo = max(0, (h + P - K + S) / S), ho_hi = min(Ho - 1, (h + P) / S);
+    const int wo_lo = max(0, (w + P - K + S) / S), wo_hi = min(Wo - 1, (w + P) / S);
+    for (int ho = ho_lo; ho <= ho_hi; ++ho)
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        const int64_t o = (((int64_t)n * Ho + ho) * Wo + wo) * C + c;
+        if (idx[o] == flat) acc += to_f32(gy[o]);
+      }
+    gx[i] = from_f32<T>(acc);
+  }
+}
+
+// ---- global average pool: block per (n, c-slab); thread per channel -------
+template <typename T>
+__global__ void gap_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                               int HW, int C, float inv_hw) {
+  const int n = blockIdx.x;
+  for (int c = blockIdx.y * blockDim.x + threadIdx.x; c < C;
+       c += gridDim.y * blockDim.x) {
+    const T* base = x + (int64_t)n * HW * C + c;
+    float acc = 0.f;
+    for (int i = 0; i < HW; ++i) acc += to_f32(base[(int64_t)i * C]);
+    y[(int64_t)n * C + c] = from_f32<T>(acc * inv_hw);
+  }
+}
+
+template <typename T>
+__global__ void gap_bwd_kernel(const T* __restrict__ gy, T* __restrict__ gx,
+                               int64_t total, int HW, int C, float inv_hw) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    const int64_t n = i / ((int64_t)HW * C);
+    gx[i] = from_f32<T>(to_f32(gy[n * C + c]) * inv_hw);
+  }
+}
+
+// ---- avg pool KxK stride S (no padding; densenet/botnet use 2x2 s2) -------
+template <typename T>
+__global__ void avgpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int N, int H, int W, int C, int Ho, int Wo,
+                                   int K, int S, float inv_kk) {
+  const int64_t total = (int64_t)N * Ho * Wo * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int wo = t % Wo;
+    t /= Wo;
+    const int ho = t % Ho;
+    const int n = t / Ho;
+    float acc = 0.f;
+    for (int kh = 0; kh < K; ++kh) {
+      const int h = ho * S + kh;
+      if (h >= H) continue;
+      for (int kw = 0; kw < K; ++kw) {
+        const int w = wo * S + kw;
+        if (w >= W) continue;
+        acc += to_f32(x[(((int64_t)n * H + h) * W + w) * C + c]);
+      }
+    }
+    y[i] = from_f32<T>(acc * inv_kk);
+  }
+}
+
+template <typename T>
+__global__ void avgpool_bwd_kernel(const T* __restrict__ gy, T* __restrict__ gx,
+                                   int N, int H, int W, int C, int Ho, int Wo,
+                                   int K, int S, float inv_kk) {
+  const int64_t total = (int64_t)N * H * W * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int w = t % W;
+    t /= W;
+    const int h = t % H;
+    const int n = t / H;
+    float acc = 0.f;
+    const int ho_lo = max(0, (h - K + S) / S), ho_hi = min(Ho - 1, h / S);
+    const int wo_lo = max(0, (w - K + S) / S), wo_hi = min(Wo - 1, w / S);
+    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+      if (h - ho * S >= K) continue;
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        if (w - wo * S >= K) continue;
+        acc += to_f32(gy[(((int64_t)n * Ho + ho) * Wo + wo) * C + c]);
+      }
+    }
+    gx[i] = from_f32<T>(acc * inv_kk);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
+                                    int64_t P) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int Ho = (H + 2 * P - K) / S + 1, Wo = (W + 2 * P - K) / S + 1;
+  auto y = at::empty({N, C, Ho, Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = at::empty({N, C, Ho, Wo},
+                       x.options()
+                           .dtype(at::kInt)
+                           .memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * Ho * Wo * C;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "maxpool_fwd", [&] {
+    hipLaunchKernelGGL((maxpool_fwd_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                       idx.data_ptr<int>(), N, H, W, C, Ho, Wo, K, S, P);
+  });
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor gy, at::Tensor idx, int64_t H, int64_t W,
+                       int64_t K, int64_t S, int64_t P) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int N = gy.size(0), C = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
+  auto gx = at::empty({N, C, H, W},
+                      gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * H * W * C;
+  DISPATCH_FLOAT_AND_BF16(gy.scalar_type(), "maxpool_bwd", [&] {
+    hipLaunchKernelGGL((maxpool_bwd_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)gy.data_ptr(), idx.data_ptr<int>(),
+                       (scalar_t*)gx.data_ptr(), N, H, W, C, Ho, Wo, K, S, P);
+  });
+  return gx;
+}
+
+at::Tensor gap_fwd(at::Tensor x) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  auto y = at::empty({N, C, 1, 1}, x.options());
+  const int cblocks = (int)std::min<int64_t>(ceil_div(C, 256), 8);
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "gap_fwd", [&] {
+    hipLaunchKernelGGL((gap_fwd_kernel<scalar_t>), dim3(N, cblocks), dim3(256),
+                       0, cur_stream(), (const scalar_t*)x.data_ptr(),
+                       (scalar_t*)y.data_ptr(), H * W, C, 1.f / (H * W));
+  });
+  return y;
+}
+
+at::Tensor gap_bwd(at::Tensor gy, int64_t H, int64_t W) {
+  CHECK_GPU(gy);
+  const int N = gy.size(0), C = gy.size(1);
+  auto gx = at::empty({N, C, H, W},
+                      gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * H * W * C;
+  DISPATCH_FLOAT_AND_BF16(gy.scalar_type(), "gap_bwd", [&] {
+    hipLaunchKernelGGL((gap_bwd_kernel<scalar_t>), dim3(grid_1d(total, 256)),
+                       dim3(256), 0, cur_stream(),
+                       (const scalar_t*)gy.data_ptr(), (scalar_t*)gx.data_ptr(),
+                       total, H * W, C, 1.f / (H * W));
+  });
+  return gx;
+}
+
+at::Tensor avgpool_fwd(at::Tensor x, int64_t K, int64_t S) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int Ho = (H - K) / S + 1, Wo = (W - K) / S + 1;
+  auto y = at::empty({N, C, Ho, Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * Ho * Wo * C;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "avgpool_fwd", [&] {
+    hipLaunchKernelGGL((avgpool_fwd_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                       N, H, W, C, Ho, Wo, K, S, 1.f / (K * K));
+  });
+  return y;
+}
+
+at::Tensor avgpool_bwd(at::Tensor gy, int64_t K, int64_t S, int64_t H,
+                       int64_t W) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int N = gy.size(0), C = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
+  auto gx = at::empty({N, C, H, W},
+                      gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * H * W * C;
+  DISPATCH_FLOAT_AND_BF16(gy.scalar_type(), "avgpool_bwd", [&] {
+    hipLaunchKernelGGL((avgpool_bwd_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)gy.data_ptr(), (scalar_t*)gx.data_ptr(),
+                       N, H, W, C, Ho, Wo, K, S, 1.f / (K * K));
+  });
+  return gx;
+}

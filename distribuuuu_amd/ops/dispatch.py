@@ -1,0 +1,95 @@
+"""Extension loading & per-op dispatch policy.
+
+The gfx950 HIP extension is built IN-TREE (``python setup.py build_ext --inplace``
+→ ``distribuuuu_amd/_hip_ops*.so``) so the binary travels with the repo snapshot.
+There is no JIT cache dependency and no multi-backend dispatch: gfx950 or CPU.
+
+Policy on a GPU box:
+* extension missing entirely  -> RuntimeError at the first op call (loud).
+* a specific kernel missing   -> RuntimeError too, unless the op is explicitly
+  listed in DISTRIBUUUU_SOFT_OPS (comma list; dev bring-up only) in which case a
+  one-time warning is emitted and ATen runs.
+"""
+
+import os
+import warnings
+
+import torch
+
+_EXT = None
+_EXT_TRIED = False
+_WARNED = set()
+
+
+def ext():
+    """Return the loaded _hip_ops module, or None when unavailable (CPU-only box)."""
+    global _EXT, _EXT_TRIED
+    if _EXT_TRIED:
+        return _EXT
+    _EXT_TRIED = True
+    try:
+        from distribuuuu_amd import _hip_ops  # built in-tree by setup.py
+
+        _EXT = _hip_ops
+    except ImportError:
+        _EXT = None
+    return _EXT
+
+
+def require_ext():
+    """On a GPU machine the HIP extension must be present — fail loudly, never
+    fall back to ATen silently on the hot path."""
+    e = ext()
+    if e is None and torch.cuda.is_available():
+        raise RuntimeError(
+            "distribuuuu_amd._hip_ops is not built but a GPU is visible. "
+            "Build it in-tree: python setup.py build_ext --inplace "
+            "(PYTORCH_ROCM_ARCH=gfx950)."
+        )
+    return e
+
+
+def hip_op_available(name):
+    if os.environ.get("DISTRIBUUUU_FORCE_TORCH", "0") == "1":
+        return False
+    e = ext()
+    return e is not None and hasattr(e, name)
+
+
+# Ops whose HIP kernels are not yet implemented: ATen fallback allowed WITH a
+# warning. This set shrinks as kernels land; it must be empty for the hot path
+# by release.
+_DEFAULT_SOFT = {
+    "mhsa_relpos_fwd",  # fused BoT attention kernel: pending
+    "conv2d_fwd",       # implicit-GEMM conv: pending
+    "linear_fwd",       # MFMA GEMM fc: pending
+}
+
+
+def _soft_ops():
+    return _DEFAULT_SOFT | set(
+        filter(None, os.environ.get("DISTRIBUUUU_SOFT_OPS", "").split(","))
+    )
+
+
+def use_hip(x, name):
+    """Dispatch decision: HIP kernel iff the tensor lives on GPU and the kernel exists."""
+    if not x.is_cuda:
+        return False
+    if os.environ.get("DISTRIBUUUU_FORCE_TORCH", "0") == "1":
+        return False
+    e = require_ext()
+    if e is not None and hasattr(e, name):
+        return True
+    if name in _soft_ops() or e is None:
+        if name not in _WARNED:
+            _WARNED.add(name)
+            warnings.warn(
+                f"HIP kernel '{name}' unavailable; ATen fallback in use (dev only)."
+            )
+        return False
+    raise RuntimeError(
+        f"HIP kernel '{name}' missing from _hip_ops on a GPU tensor. "
+        "Rebuild the extension (python setup.py build_ext --inplace) or, for "
+        f"debugging only, add it to DISTRIBUUUU_SOFT_OPS."
+    )

@@ -1,0 +1,309 @@
+"""Functional op layer: CPU = plain ATen composition (autograd handles backward);
+GPU = custom autograd.Functions over the gfx950 HIP kernels.
+
+Kernel inventory mirrors SURVEY.md §2b (K1-K20). All GPU compute tensors are
+NHWC (PyTorch ``channels_last``): logical NCHW with [N,H,W,C] physical layout,
+weights [K,C,R,S] logical → [K,R,S,C] physical.
+"""
+
+import torch
+import torch.nn.functional as F
+
+from .dispatch import ext, use_hip
+
+_ACTS = {"none": 0, "relu": 1, "silu": 2, "sigmoid": 3}
+
+
+def _cl(x):
+    """Ensure channels_last physical layout for 4-D GPU tensors."""
+    if x.dim() == 4:
+        return x.contiguous(memory_format=torch.channels_last)
+    return x.contiguous()
+
+
+# ---------------------------------------------------------------------------
+# Convolution (K1-K5): implicit-GEMM MFMA kernels on GPU
+# ---------------------------------------------------------------------------
+class _HIPConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, padding, dilation, groups):
+        x = _cl(x)
+        w = _cl(w)
+        y = ext().conv2d_fwd(x, w, stride[0], stride[1], padding[0], padding[1],
+                             dilation[0], dilation[1], groups)
+        if bias is not None:
+            y = y + bias.reshape(1, -1, 1, 1)
+        ctx.save_for_backward(x, w)
+        ctx.conf = (stride, padding, dilation, groups, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        stride, padding, dilation, groups, has_bias = ctx.conf
+        gy = _cl(gy)
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = ext().conv2d_dgrad(gy, w, x.shape[2], x.shape[3],
+                                    stride[0], stride[1], padding[0], padding[1],
+                                    dilation[0], dilation[1], groups)
+        if ctx.needs_input_grad[1]:
+            gw = ext().conv2d_wgrad(gy, x, w.shape[2], w.shape[3],
+                                    stride[0], stride[1], padding[0], padding[1],
+                                    dilation[0], dilation[1], groups)
+        if has_bias and ctx.needs_input_grad[2]:
+            gb = gy.sum(dim=(0, 2, 3))
+        return gx, gw, gb, None, None, None, None
+
+
+def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
+           groups=1):
+    if use_hip(x, "conv2d_fwd"):
+        return _HIPConv2d.apply(x, weight, bias, stride, padding, dilation, groups)
+    return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm (+residual add + activation) — K6/K8/K20 fused
+# ---------------------------------------------------------------------------
+class _HIPBatchNormAct(torch.autograd.Function):
+    """Training-mode fused BN: one per-channel fp32 sums reduction kernel +
+    one normalize(+add)(+act) elementwise kernel; backward is one grad-stats
+    reduction + one dx kernel. Running stats updated in fp32."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var, training,
+                momentum, eps, act, residual):
+        x = _cl(x)
+        e = ext()
+        act_id = _ACTS[act]
+        c = x.shape[1]
+        if training:
+            s, ss = e.bn_sums(x)  # fp32 [C]
+            n = x.numel() / c
+            mean = s / n
+            var = (ss / n - mean * mean).clamp_(min=0)
+            if running_mean is not None:
+                with torch.no_grad():
+                    unbiased = var * (n / max(n - 1, 1))
+                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                    running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        else:
+            mean = running_mean.float()
+            var = running_var.float()
+        rstd = (var + eps).rsqrt()
+        gamma = weight.float()
+        beta = bias.float()
+        scale = gamma * rstd
+        shift = beta - mean * scale
+        res = _cl(residual) if residual is not None else None
+        y = e.bn_apply_act(x, scale, shift, act_id, res)
+        ctx.save_for_backward(x, gamma, beta, mean, rstd, y,
+                              res if res is not None else x.new_empty(0))
+        ctx.act_id = act_id
+        ctx.training = training
+        ctx.has_res = residual is not None
+        ctx.w_dtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, gamma, beta, mean, rstd, y, res = ctx.saved_tensors
+        e = ext()
+        gy = _cl(gy)
+        gx, gw, gb, gres = e.bn_bwd(
+            gy, x, y, res if ctx.has_res else None, mean, rstd, gamma, beta,
+            ctx.act_id, ctx.training, ctx.has_res)
+        return (gx, gw.to(ctx.w_dtype), gb.to(ctx.w_dtype), None, None, None,
+                None, None, None, gres if ctx.has_res else None)
+
+
+def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
+                   momentum=0.1, eps=1e-5, act="none", residual=None):
+    if use_hip(x, "bn_sums"):
+        return _HIPBatchNormAct.apply(x, weight, bias, running_mean, running_var,
+                                      training, momentum, eps, act, residual)
+    y = F.batch_norm(x, running_mean, running_var, weight, bias, training,
+                     momentum, eps)
+    if residual is not None:
+        y = y + residual
+    return _apply_act(y, act)
+
+
+def _apply_act(y, act):
+    if act == "relu":
+        return F.relu(y, inplace=True)
+    if act == "silu":
+        return F.silu(y, inplace=True)
+    if act == "sigmoid":
+        return torch.sigmoid(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# Elementwise residual add + ReLU (K20/K8)
+# ---------------------------------------------------------------------------
+class _HIPAddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        y = ext().add_relu_fwd(_cl(a), _cl(b))
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (y,) = ctx.saved_tensors
+        g = ext().relu_bwd(_cl(gy), y)
+        return g, g
+
+
+def add_relu(a, b):
+    if use_hip(a, "add_relu_fwd"):
+        return _HIPAddReLU.apply(a, b)
+    return F.relu(a + b)
+
+
+class _HIPReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = ext().relu_fwd(_cl(x))
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (y,) = ctx.saved_tensors
+        return ext().relu_bwd(_cl(gy), y)
+
+
+def relu(x):
+    if use_hip(x, "relu_fwd"):
+        return _HIPReLU.apply(x)
+    return F.relu(x)
+
+
+# ---------------------------------------------------------------------------
+# Pooling (K9/K10)
+# ---------------------------------------------------------------------------
+class _HIPMaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        x = _cl(x)
+        y, idx = ext().maxpool_fwd(x, kernel, stride, padding)
+        ctx.save_for_backward(idx)
+        ctx.conf = (x.shape, kernel, stride, padding)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (idx,) = ctx.saved_tensors
+        shape, kernel, stride, padding = ctx.conf
+        gx = ext().maxpool_bwd(_cl(gy), idx, shape[2], shape[3], kernel,
+                               stride, padding)
+        return gx, None, None, None
+
+
+def max_pool2d(x, kernel_size, stride, padding):
+    if use_hip(x, "maxpool_fwd"):
+        return _HIPMaxPool2d.apply(x, kernel_size, stride, padding)
+    return F.max_pool2d(x, kernel_size, stride, padding)
+
+
+class _HIPGlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = _cl(x)
+        ctx.in_shape = x.shape
+        return ext().gap_fwd(x)  # [N, C, 1, 1]
+
+    @staticmethod
+    def backward(ctx, gy):
+        n, c, h, w = ctx.in_shape
+        return ext().gap_bwd(gy.contiguous(), h, w)
+
+
+def adaptive_avg_pool2d(x, output_size=1):
+    if output_size in (1, (1, 1)) and use_hip(x, "gap_fwd"):
+        return _HIPGlobalAvgPool.apply(x)
+    return F.adaptive_avg_pool2d(x, output_size)
+
+
+def avg_pool2d(x, kernel_size, stride=None):
+    if use_hip(x, "avgpool_fwd"):
+        return _HIPAvgPool2d.apply(x, kernel_size, stride or kernel_size)
+    return F.avg_pool2d(x, kernel_size, stride)
+
+
+class _HIPAvgPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride):
+        x = _cl(x)
+        ctx.conf = (x.shape, kernel, stride)
+        return ext().avgpool_fwd(x, kernel, stride)
+
+    @staticmethod
+    def backward(ctx, gy):
+        shape, kernel, stride = ctx.conf
+        gx = ext().avgpool_bwd(_cl(gy), kernel, stride, shape[2], shape[3])
+        return gx, None, None
+
+
+# ---------------------------------------------------------------------------
+# Linear / fc (K11) — MFMA GEMM
+# ---------------------------------------------------------------------------
+class _HIPLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        x2 = x.contiguous()
+        y = ext().linear_fwd(x2, w.contiguous(), b if b is None else b.contiguous())
+        ctx.save_for_backward(x2, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        gy = gy.contiguous()
+        gx = ext().linear_dgrad(gy, w)
+        gw = ext().linear_wgrad(gy, x)
+        gb = gy.sum(0) if ctx.has_bias else None
+        return gx, gw, gb
+
+
+def linear(x, weight, bias=None):
+    if use_hip(x, "linear_fwd"):
+        return _HIPLinear.apply(x, weight, bias)
+    return F.linear(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# Cross-entropy (K12)
+# ---------------------------------------------------------------------------
+class _HIPCrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        logits = logits.contiguous()
+        loss, lse = ext().ce_fwd(logits, target)
+        ctx.save_for_backward(logits, target, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gl):
+        logits, target, lse = ctx.saved_tensors
+        gx = ext().ce_bwd(logits, target, lse, gl.contiguous())
+        return gx, None
+
+
+def cross_entropy(logits, target):
+    if use_hip(logits, "ce_fwd"):
+        return _HIPCrossEntropy.apply(logits, target)
+    return F.cross_entropy(logits, target)
+
+
+# ---------------------------------------------------------------------------
+# Dropout (K19)
+# ---------------------------------------------------------------------------
+def dropout(x, p, training):
+    if p == 0.0 or not training:
+        return x
+    return F.dropout(x, p, training)

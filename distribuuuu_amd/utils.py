@@ -1,0 +1,338 @@
+"""Distributed init, seeding, metrics, LR schedules, and checkpoint I/O.
+
+Capability-parity with the reference's flat utility bag
+(`/root/reference/distribuuuu/utils.py:19-410`), rebuilt MI355X-first:
+
+* ``setup_distributed`` — dual-mode rendezvous (Slurm env or launcher env) into
+  ``init_process_group(backend="nccl")``, which on ROCm IS RCCL over xGMI
+  (reference utils.py:19-51).
+* ``scaled_all_reduce`` — the reference issues one async all-reduce per metric
+  tensor (utils.py:85-106); here the tensors are flattened into ONE RCCL
+  all-reduce launch per call (collective C3 of SURVEY.md §2c, batched).
+* LR schedules, meters, accuracy, checkpoint layout — behavior-equivalent
+  (checkpoint layout is part of the public contract: OUT_DIR/checkpoints/
+  ckpt_ep_{E:03d}.pth.tar, best.pth.tar, auto-resume by newest).
+"""
+
+import math
+import os
+import random
+import subprocess
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from .config import cfg, dump_cfg
+from .logger import logger, setup_logger  # noqa: F401  (re-export)
+
+_DIR_NAME = "checkpoints"
+_NAME_PREFIX = "ckpt_ep_"
+
+
+# ---------------------------------------------------------------------------
+# Distributed init (reference utils.py:19-51)
+# ---------------------------------------------------------------------------
+def setup_distributed(backend="nccl", port=None):
+    """Initialize the process group from Slurm or launcher environment.
+
+    Slurm mode (SLURM_JOB_ID present): derive RANK/WORLD_SIZE from
+    SLURM_PROCID/SLURM_NTASKS and resolve MASTER_ADDR via `scontrol show
+    hostname` on the nodelist (reference utils.py:26-40); default port 29566.
+    Launcher mode: RANK/WORLD_SIZE already exported by torch.distributed.run.
+    """
+    num_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 1
+
+    if "SLURM_JOB_ID" in os.environ:
+        rank = int(os.environ["SLURM_PROCID"])
+        world_size = int(os.environ["SLURM_NTASKS"])
+        node_list = os.environ["SLURM_NODELIST"]
+        addr = subprocess.getoutput(f"scontrol show hostname {node_list} | head -n1")
+        os.environ.setdefault("MASTER_PORT", str(port or 29566))
+        os.environ.setdefault("MASTER_ADDR", addr)
+        os.environ["WORLD_SIZE"] = str(world_size)
+        os.environ["LOCAL_RANK"] = str(rank % num_gpus)
+        os.environ["RANK"] = str(rank)
+    else:
+        rank = int(os.environ.get("RANK", 0))
+        world_size = int(os.environ.get("WORLD_SIZE", 1))
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", str(port or 29566))
+        os.environ.setdefault("RANK", str(rank))
+        os.environ.setdefault("WORLD_SIZE", str(world_size))
+
+    if torch.cuda.is_available():
+        torch.cuda.set_device(rank % num_gpus)
+    else:
+        backend = "gloo"
+
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend,
+            world_size=int(os.environ["WORLD_SIZE"]),
+            rank=int(os.environ["RANK"]),
+        )
+    return int(os.environ["RANK"]), int(os.environ.get("LOCAL_RANK", rank % num_gpus))
+
+
+def get_rank():
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def get_world_size():
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+# ---------------------------------------------------------------------------
+# Seeding / determinism (reference utils.py:54-68)
+# ---------------------------------------------------------------------------
+def setup_seed(rank=0):
+    """Rank 0 creates OUT_DIR and dumps the config; seed with RNG_SEED+rank."""
+    if rank == 0:
+        os.makedirs(cfg.OUT_DIR, exist_ok=True)
+        dump_cfg()
+    if cfg.RNG_SEED is not None:
+        seed = int(cfg.RNG_SEED) + rank
+        np.random.seed(seed)
+        torch.manual_seed(seed)
+        random.seed(seed)
+        torch.backends.cudnn.deterministic = True
+        torch.backends.cudnn.benchmark = False
+    else:
+        torch.backends.cudnn.benchmark = bool(cfg.CUDNN.BENCHMARK)
+        torch.backends.cudnn.deterministic = bool(cfg.CUDNN.DETERMINISTIC)
+
+
+# ---------------------------------------------------------------------------
+# Metric all-reduce (reference utils.py:85-106) — batched into ONE collective
+# ---------------------------------------------------------------------------
+def scaled_all_reduce(tensors):
+    """Average a list of scalar metric tensors across ranks, in place.
+
+    The reference launches one async NCCL all-reduce per tensor; on xGMI the
+    latency of 3 tiny collectives dominates, so we flatten into a single
+    all-reduce and scatter back.
+    """
+    world_size = get_world_size()
+    if world_size == 1:
+        return tensors
+    flat = torch.stack([t.detach().float().reshape(()) for t in tensors])
+    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    flat /= world_size
+    for t, v in zip(tensors, flat):
+        t.copy_(v.to(t.dtype))
+    return tensors
+
+
+# ---------------------------------------------------------------------------
+# Meters (reference utils.py:199-262)
+# ---------------------------------------------------------------------------
+class AverageMeter:
+    """Tracks current value, running average, sum and count."""
+
+    def __init__(self, name, fmt=":f"):
+        self.name = name
+        self.fmt = fmt
+        self.reset()
+
+    def reset(self):
+        self.val = 0.0
+        self.avg = 0.0
+        self.sum = 0.0
+        self.count = 0
+
+    def update(self, val, n=1):
+        self.val = val
+        self.sum += val * n
+        self.count += n
+        self.avg = self.sum / max(self.count, 1)
+
+    def __str__(self):
+        fmtstr = "{name} {val" + self.fmt + "} ({avg" + self.fmt + "})"
+        return fmtstr.format(name=self.name, val=self.val, avg=self.avg)
+
+
+class ProgressMeter:
+    """Joins meters into one display line with an ETA estimate
+    (reference utils.py:227-252: eta = elapsed / ratio_done * ratio_remaining)."""
+
+    def __init__(self, num_batches, meters, prefix=""):
+        self.fmtstr = self._get_batch_fmtstr(num_batches)
+        self.meters = meters
+        self.prefix = prefix
+        self.num_batches = num_batches
+        self.start_time = time.time()
+
+    def display(self, batch):
+        entries = [self.prefix + self.fmtstr.format(batch)]
+        entries += [str(m) for m in self.meters]
+        ratio = max((batch + 1) / max(self.num_batches, 1), 1e-9)
+        elapsed = time.time() - self.start_time
+        eta = elapsed / ratio * (1.0 - ratio)
+        entries.append(f"ETA {int(eta) // 60:02d}:{int(eta) % 60:02d}")
+        logger.info(" | ".join(entries))
+
+    @staticmethod
+    def _get_batch_fmtstr(num_batches):
+        num_digits = len(str(num_batches // 1))
+        f = "{:" + str(num_digits) + "d}"
+        return "[" + f + "/" + f.format(num_batches) + "]"
+
+
+def construct_meters(num_batches, prefix, topk=5):
+    """Time/Data/Loss/Acc@1/Acc@k meter set (reference utils.py:255-262)."""
+    batch_time = AverageMeter("Time", ":.3f")
+    data_time = AverageMeter("Data", ":.3f")
+    losses = AverageMeter("Loss", ":.4e")
+    top1 = AverageMeter("Acc@1", ":6.2f")
+    topk_m = AverageMeter(f"Acc@{topk}", ":6.2f")
+    progress = ProgressMeter(
+        num_batches, [batch_time, data_time, losses, top1, topk_m], prefix=prefix
+    )
+    return batch_time, data_time, losses, top1, topk_m, progress
+
+
+# ---------------------------------------------------------------------------
+# Accuracy (reference utils.py:265-277)
+# ---------------------------------------------------------------------------
+def accuracy(output, target, topk=(1,)):
+    """Top-k accuracy as percentages."""
+    with torch.no_grad():
+        maxk = max(topk)
+        batch_size = target.size(0)
+        _, pred = output.topk(maxk, 1, True, True)
+        pred = pred.t()
+        correct = pred.eq(target.reshape(1, -1).expand_as(pred))
+        res = []
+        for k in topk:
+            correct_k = correct[:k].reshape(-1).float().sum(0, keepdim=True)
+            res.append(correct_k.mul_(100.0 / batch_size))
+        return res
+
+
+# ---------------------------------------------------------------------------
+# LR schedules (reference utils.py:280-316) — pure functions of epoch
+# ---------------------------------------------------------------------------
+def lr_fun_steps(cur_epoch):
+    ind = sum(1 for s in cfg.OPTIM.STEPS if cur_epoch >= s)
+    return cfg.OPTIM.BASE_LR * (cfg.OPTIM.LR_MULT ** ind)
+
+
+def lr_fun_cos(cur_epoch):
+    lr = 0.5 * (1.0 + math.cos(math.pi * cur_epoch / cfg.OPTIM.MAX_EPOCH))
+    return (cfg.OPTIM.BASE_LR - cfg.OPTIM.MIN_LR) * lr + cfg.OPTIM.MIN_LR
+
+
+def get_lr_fun():
+    name = "lr_fun_" + cfg.OPTIM.LR_POLICY
+    if name not in globals():
+        raise NotImplementedError(f"Unknown LR policy: {cfg.OPTIM.LR_POLICY}")
+    return globals()[name]
+
+
+def get_epoch_lr(cur_epoch):
+    """Schedule LR with linear warmup over WARMUP_EPOCHS from WARMUP_FACTOR."""
+    lr = get_lr_fun()(cur_epoch)
+    if cur_epoch < cfg.OPTIM.WARMUP_EPOCHS:
+        alpha = cur_epoch / cfg.OPTIM.WARMUP_EPOCHS
+        warmup_factor = cfg.OPTIM.WARMUP_FACTOR * (1.0 - alpha) + alpha
+        lr *= warmup_factor
+    return lr
+
+
+def set_lr(optimizer, new_lr):
+    for group in optimizer.param_groups:
+        group["lr"] = new_lr
+
+
+# ---------------------------------------------------------------------------
+# Optimizer factory (reference utils.py:187-196)
+# ---------------------------------------------------------------------------
+def construct_optimizer(model):
+    """SGD with momentum/nesterov/weight-decay; on GPU the fused multi-tensor
+    HIP step (ops.fused_sgd) replaces the per-tensor ATen step."""
+    from .ops.optim import HIPSGD
+
+    use_fused = bool(cfg.OPTIM.FUSED_SGD) and torch.cuda.is_available()
+    klass = HIPSGD if use_fused else torch.optim.SGD
+    return klass(
+        model.parameters(),
+        lr=cfg.OPTIM.BASE_LR,
+        momentum=cfg.OPTIM.MOMENTUM,
+        weight_decay=cfg.OPTIM.WEIGHT_DECAY,
+        dampening=cfg.OPTIM.DAMPENING,
+        nesterov=cfg.OPTIM.NESTEROV,
+    )
+
+
+# ---------------------------------------------------------------------------
+# Parameter count (reference utils.py:353-357)
+# ---------------------------------------------------------------------------
+def count_parameters(model):
+    cnt = sum(p.numel() for p in model.parameters() if p.requires_grad)
+    return cnt, cnt * 4 / 1024 ** 2  # fp32 MB, matching the reference's report
+
+
+def unwrap_model(model):
+    """Remove a DDP wrapper if present (reference utils.py:360-363)."""
+    return model.module if hasattr(model, "module") else model
+
+
+# ---------------------------------------------------------------------------
+# Checkpoint I/O (reference utils.py:319-410) — layout is a public contract
+# ---------------------------------------------------------------------------
+def get_checkpoint_dir():
+    return os.path.join(cfg.OUT_DIR, _DIR_NAME)
+
+
+def get_checkpoint(epoch):
+    name = f"{_NAME_PREFIX}{epoch:03d}.pth.tar"
+    return os.path.join(get_checkpoint_dir(), name)
+
+
+def get_last_checkpoint():
+    d = get_checkpoint_dir()
+    names = [f for f in os.listdir(d) if _NAME_PREFIX in f]
+    if not names:
+        raise RuntimeError(f"No checkpoints in {d}")
+    return os.path.join(d, sorted(names)[-1])
+
+
+def has_checkpoint():
+    d = get_checkpoint_dir()
+    return os.path.exists(d) and any(_NAME_PREFIX in f for f in os.listdir(d))
+
+
+def save_checkpoint(model, optimizer, epoch, best_acc1, best=False):
+    """Rank-0-only save; dict {epoch, state_dict (unwrapped), optimizer, best_acc1};
+    best model saved as bare state_dict OUT_DIR/best.pth.tar."""
+    if get_rank() != 0:
+        return None
+    os.makedirs(get_checkpoint_dir(), exist_ok=True)
+    sd = unwrap_model(model).state_dict()
+    checkpoint = {
+        "epoch": epoch,
+        "state_dict": sd,
+        "optimizer": optimizer.state_dict(),
+        "best_acc1": best_acc1,
+    }
+    path = get_checkpoint(epoch + 1)
+    torch.save(checkpoint, path)
+    if best:
+        torch.save(sd, os.path.join(cfg.OUT_DIR, "best.pth.tar"))
+    return path
+
+
+def load_checkpoint(checkpoint_file, model, optimizer=None):
+    """CPU-mapped load tolerant of bare state_dicts and missing optimizer state.
+    Returns the next epoch to run (0 for bare state_dicts)."""
+    ckpt = torch.load(checkpoint_file, map_location="cpu", weights_only=False)
+    target = unwrap_model(model)
+    if isinstance(ckpt, dict) and "state_dict" in ckpt:
+        target.load_state_dict(ckpt["state_dict"])
+        if optimizer is not None and "optimizer" in ckpt and cfg.TRAIN.LOAD_OPT:
+            optimizer.load_state_dict(ckpt["optimizer"])
+        return ckpt.get("epoch", -1) + 1, ckpt.get("best_acc1", 0.0)
+    target.load_state_dict(ckpt)
+    return 0, 0.0

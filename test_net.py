@@ -1,0 +1,19 @@
+"""Evaluation entry point (reference test_net.py:1-13 parity).
+
+Usage:
+    python -m torch.distributed.run --nproc-per-node 8 --master-addr 127.0.0.1 \
+        test_net.py --cfg config/resnet50.yaml MODEL.WEIGHTS ckpt.pth.tar
+"""
+
+import distribuuuu_amd.trainer as trainer
+from distribuuuu_amd.config import cfg, load_cfg_fom_args
+
+
+def main():
+    load_cfg_fom_args("Evaluate a classification model.")
+    cfg.freeze()
+    trainer.test_model()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,32 @@
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires a GPU (run on an MI355X box)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU on this machine")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture(autouse=True)
+def fresh_cfg():
+    """Reset the global config around every test."""
+    from distribuuuu_amd.config import cfg, reset_cfg
+
+    reset_cfg()
+    cfg.defrost()
+    yield cfg
+    reset_cfg()

@@ -1,0 +1,111 @@
+"""LR schedules, meters, accuracy, checkpoint layout (reference utils.py parity)."""
+
+import os
+
+import pytest
+import torch
+
+from distribuuuu_amd import utils
+from distribuuuu_amd.config import cfg
+
+
+def test_lr_cos_schedule():
+    cfg.OPTIM.BASE_LR = 0.2
+    cfg.OPTIM.MAX_EPOCH = 100
+    cfg.OPTIM.WARMUP_EPOCHS = 5
+    cfg.OPTIM.WARMUP_FACTOR = 0.1
+    # warmup start: lr * warmup_factor
+    assert utils.get_epoch_lr(0) == pytest.approx(0.2 * 0.1)
+    # end of warmup reaches full cosine value
+    import math
+
+    lr5 = 0.2 * 0.5 * (1 + math.cos(math.pi * 5 / 100))
+    assert utils.get_epoch_lr(5) == pytest.approx(lr5)
+    # final epoch approaches MIN_LR
+    assert utils.get_epoch_lr(99) < 0.001
+
+
+def test_lr_steps_schedule():
+    cfg.OPTIM.LR_POLICY = "steps"
+    cfg.OPTIM.BASE_LR = 0.1
+    cfg.OPTIM.STEPS = [30, 60]
+    cfg.OPTIM.LR_MULT = 0.1
+    cfg.OPTIM.WARMUP_EPOCHS = 0
+    assert utils.get_epoch_lr(10) == pytest.approx(0.1)
+    assert utils.get_epoch_lr(30) == pytest.approx(0.01)
+    assert utils.get_epoch_lr(60) == pytest.approx(0.001)
+
+
+def test_accuracy():
+    out = torch.tensor([[0.1, 0.9, 0.0], [0.8, 0.1, 0.1]])
+    tgt = torch.tensor([1, 2])
+    a1, a2 = utils.accuracy(out, tgt, topk=(1, 2))
+    assert a1.item() == pytest.approx(50.0)
+    assert a2.item() == pytest.approx(50.0)
+
+
+def test_meters():
+    m = utils.AverageMeter("x", ":.2f")
+    m.update(1.0, 2)
+    m.update(3.0, 2)
+    assert m.avg == pytest.approx(2.0)
+    assert "x" in str(m)
+
+
+def test_scaled_all_reduce_single_rank():
+    t = [torch.tensor(2.0)]
+    out = utils.scaled_all_reduce(t)
+    assert out[0].item() == 2.0
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    cfg.OUT_DIR = str(tmp_path)
+    from distribuuuu_amd import models
+
+    net = models.build_model("resnet18", num_classes=10)
+    opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9)
+    path = utils.save_checkpoint(net, opt, epoch=3, best_acc1=12.3, best=True)
+    assert os.path.basename(path) == "ckpt_ep_004.pth.tar"
+    assert os.path.exists(os.path.join(str(tmp_path), "best.pth.tar"))
+    assert utils.has_checkpoint()
+    assert utils.get_last_checkpoint() == path
+
+    net2 = models.build_model("resnet18", num_classes=10)
+    opt2 = torch.optim.SGD(net2.parameters(), lr=0.1, momentum=0.9)
+    next_epoch, best = utils.load_checkpoint(path, net2, opt2)
+    assert next_epoch == 4
+    assert best == pytest.approx(12.3)
+    for a, b in zip(net.parameters(), net2.parameters()):
+        assert torch.equal(a, b)
+
+
+def test_checkpoint_bare_state_dict(tmp_path):
+    cfg.OUT_DIR = str(tmp_path)
+    from distribuuuu_amd import models
+
+    net = models.build_model("resnet18", num_classes=10)
+    p = tmp_path / "bare.pth.tar"
+    torch.save(net.state_dict(), str(p))
+    net2 = models.build_model("resnet18", num_classes=10)
+    next_epoch, _ = utils.load_checkpoint(str(p), net2)
+    assert next_epoch == 0
+    for a, b in zip(net.parameters(), net2.parameters()):
+        assert torch.equal(a, b)
+
+
+def test_count_parameters():
+    from distribuuuu_amd import models
+
+    n, mb = utils.count_parameters(models.build_model("resnet18"))
+    assert n == 11689512
+
+
+def test_setup_seed_writes_config(tmp_path):
+    cfg.OUT_DIR = str(tmp_path)
+    cfg.RNG_SEED = 1
+    utils.setup_seed(rank=0)
+    assert os.path.exists(os.path.join(str(tmp_path), "config.yaml"))
+    a = torch.randn(3)
+    utils.setup_seed(rank=0)
+    b = torch.randn(3)
+    assert torch.equal(a, b)
