@@ -30,6 +30,17 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
                   at::Tensor gl);
 std::vector<at::Tensor> topk_acc(at::Tensor logits, at::Tensor target,
                                  int64_t topk);
+// conv.hip
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                      int64_t groups);
+at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                        int64_t dh, int64_t dw, int64_t groups);
+at::Tensor gemm_nt(at::Tensor a, at::Tensor b);
+at::Tensor weight_flip_t(at::Tensor w, int64_t groups);
+at::Tensor dilate_nhwc(at::Tensor x, int64_t sh, int64_t sw);
+at::Tensor pad_channels(at::Tensor x, int64_t Cn);
 // sgd.hip
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
@@ -57,4 +68,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("topk_acc", &topk_acc);
   m.def("sgd_step", &sgd_step);
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("gemm_nt", &gemm_nt);
+  m.def("weight_flip_t", &weight_flip_t);
+  m.def("dilate_nhwc", &dilate_nhwc);
+  m.def("pad_channels", &pad_channels);
 }

@@ -1,0 +1,385 @@
+// Implicit-GEMM convolution for gfx950: NHWC bf16, MFMA
+// 16x16x32, LDS-staged double-buffered tiles, fp32 accumulation.
+//
+// GEMM view of conv fwd: C[M, Kg] = A[M, R*S*Cg] @ B[Kg, R*S*Cg]^T per group,
+// where M = N*Ho*Wo output pixels, A rows are gathered input patches (NHWC
+// makes each (r, s-span) contiguous in memory) and B is the weight tensor in
+// its [K][R][S][C] channels_last layout (already reduction-contiguous).
+//
+// The same kernel computes dgrad: conv_fwd(dilate(gy), flipT(w)) — the Python
+// wrapper materializes the transformed weight [C,R,S,K] and zero-dilated gy,
+// and the kernel writes into a larger (HoA, WoA)-strided output region to
+// honor output_padding. Linear layers are the R=S=1, H=W=1 case.
+//
+// Tile: BM=128 x BN=128 x BK=32, 4 waves (2x2), each wave 64x64 via 4x4
+// fragments of mfma_f32_16x16x32_bf16. Register-staged global->LDS with
+// boundary predication (zero-fill); one-step software pipeline (T14 G15:
+// loads for step k+1 issued before compute of step k, LDS write after the
+// barrier).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int LDS_K = BK + 8;  // +8 bf16 pad: break 64B-row bank alignment
+
+struct ConvParams {
+  const __hip_bfloat16* x;  // [N,H,W,Ct] NHWC
+  const __hip_bfloat16* w;  // [Kt,R,S,Cg]
+  __hip_bfloat16* y;        // [N,HoA,WoA,Kt] NHWC
+  int N, H, W, Ct;          // Ct = total input channels
+  int Kt;                   // total output channels
+  int R, S, Cg, Kg;         // per-group
+  int sh, sw, ph, pw, dh, dw;
+  int Ho, Wo;               // logical output (M = N*Ho*Wo)
+  int HoA, WoA;             // allocated output strides (>= Ho,Wo)
+  int M, nspan, ksteps;     // nspan = ceil(S*Cg/BK), ksteps = R*nspan
+  int tiles_m;              // for XCD swizzle
+};
+
+__global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
+  const int g = blockIdx.z;
+  // XCD-aware swizzle over m-tiles (T1; bijective form)
+  int tile_m = blockIdx.x, tile_n = blockIdx.y;
+  {
+    const int nwg = p.tiles_m;
+    const int q = nwg / 8, r8 = nwg % 8;
+    const int xcd = tile_m % 8, idx = tile_m / 8;
+    tile_m = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q) + idx;
+  }
+
+  __shared__ __hip_bfloat16 ldsA[2][BM][LDS_K];
+  __shared__ __hip_bfloat16 ldsB[2][BN][LDS_K];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;  // 2x2 wave grid
+  const int il = lane & 15, kg = lane >> 4;
+
+  // ---- per-thread staging coordinates (fixed across k-steps) ----
+  // Each thread stages 2 rows of A and 2 rows of B per k-step (rows t/4 and
+  // t/4+64), 8 contiguous elements each at span offset (t%4)*8.
+  const int srow = tid >> 2;           // 0..63
+  const int scol8 = (tid & 3) << 3;    // 0,8,16,24
+  int an[2], ahbase[2], awbase[2];
+  bool arow_ok[2];
+#pragma unroll
+  for (int pss = 0; pss < 2; ++pss) {
+    const int m = tile_m * BM + srow + pss * 64;
+    arow_ok[pss] = m < p.M;
+    const int mm = arow_ok[pss] ? m : 0;
+    const int n = mm / (p.Ho * p.Wo);
+    const int rem = mm % (p.Ho * p.Wo);
+    an[pss] = n;
+    ahbase[pss] = (rem / p.Wo) * p.sh - p.ph;
+    awbase[pss] = (rem % p.Wo) * p.sw - p.pw;
+  }
+  const int SCg = p.S * p.Cg;
+  int bk[2];
+  bool brow_ok[2];
+#pragma unroll
+  for (int pss = 0; pss < 2; ++pss) {
+    const int k = tile_n * BN + srow + pss * 64;
+    brow_ok[pss] = k < p.Kg;
+    bk[pss] = g * p.Kg + (brow_ok[pss] ? k : 0);
+  }
+
+  // ---- staging helpers -----------------------------------------------------
+  // loads for k-step ks into registers
+  uint4 regA[2], regB[2];
+  auto stage_load = [&](int ks) {
+    const int r = ks / p.nspan;
+    const int span0 = (ks % p.nspan) * BK;
+    const int span = span0 + scol8;
+    const bool span_ok = span < SCg;
+    const int s = span_ok ? span / p.Cg : 0;
+    const int c = span - s * p.Cg;
+#pragma unroll
+    for (int pss = 0; pss < 2; ++pss) {
+      // A: x[n, hbase + r*dh, wbase + s*dw, gCg + c .. +8]
+      const int h = ahbase[pss] + r * p.dh;
+      const int w_ = awbase[pss] + s * p.dw;
+      const bool ok = arow_ok[pss] && span_ok && h >= 0 && h < p.H &&
+                      w_ >= 0 && w_ < p.W;
+      regA[pss] = ok ? *reinterpret_cast<const uint4*>(
+                           p.x + (((int64_t)an[pss] * p.H + h) * p.W + w_) *
+                                     p.Ct +
+                           g * p.Cg + c)
+                     : uint4{0, 0, 0, 0};
+      // B: w[k, r, span .. +8]
+      const bool okb = brow_ok[pss] && span_ok;
+      regB[pss] = okb ? *reinterpret_cast<const uint4*>(
+                            p.w + ((int64_t)bk[pss] * p.R + r) * SCg + span)
+                      : uint4{0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int pss = 0; pss < 2; ++pss) {
+      *reinterpret_cast<uint4*>(&ldsA[buf][srow + pss * 64][scol8]) = regA[pss];
+      *reinterpret_cast<uint4*>(&ldsB[buf][srow + pss * 64][scol8]) = regB[pss];
+    }
+  };
+
+  // ---- main loop -----------------------------------------------------------
+  f32x4 acc[4][4] = {};
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (p.ksteps > 1) stage_load(1);
+
+  int cur = 0;
+  for (int ks = 0; ks < p.ksteps; ++ks) {
+    // fragments: A rows wm*64 + mi*16 + il, k = kg*8; B rows wn*64 + ni*16+il
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      afrag[mi] = *reinterpret_cast<const bf16x8*>(
+          &ldsA[cur][wm * 64 + mi * 16 + il][kg * 8]);
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+          &ldsB[cur][wn * 64 + ni * 16 + il][kg * 8]);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+    if (ks + 1 < p.ksteps) {
+      stage_write(cur ^ 1);
+      if (ks + 2 < p.ksteps) stage_load(ks + 2);
+      __syncthreads();
+    }
+    cur ^= 1;
+  }
+
+  // ---- epilogue: scalar bf16 stores (C layout: col=lane&15, row=kg*4+reg) --
+  const int HoWo = p.Ho * p.Wo;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int m = tile_m * BM + wm * 64 + mi * 16 + kg * 4 + rr;
+      if (m >= p.M) continue;
+      const int n = m / HoWo;
+      const int rem = m % HoWo;
+      const int64_t obase =
+          (((int64_t)n * p.HoA + rem / p.Wo) * p.WoA + rem % p.Wo) * p.Kt +
+          g * p.Kg;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int k = tile_n * BN + wn * 64 + ni * 16 + il;
+        if (k < p.Kg)
+          p.y[obase + k] = from_f32<__hip_bfloat16>(acc[mi][ni][rr]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
+                           int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
+                           int64_t pw, int64_t dh, int64_t dw, int64_t groups) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16 only");
+  check_nhwc(x, "x");
+  check_nhwc(w, "w");
+  const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(Ct == Cg * groups, "channel/group mismatch");
+  TORCH_CHECK(Cg % 8 == 0, "conv2d_fwd: per-group C must be a multiple of 8");
+  const int Kg = Kt / groups;
+  ConvParams p;
+  p.x = (const __hip_bfloat16*)x.data_ptr();
+  p.w = (const __hip_bfloat16*)w.data_ptr();
+  p.y = (__hip_bfloat16*)y.data_ptr();
+  p.N = N; p.H = H; p.W = W; p.Ct = Ct; p.Kt = Kt;
+  p.R = R; p.S = S; p.Cg = Cg; p.Kg = Kg;
+  p.sh = sh; p.sw = sw; p.ph = ph; p.pw = pw; p.dh = dh; p.dw = dw;
+  p.Ho = Ho; p.Wo = Wo;
+  p.HoA = y.size(2); p.WoA = y.size(3);
+  p.M = N * Ho * Wo;
+  p.nspan = (S * Cg + BK - 1) / BK;
+  p.ksteps = R * p.nspan;
+  p.tiles_m = (p.M + BM - 1) / BM;
+  dim3 grid(p.tiles_m, (Kg + BN - 1) / BN, groups);
+  hipLaunchKernelGGL(conv_igemm_fwd_kernel, grid, dim3(256), 0, cur_stream(),
+                     p);
+  return y;
+}
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                      int64_t groups) {
+  const int N = x.size(0), H = x.size(2), W = x.size(3);
+  const int Kt = w.size(0), R = w.size(2), S = w.size(3);
+  const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
+  const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  auto y = at::empty({N, Kt, Ho, Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups);
+}
+
+// NT GEMM through the conv kernel: y[M, N] = a[M, K] @ b[N, K]^T
+at::Tensor gemm_nt(at::Tensor a, at::Tensor b) {
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm_nt expects 2-D");
+  const int M = a.size(0), K = a.size(1), Nc = b.size(0);
+  // NHWC 4-D with H=W=1: logical [M,K,1,1] channels_last == [M][K] rows
+  auto a_cl = a.reshape({M, K, 1, 1}).contiguous(at::MemoryFormat::ChannelsLast);
+  auto b_cl = b.reshape({Nc, K, 1, 1}).contiguous(at::MemoryFormat::ChannelsLast);
+  auto y = at::empty({M, Nc, 1, 1},
+                     a.options().memory_format(at::MemoryFormat::ChannelsLast));
+  conv2d_fwd_into(a_cl, b_cl, y, 1, 1, 1, 1, 0, 0, 1, 1, 1);
+  return y.reshape({M, Nc});
+}
+
+// ---------------------------------------------------------------------------
+// dgrad transforms
+// ---------------------------------------------------------------------------
+namespace {
+
+// wt[g*Cg + c][r][s][Kg] = w[g*Kg + k][R-1-r][S-1-s][c]  (per group)
+template <typename T>
+__global__ void weight_flip_t_kernel(const T* __restrict__ w, T* __restrict__ wt,
+                                     int Kg, int Cg, int R, int S, int G) {
+  const int64_t total = (int64_t)G * Cg * R * S * Kg;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = i;
+    const int k = t % Kg;
+    t /= Kg;
+    const int s = t % S;
+    t /= S;
+    const int r = t % R;
+    t /= R;
+    const int c = t % Cg;
+    const int g = t / Cg;
+    wt[i] = w[((((int64_t)(g * Kg + k) * R) + (R - 1 - r)) * S + (S - 1 - s)) *
+                  Cg +
+              c];
+  }
+}
+
+// zero-dilate NHWC: out[n, ho*sh, wo*sw, c] = x[n, ho, wo, c]
+template <typename T>
+__global__ void dilate_nhwc_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int N, int Ho, int Wo, int C, int sh,
+                                   int sw, int Hd, int Wd) {
+  const int64_t total = (int64_t)N * Ho * Wo * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int wo = t % Wo;
+    t /= Wo;
+    const int ho = t % Ho;
+    const int n = t / Ho;
+    y[(((int64_t)n * Hd + ho * sh) * Wd + wo * sw) * C + c] = x[i];
+  }
+}
+
+}  // namespace
+
+// weight transform for dgrad: [Kt,Cg,R,S]cl -> [Ct,Kg,R,S]cl flipped
+at::Tensor weight_flip_t(at::Tensor w, int64_t groups) {
+  CHECK_GPU(w);
+  check_nhwc(w, "w");
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Kg = Kt / groups;
+  auto wt = at::empty({Cg * groups, Kg, R, S},
+                      w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)Kt * Cg * R * S;
+  DISPATCH_FLOAT_AND_BF16(w.scalar_type(), "weight_flip_t", [&] {
+    hipLaunchKernelGGL((weight_flip_t_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)w.data_ptr(), (scalar_t*)wt.data_ptr(),
+                       Kg, Cg, R, S, (int)groups);
+  });
+  return wt;
+}
+
+at::Tensor dilate_nhwc(at::Tensor x, int64_t sh, int64_t sw) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), Ho = x.size(2), Wo = x.size(3);
+  const int Hd = (Ho - 1) * sh + 1, Wd = (Wo - 1) * sw + 1;
+  auto y = at::zeros({N, C, Hd, Wd},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * Ho * Wo * C;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "dilate_nhwc", [&] {
+    hipLaunchKernelGGL((dilate_nhwc_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                       N, Ho, Wo, C, sh, sw, Hd, Wd);
+  });
+  return y;
+}
+
+// pad the channel dim (NHWC innermost) with zeros: stem C=3 -> 8 so the
+// MFMA kernel's Cg%8 requirement holds
+namespace {
+template <typename T>
+__global__ void pad_channels_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                    int64_t rows, int C, int Cn) {
+  const int64_t total = rows * Cn;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % Cn;
+    const int64_t r = i / Cn;
+    y[i] = (c < C) ? x[r * C + c] : from_f32<T>(0.f);
+  }
+}
+}  // namespace
+
+at::Tensor pad_channels(at::Tensor x, int64_t Cn) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  auto y = at::empty({N, Cn, H, W},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const int64_t rows = (int64_t)N * H * W;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "pad_channels", [&] {
+    hipLaunchKernelGGL((pad_channels_kernel<scalar_t>),
+                       dim3(grid_1d(rows * Cn, 256)), dim3(256), 0,
+                       cur_stream(), (const scalar_t*)x.data_ptr(),
+                       (scalar_t*)y.data_ptr(), rows, C, (int)Cn);
+  });
+  return y;
+}
+
+// dgrad: gx = conv(dilate(gy), flipT(w)) written into a zeroed [H, W] canvas
+at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                        int64_t dh, int64_t dw, int64_t groups) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int N = gy.size(0);
+  const int Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Ct = Cg * groups;
+  auto wt = weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
+  auto gyd = (sh == 1 && sw == 1) ? gy : dilate_nhwc(gy, sh, sw);
+  const int Hd = gyd.size(2), Wd = gyd.size(3);
+  const int pph = dh * (R - 1) - ph, ppw = dw * (S - 1) - pw;
+  TORCH_CHECK(pph >= 0 && ppw >= 0, "dgrad pad underflow");
+  const int out_h = Hd + 2 * pph - dh * (R - 1);
+  const int out_w = Wd + 2 * ppw - dw * (S - 1);
+  TORCH_CHECK(out_h <= H && out_w <= W, "dgrad output overflow");
+  auto gx = (out_h == H && out_w == W)
+                ? at::empty({N, Ct, H, W}, gy.options().memory_format(
+                                               at::MemoryFormat::ChannelsLast))
+                : at::zeros({N, Ct, H, W}, gy.options().memory_format(
+                                               at::MemoryFormat::ChannelsLast));
+  conv2d_fwd_into(gyd, wt, gx, out_h, out_w, 1, 1, pph, ppw, dh, dw, groups);
+  return gx;
+}

@@ -61,8 +61,6 @@ def hip_op_available(name):
 # by release.
 _DEFAULT_SOFT = {
     "mhsa_relpos_fwd",  # fused BoT attention kernel: pending
-    "conv2d_fwd",       # implicit-GEMM conv: pending
-    "linear_fwd",       # MFMA GEMM fc: pending
 }
 
 

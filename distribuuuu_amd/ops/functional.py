@@ -29,36 +29,64 @@ class _HIPConv2d(torch.autograd.Function):
     def forward(ctx, x, w, bias, stride, padding, dilation, groups):
         x = _cl(x)
         w = _cl(w)
-        y = ext().conv2d_fwd(x, w, stride[0], stride[1], padding[0], padding[1],
-                             dilation[0], dilation[1], groups)
+        e = ext()
+        cin = x.shape[1]
+        xp, wp = x, w
+        if cin < 8 and groups == 1:  # stem C=3: zero-pad channels to 8
+            xp = e.pad_channels(x, 8)
+            wp = e.pad_channels(w, 8)
+        y = e.conv2d_fwd(xp, wp, stride[0], stride[1], padding[0], padding[1],
+                         dilation[0], dilation[1], groups)
         if bias is not None:
             y = y + bias.reshape(1, -1, 1, 1)
         ctx.save_for_backward(x, w)
-        ctx.conf = (stride, padding, dilation, groups, bias is not None)
+        ctx.conf = (stride, padding, dilation, groups, bias is not None, cin)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
-        stride, padding, dilation, groups, has_bias = ctx.conf
+        stride, padding, dilation, groups, has_bias, cin = ctx.conf
         gy = _cl(gy)
+        e = ext()
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            gx = ext().conv2d_dgrad(gy, w, x.shape[2], x.shape[3],
-                                    stride[0], stride[1], padding[0], padding[1],
-                                    dilation[0], dilation[1], groups)
+            if cin < 8 and groups == 1:
+                raise RuntimeError("dgrad through a <8-channel conv "
+                                   "(stem input never requires grad)")
+            gx = e.conv2d_dgrad(gy, w, x.shape[2], x.shape[3],
+                                stride[0], stride[1], padding[0], padding[1],
+                                dilation[0], dilation[1], groups)
         if ctx.needs_input_grad[1]:
-            gw = ext().conv2d_wgrad(gy, x, w.shape[2], w.shape[3],
-                                    stride[0], stride[1], padding[0], padding[1],
-                                    dilation[0], dilation[1], groups)
+            if hasattr(e, "conv2d_wgrad") and (cin >= 8 or groups > 1):
+                gw = e.conv2d_wgrad(gy, x, w.shape[2], w.shape[3],
+                                    stride[0], stride[1], padding[0],
+                                    padding[1], dilation[0], dilation[1],
+                                    groups)
+            else:
+                gw = torch.nn.grad.conv2d_weight(
+                    x, w.shape, gy, stride, padding, dilation, groups)
+                gw = _cl(gw)
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
         return gx, gw, gb, None, None, None, None
 
 
+def _hip_conv_ok(x, weight, groups):
+    if x.dtype != torch.bfloat16:
+        return False
+    cg = weight.shape[1]
+    kg = weight.shape[0] // groups
+    if cg % 8 != 0 and not (cg < 8 and groups == 1):
+        return False  # depthwise & odd group widths: dedicated kernels pending
+    if kg % 8 != 0:
+        return False
+    return True
+
+
 def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
            groups=1):
-    if use_hip(x, "conv2d_fwd"):
+    if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
         return _HIPConv2d.apply(x, weight, bias, stride, padding, dilation, groups)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
 
@@ -252,11 +280,17 @@ class _HIPAvgPool2d(torch.autograd.Function):
 # Linear / fc (K11) — MFMA GEMM
 # ---------------------------------------------------------------------------
 class _HIPLinear(torch.autograd.Function):
+    """fc via the MFMA NT GEMM (K11). The fc is tiny next to the convs; the
+    transposes in backward are ATen copies of <2 MB tensors."""
+
     @staticmethod
     def forward(ctx, x, w, b):
         x2 = x.contiguous()
-        y = ext().linear_fwd(x2, w.contiguous(), b if b is None else b.contiguous())
-        ctx.save_for_backward(x2, w)
+        w2 = w.contiguous()
+        y = ext().gemm_nt(x2, w2)
+        if b is not None:
+            y = y + b
+        ctx.save_for_backward(x2, w2)
         ctx.has_bias = b is not None
         return y
 
@@ -264,14 +298,16 @@ class _HIPLinear(torch.autograd.Function):
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
         gy = gy.contiguous()
-        gx = ext().linear_dgrad(gy, w)
-        gw = ext().linear_wgrad(gy, x)
+        e = ext()
+        gx = e.gemm_nt(gy, w.t().contiguous())
+        gw = e.gemm_nt(gy.t().contiguous(), x.t().contiguous())
         gb = gy.sum(0) if ctx.has_bias else None
         return gx, gw, gb
 
 
 def linear(x, weight, bias=None):
-    if use_hip(x, "linear_fwd"):
+    if (use_hip(x, "gemm_nt") and x.dtype == torch.bfloat16 and x.dim() == 2
+            and x.shape[1] % 8 == 0):
         return _HIPLinear.apply(x, weight, bias)
     return F.linear(x, weight, bias)
 
