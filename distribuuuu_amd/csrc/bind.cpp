@@ -7,6 +7,11 @@ at::Tensor relu_bwd(at::Tensor gy, at::Tensor y);
 at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b);
 // bn.hip
 std::vector<at::Tensor> bn_sums(at::Tensor x);
+std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
+                                 at::Tensor beta,
+                                 c10::optional<at::Tensor> rm_opt,
+                                 c10::optional<at::Tensor> rv_opt,
+                                 double momentum, double eps, bool training);
 at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
                         int64_t act, c10::optional<at::Tensor> res);
 std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
@@ -41,6 +46,9 @@ at::Tensor gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor weight_flip_t(at::Tensor w, int64_t groups);
 at::Tensor dilate_nhwc(at::Tensor x, int64_t sh, int64_t sw);
 at::Tensor pad_channels(at::Tensor x, int64_t Cn);
+at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                        int64_t dh, int64_t dw, int64_t groups);
 // sgd.hip
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
@@ -52,6 +60,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("bn_sums", &bn_sums);
+  m.def("bn_stats", &bn_stats, py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("rm") = py::none(), py::arg("rv") = py::none(),
+        py::arg("momentum") = 0.1, py::arg("eps") = 1e-5,
+        py::arg("training") = true);
   m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
   m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"), py::arg("y"),
@@ -70,6 +82,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("gemm_nt", &gemm_nt);
   m.def("weight_flip_t", &weight_flip_t);
   m.def("dilate_nhwc", &dilate_nhwc);

@@ -3,6 +3,12 @@
 // fused normalize(+residual)(+activation) apply, and backward reductions.
 // Stats and parameters are fp32 regardless of the compute dtype (bf16-accuracy
 // requirement, SURVEY.md §7 hard-part 4).
+//
+// Reduction layout: NHWC rows x C columns; every thread owns one 8-channel
+// pack column for a slab of rows (16-byte loads, coalesced across the whole
+// row; Guideline 13), accumulates in fp32 registers, then one atomicAdd per
+// channel. One finalize kernel computes mean/rstd/scale/shift AND updates the
+// running stats in place — replacing ~8 tiny ATen glue launches per BN layer.
 #include "common_hip.h"
 
 namespace {
@@ -17,8 +23,6 @@ DEV_INLINE float act_apply(float z, int act) {
   }
 }
 
-// dAct/dz expressed with what each case can reach:
-//   relu: from post-act y; silu/sigmoid: from pre-act z.
 DEV_INLINE float act_grad(float y, float z, int act) {
   switch (act) {
     case 1: return y > 0.f ? 1.f : 0.f;
@@ -34,24 +38,134 @@ DEV_INLINE float act_grad(float y, float z, int act) {
   }
 }
 
-// ---- per-channel sum / sum-of-squares over N*H*W rows (NHWC: C fastest) ----
-// Each block owns a slab of rows × all C channels; fp32 atomics merge blocks.
+// ---- per-channel sum / sum-of-squares ------------------------------------
+// thread -> (pack column cp, row lane); block strides over a row slab.
+// Row lanes are LDS-tree-reduced per block so only ONE atomicAdd per channel
+// per block reaches HBM (per-address atomic serialization otherwise dominates).
+DEV_INLINE int p2_floor(int v) { return 1 << (31 - __builtin_clz(v)); }
+
 template <typename T>
 __global__ void bn_sums_kernel(const T* __restrict__ x, float* __restrict__ s,
                                float* __restrict__ ss, int64_t rows, int C,
-                               int rows_per_block) {
+                               int64_t rows_per_block) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
   const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
   const int64_t row1 = min(row0 + rows_per_block, rows);
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float acc = 0.f, acc2 = 0.f;
-    for (int64_t r = row0; r < row1; ++r) {
-      float v = to_f32(x[r * C + c]);
-      acc += v;
-      acc2 += v * v;
+  const P* xp = reinterpret_cast<const P*>(x);
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float acc[V] = {}, acc2[V] = {};
+    if (active) {
+      // 8-deep unroll: 8 independent 16-B loads in flight per thread hides
+      // HBM latency (1 outstanding load caps the chip at ~0.2 TB/s)
+      int64_t r = row0 + rl;
+      for (; r + 7 * (int64_t)nrl < row1; r += 8 * (int64_t)nrl) {
+        P pk[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) pk[u] = xp[(r + u * nrl) * cpacks + cp];
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            float v = to_f32(pk[u].v[j]);
+            acc[j] += v;
+            acc2[j] += v * v;
+          }
+      }
+      for (; r < row1; r += nrl) {
+        P p = xp[r * cpacks + cp];
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float v = to_f32(p.v[j]);
+          acc[j] += v;
+          acc2[j] += v * v;
+        }
+      }
     }
-    atomicAdd(&s[c], acc);
-    atomicAdd(&ss[c], acc2);
+    float* slot = &red[(rl * ncp + cp0) * 2 * V];
+    if (active) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        slot[j] = acc[j];
+        slot[V + j] = acc2[j];
+      }
+    }
+    __syncthreads();
+    for (int st = nrl >> 1; st > 0; st >>= 1) {
+      if (active && rl < st) {
+        const float* other = &red[((rl + st) * ncp + cp0) * 2 * V];
+#pragma unroll
+        for (int j = 0; j < 2 * V; ++j) slot[j] += other[j];
+      }
+      __syncthreads();
+    }
+    if (active && rl == 0) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        atomicAdd(&s[cp * V + j], slot[j]);
+        atomicAdd(&ss[cp * V + j], slot[V + j]);
+      }
+    }
+    __syncthreads();
   }
+}
+
+// ---- finalize: mean/rstd/scale/shift + running-stat update ----------------
+__global__ void bn_finalize_kernel(const float* __restrict__ s,
+                                   const float* __restrict__ ss,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ rstd,
+                                   float* __restrict__ scale,
+                                   float* __restrict__ shift, int C,
+                                   float inv_cnt, float unbias, float momentum,
+                                   float eps, int update_running) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float m = s[c] * inv_cnt;
+  float v = ss[c] * inv_cnt - m * m;
+  v = fmaxf(v, 0.f);
+  const float r = rsqrtf(v + eps);
+  mean[c] = m;
+  rstd[c] = r;
+  const float sc = gamma[c] * r;
+  scale[c] = sc;
+  shift[c] = beta[c] - m * sc;
+  if (update_running) {
+    running_mean[c] = running_mean[c] * (1.f - momentum) + m * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + v * unbias * momentum;
+  }
+}
+
+// eval-mode prep: scale/shift from running stats
+__global__ void bn_eval_prep_kernel(const float* __restrict__ running_mean,
+                                    const float* __restrict__ running_var,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ beta,
+                                    float* __restrict__ mean,
+                                    float* __restrict__ rstd,
+                                    float* __restrict__ scale,
+                                    float* __restrict__ shift, int C,
+                                    float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float m = running_mean[c];
+  const float r = rsqrtf(running_var[c] + eps);
+  mean[c] = m;
+  rstd[c] = r;
+  const float sc = gamma[c] * r;
+  scale[c] = sc;
+  shift[c] = beta[c] - m * sc;
 }
 
 // ---- fused apply: y = act(x*scale[c] + shift[c] (+ res)) -------------------
@@ -82,34 +196,119 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 }
 
 // ---- backward reduction: per-channel sum(g), sum(g * xhat) -----------------
-// g = gy * act'(...); xhat = (x - mean) * rstd
+// g = gy * act'(...); xhat = (x - mean) * rstd. Same pack-column layout.
 template <typename T, bool HAS_RES>
 __global__ void bn_bwd_reduce_kernel(
     const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
     const T* __restrict__ res, const float* __restrict__ mean,
     const float* __restrict__ rstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ sum_g,
-    float* __restrict__ sum_gxh, int64_t rows, int C, int rows_per_block,
+    float* __restrict__ sum_gxh, int64_t rows, int C, int64_t rows_per_block,
     int act) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
   const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
   const int64_t row1 = min(row0 + rows_per_block, rows);
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    const float m = mean[c], r = rstd[c], w = gamma[c], b = beta[c];
-    float acc = 0.f, acc2 = 0.f;
-    for (int64_t i = row0; i < row1; ++i) {
-      const int64_t off = i * C + c;
-      float xh = (to_f32(x[off]) - m) * r;
-      float g = to_f32(gy[off]);
-      if (act != 0) {
-        float z = xh * w + b;
-        if (HAS_RES) z += to_f32(res[off]);
-        g *= act_grad(to_f32(y[off]), z, act);
+  const P* gp = reinterpret_cast<const P*>(gy);
+  const P* xp = reinterpret_cast<const P*>(x);
+  const P* ypk = reinterpret_cast<const P*>(y);
+  const P* rp = reinterpret_cast<const P*>(res);
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float accg[V] = {}, accgx[V] = {};
+    if (active) {
+      float m[V], r[V], w[V], b[V];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const int c = cp * V + j;
+        m[j] = mean[c];
+        r[j] = rstd[c];
+        w[j] = gamma[c];
+        b[j] = beta[c];
       }
-      acc += g;
-      acc2 += g * xh;
+      // 4-deep unroll x 2-3 tensors = 8-12 loads in flight per thread
+      int64_t row = row0 + rl;
+      for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+        P pg4[4], px4[4], py4[4], pr4[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int64_t off = (row + u * nrl) * cpacks + cp;
+          pg4[u] = gp[off];
+          px4[u] = xp[off];
+          if (act != 0) py4[u] = ypk[off];
+          if (HAS_RES && act >= 2) pr4[u] = rp[off];
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            float xh = (to_f32(px4[u].v[j]) - m[j]) * r[j];
+            float g = to_f32(pg4[u].v[j]);
+            if (act != 0) {
+              float z = 0.f;
+              if (act >= 2) {
+                z = xh * w[j] + b[j];
+                if (HAS_RES) z += to_f32(pr4[u].v[j]);
+              }
+              g *= act_grad(to_f32(py4[u].v[j]), z, act);
+            }
+            accg[j] += g;
+            accgx[j] += g * xh;
+          }
+      }
+      for (; row < row1; row += nrl) {
+        const int64_t off = row * cpacks + cp;
+        P pg = gp[off], px = xp[off], py, prr;
+        if (act != 0) py = ypk[off];
+        if (HAS_RES && act >= 2) prr = rp[off];
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float xh = (to_f32(px.v[j]) - m[j]) * r[j];
+          float g = to_f32(pg.v[j]);
+          if (act != 0) {
+            float z = 0.f;
+            if (act >= 2) {
+              z = xh * w[j] + b[j];
+              if (HAS_RES) z += to_f32(prr.v[j]);
+            }
+            g *= act_grad(to_f32(py.v[j]), z, act);
+          }
+          accg[j] += g;
+          accgx[j] += g * xh;
+        }
+      }
     }
-    atomicAdd(&sum_g[c], acc);
-    atomicAdd(&sum_gxh[c], acc2);
+    float* slot = &red[(rl * ncp + cp0) * 2 * V];
+    if (active) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        slot[j] = accg[j];
+        slot[V + j] = accgx[j];
+      }
+    }
+    __syncthreads();
+    for (int st = nrl >> 1; st > 0; st >>= 1) {
+      if (active && rl < st) {
+        const float* other = &red[((rl + st) * ncp + cp0) * 2 * V];
+#pragma unroll
+        for (int j = 0; j < 2 * V; ++j) slot[j] += other[j];
+      }
+      __syncthreads();
+    }
+    if (active && rl == 0) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        atomicAdd(&sum_g[cp * V + j], slot[j]);
+        atomicAdd(&sum_gxh[cp * V + j], slot[V + j]);
+      }
+    }
+    __syncthreads();
   }
 }
 
@@ -136,7 +335,7 @@ __global__ void bn_bwd_dx_kernel(
     const int c0 = (int)(i % cpacks) * V;
     P pg = gp[i], px = xp[i], py, pr;
     if (act != 0) py = ypk[i];
-    if (HAS_RES && act != 0) pr = rp[i];
+    if (HAS_RES && act >= 2) pr = rp[i];
     P ox, orr;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -145,8 +344,11 @@ __global__ void bn_bwd_dx_kernel(
       float xh = (to_f32(px.v[j]) - m) * r;
       float g = to_f32(pg.v[j]);
       if (act != 0) {
-        float z = xh * w + beta[c];
-        if (HAS_RES) z += to_f32(pr.v[j]);
+        float z = 0.f;
+        if (act >= 2) {
+          z = xh * w + beta[c];
+          if (HAS_RES) z += to_f32(pr.v[j]);
+        }
         g *= act_grad(to_f32(py.v[j]), z, act);
       }
       if (HAS_RES) orr.v[j] = from_f32<T>(g);
@@ -162,9 +364,10 @@ __global__ void bn_bwd_dx_kernel(
   }
 }
 
-int pick_rows_per_block(int64_t rows) {
-  // target ~1024 reduction blocks
-  return (int)std::max<int64_t>(ceil_div(rows, 1024), 8);
+int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
+  // target ~2048 reduction blocks (8 per CU), slab a multiple of the per-iter row count
+  int64_t rpb = std::max<int64_t>(ceil_div(rows, 2048), rows_per_iter);
+  return ceil_div(rpb, rows_per_iter) * rows_per_iter;
 }
 
 template <typename scalar_t, int V, bool HR, bool TR>
@@ -196,14 +399,62 @@ std::vector<at::Tensor> bn_sums(at::Tensor x) {
   auto opts = x.options().dtype(at::kFloat);
   auto s = at::zeros({C}, opts);
   auto ss = at::zeros({C}, opts);
-  int rpb = pick_rows_per_block(rows);
-  int grid = (int)ceil_div(rows, rpb);
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_sums", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int grid = (int)ceil_div(rows, rpb);
     hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(grid), dim3(256), 0,
                        cur_stream(), (const scalar_t*)x.data_ptr(),
                        s.data_ptr<float>(), ss.data_ptr<float>(), rows, C, rpb);
   });
   return {s, ss};
+}
+
+// One-shot stats: sums -> (mean, rstd, scale, shift) + running update.
+// Returns {mean, rstd, scale, shift}.
+std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
+                                 at::Tensor beta,
+                                 c10::optional<at::Tensor> rm_opt,
+                                 c10::optional<at::Tensor> rv_opt,
+                                 double momentum, double eps, bool training) {
+  at::Tensor running_mean = rm_opt.has_value() ? *rm_opt : at::Tensor();
+  at::Tensor running_var = rv_opt.has_value() ? *rv_opt : at::Tensor();
+  const int C = x.size(1);
+  auto opts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({C}, opts);
+  auto rstd = at::empty({C}, opts);
+  auto scale = at::empty({C}, opts);
+  auto shift = at::empty({C}, opts);
+  const int grid = (int)ceil_div(C, 256);
+  if (training) {
+    auto sums = bn_sums(x);
+    const int64_t rows = x.numel() / C;
+    const float inv_cnt = 1.f / (float)rows;
+    const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(grid), dim3(256), 0,
+                       cur_stream(), sums[0].data_ptr<float>(),
+                       sums[1].data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(),
+                       running_mean.defined() ? running_mean.data_ptr<float>()
+                                              : nullptr,
+                       running_var.defined() ? running_var.data_ptr<float>()
+                                             : nullptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(), C,
+                       inv_cnt, unbias, (float)momentum, (float)eps,
+                       running_mean.defined() ? 1 : 0);
+  } else {
+    hipLaunchKernelGGL(bn_eval_prep_kernel, dim3(grid), dim3(256), 0,
+                       cur_stream(), running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), C, (float)eps);
+  }
+  return {mean, rstd, scale, shift};
 }
 
 at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
@@ -256,32 +507,31 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
     constexpr int V = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
     auto stream = cur_stream();
-    int rpb = pick_rows_per_block(rows);
-    int rgrid = (int)ceil_div(rows, rpb);
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
-    {  // grad-stat reduction runs in train AND eval (gw/gb need it)
-      if (has_res)
-        hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true>), dim3(rgrid),
-                           dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                           (const scalar_t*)x.data_ptr(),
-                           (const scalar_t*)y.data_ptr(), resp,
-                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                           sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
-                           rows, C, rpb, (int)act);
-      else
-        hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
-                           dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                           (const scalar_t*)x.data_ptr(),
-                           (const scalar_t*)y.data_ptr(), resp,
-                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                           sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
-                           rows, C, rpb, (int)act);
-    }
+    if (has_res)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true>), dim3(rgrid),
+                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), resp,
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                         sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
+                         rows, C, rpb, (int)act);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
+                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), resp,
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                         sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
+                         rows, C, rpb, (int)act);
     int64_t npacks = x.numel() / V;
-    int cpacks = C / V;
     int grid = grid_1d(npacks, 256);
     float inv_cnt = 1.f / (float)rows;
     scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;

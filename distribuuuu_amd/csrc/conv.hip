@@ -314,8 +314,9 @@ at::Tensor dilate_nhwc(at::Tensor x, int64_t sh, int64_t sw) {
   check_nhwc(x, "x");
   const int N = x.size(0), C = x.size(1), Ho = x.size(2), Wo = x.size(3);
   const int Hd = (Ho - 1) * sh + 1, Wd = (Wo - 1) * sw + 1;
-  auto y = at::zeros({N, C, Hd, Wd},
+  auto y = at::empty({N, C, Hd, Wd},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  y.zero_();
   int64_t total = (int64_t)N * Ho * Wo * C;
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "dilate_nhwc", [&] {
     hipLaunchKernelGGL((dilate_nhwc_kernel<scalar_t>),
@@ -372,14 +373,13 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
   const int Hd = gyd.size(2), Wd = gyd.size(3);
   const int pph = dh * (R - 1) - ph, ppw = dw * (S - 1) - pw;
   TORCH_CHECK(pph >= 0 && ppw >= 0, "dgrad pad underflow");
-  const int out_h = Hd + 2 * pph - dh * (R - 1);
-  const int out_w = Wd + 2 * ppw - dw * (S - 1);
-  TORCH_CHECK(out_h <= H && out_w <= W, "dgrad output overflow");
-  auto gx = (out_h == H && out_w == W)
-                ? at::empty({N, Ct, H, W}, gy.options().memory_format(
-                                               at::MemoryFormat::ChannelsLast))
-                : at::zeros({N, Ct, H, W}, gy.options().memory_format(
-                                               at::MemoryFormat::ChannelsLast));
-  conv2d_fwd_into(gyd, wt, gx, out_h, out_w, 1, 1, pph, ppw, dh, dw, groups);
+  // Every input row/col receives its own output: run the conv at logical
+  // output size exactly (H, W); the kernel's bounds predication supplies the
+  // zero-padding of the dilated gy on both ends (incl. rows past the last
+  // forward window, where forward's floor division dropped input).
+  (void)Hd; (void)Wd;
+  auto gx = at::empty({N, Ct, H, W}, gy.options().memory_format(
+                                           at::MemoryFormat::ChannelsLast));
+  conv2d_fwd_into(gyd, wt, gx, H, W, 1, 1, pph, ppw, dh, dw, groups);
   return gx;
 }

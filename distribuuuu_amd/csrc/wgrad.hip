@@ -1,0 +1,212 @@
+// Weight-gradient implicit GEMM: gw[k][r][s][c] =
+// sum_m gy[m][k] * patch(x)[m][r][s][c], reduction over all output pixels m.
+//
+// Both operands are m-major in memory (gy NHWC rows, x NHWC gather), while the
+// MFMA wants the reduction (m) per-lane-contiguous — tiles are staged into LDS
+// TRANSPOSED ([k][m], [rsc][m]) by element scatter-writes, fragments then read
+// contiguously. Split-m: each block owns a pixel chunk and atomically
+// accumulates its fp32 partial tile into gw_accum[Kt][R*S*Cg]; a final cast
+// kernel produces the bf16 weight grad in channels_last layout.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4w;
+typedef __bf16 bf16x8w __attribute__((ext_vector_type(8)));
+
+namespace {
+
+constexpr int WBM = 64;   // k rows per tile
+constexpr int WBN = 64;   // rsc cols per tile
+constexpr int WBK = 64;   // pixels per k-step
+constexpr int WLDS = WBK + 8;  // m-stride with pad
+constexpr int CHUNK_STEPS = 64;  // k-steps per block chunk (4096 pixels)
+
+struct WgradParams {
+  const __hip_bfloat16* x;   // [N,H,W,Ct]
+  const __hip_bfloat16* gy;  // [N,Ho,Wo,Kt]
+  float* acc;                // [Kt, R*S*Cg] zeroed
+  int N, H, W, Ct, Kt;
+  int R, S, Cg, Kg;
+  int sh, sw, ph, pw, dh, dw;
+  int Ho, Wo;
+  int M, RSC;
+  int ktiles, ntiles, chunks;
+};
+
+__global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
+  const int g = blockIdx.z;
+  const int ktile = blockIdx.x % p.ktiles;
+  const int ntile = blockIdx.x / p.ktiles;
+  const int chunk = blockIdx.y;
+
+  __shared__ __hip_bfloat16 ldsA[2][WBM][WLDS];  // [k][m]
+  __shared__ __hip_bfloat16 ldsB[2][WBN][WLDS];  // [rsc][m]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int il = lane & 15, kq = lane >> 4;
+
+  // staging: thread t loads 8 contiguous k (or c) at pixel m_local:
+  //   m_local = t/8 (+32), col8 = (t%8)*8
+  const int sml = tid >> 3;        // 0..31
+  const int scol8 = (tid & 7) << 3;
+  const int SCg = p.S * p.Cg;
+  // B-side decomposition of this thread's rsc range
+  const int rsc = ntile * WBN + scol8;
+  const bool rsc_ok = rsc < p.RSC;
+  const int br = rsc_ok ? rsc / SCg : 0;
+  const int brem = rsc - br * SCg;
+  const int bs = brem / p.Cg;
+  const int bc = brem - bs * p.Cg;
+  const int kcol = ktile * WBM + scol8;
+  const bool k_ok = kcol < p.Kg;
+
+  const int m0 = chunk * (CHUNK_STEPS * WBK);
+  const int HoWo = p.Ho * p.Wo;
+  const int ksteps =
+      min(CHUNK_STEPS, (p.M - m0 + WBK - 1) / WBK);
+
+  __hip_bfloat16 regA[2][8], regB[2][8];
+  auto stage_load = [&](int ks) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int m = m0 + ks * WBK + sml + half * 32;
+      const bool m_ok = m < p.M;
+      const int mm = m_ok ? m : 0;
+      const int n = mm / HoWo;
+      const int rem = mm - n * HoWo;
+      const int ho = rem / p.Wo, wo = rem - (rem / p.Wo) * p.Wo;
+      // gy[m][kcol..+8]
+      if (m_ok && k_ok) {
+        *reinterpret_cast<uint4*>(&regA[half][0]) =
+            *reinterpret_cast<const uint4*>(
+                p.gy + ((int64_t)mm * p.Kt) + g * p.Kg + kcol);
+      } else {
+        *reinterpret_cast<uint4*>(&regA[half][0]) = uint4{0, 0, 0, 0};
+      }
+      // x[n, ho*sh-ph+r*dh, wo*sw-pw+s*dw, gCg + c..+8]
+      const int h = ho * p.sh - p.ph + br * p.dh;
+      const int w_ = wo * p.sw - p.pw + bs * p.dw;
+      if (m_ok && rsc_ok && h >= 0 && h < p.H && w_ >= 0 && w_ < p.W) {
+        *reinterpret_cast<uint4*>(&regB[half][0]) =
+            *reinterpret_cast<const uint4*>(
+                p.x + (((int64_t)n * p.H + h) * p.W + w_) * p.Ct + g * p.Cg +
+                bc);
+      } else {
+        *reinterpret_cast<uint4*>(&regB[half][0]) = uint4{0, 0, 0, 0};
+      }
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int m = sml + half * 32;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        ldsA[buf][scol8 + i][m] = regA[half][i];
+        ldsB[buf][scol8 + i][m] = regB[half][i];
+      }
+    }
+  };
+
+  f32x4w accv[2][2] = {};  // 2x2 fragments of 16x16 per wave
+
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (ksteps > 1) stage_load(1);
+
+  int cur = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {  // two 32-deep m sub-steps
+      bf16x8w afrag[2], bfrag[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        afrag[mi] = *reinterpret_cast<const bf16x8w*>(
+            &ldsA[cur][wm * 32 + mi * 16 + il][mc * 32 + kq * 8]);
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bfrag[ni] = *reinterpret_cast<const bf16x8w*>(
+            &ldsB[cur][wn * 32 + ni * 16 + il][mc * 32 + kq * 8]);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], accv[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      stage_write(cur ^ 1);
+      if (ks + 2 < ksteps) stage_load(ks + 2);
+      __syncthreads();
+    }
+    cur ^= 1;
+  }
+  // epilogue: fp32 atomic accumulate (D: col=lane&15, row=(lane>>4)*4+rr)
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int k = ktile * WBM + wm * 32 + mi * 16 + kq * 4 + rr;
+      if (k >= p.Kg) continue;
+      const int64_t rowbase = (int64_t)(g * p.Kg + k) * p.RSC;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int col = ntile * WBN + wn * 32 + ni * 16 + il;
+        if (col < p.RSC) atomicAdd(&p.acc[rowbase + col], accv[mi][ni][rr]);
+      }
+    }
+  }
+}
+
+__global__ void cast_acc_kernel(const float* __restrict__ acc,
+                                __hip_bfloat16* __restrict__ gw,
+                                int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x)
+    gw[i] = from_f32<__hip_bfloat16>(acc[i]);
+}
+
+}  // namespace
+
+at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                        int64_t dh, int64_t dw, int64_t groups) {
+  CHECK_GPU(gy);
+  TORCH_CHECK(gy.scalar_type() == at::kBFloat16, "wgrad: bf16 only");
+  check_nhwc(gy, "gy");
+  check_nhwc(x, "x");
+  const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
+  const int Kt = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
+  const int Cg = Ct / groups, Kg = Kt / groups;
+  TORCH_CHECK(Cg % 8 == 0 && Kg % 8 == 0, "wgrad: Cg/Kg must be multiples of 8");
+  WgradParams p;
+  p.x = (const __hip_bfloat16*)x.data_ptr();
+  p.gy = (const __hip_bfloat16*)gy.data_ptr();
+  p.N = N; p.H = H; p.W = W; p.Ct = Ct; p.Kt = Kt;
+  p.R = R; p.S = S; p.Cg = Cg; p.Kg = Kg;
+  p.sh = sh; p.sw = sw; p.ph = ph; p.pw = pw; p.dh = dh; p.dw = dw;
+  p.Ho = Ho; p.Wo = Wo;
+  p.M = N * Ho * Wo;
+  p.RSC = R * S * Cg;
+  p.ktiles = (Kg + WBM - 1) / WBM;
+  p.ntiles = (p.RSC + WBN - 1) / WBN;
+  p.chunks = (p.M + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK);
+
+  auto accbuf = at::empty({(int64_t)Kt, p.RSC}, x.options().dtype(at::kFloat));
+  accbuf.zero_();
+  p.acc = accbuf.data_ptr<float>();
+  dim3 grid(p.ktiles * p.ntiles, p.chunks, groups);
+  hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(256), 0, cur_stream(), p);
+
+  auto gw = at::empty({Kt, Cg, (int64_t)R, (int64_t)S},
+                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const int64_t total = (int64_t)Kt * p.RSC;
+  hipLaunchKernelGGL(cast_acc_kernel, dim3(grid_1d(total, 256)), dim3(256), 0,
+                     cur_stream(), p.acc, (__hip_bfloat16*)gw.data_ptr(),
+                     total);
+  return gw;
+}

@@ -58,15 +58,15 @@ class _HIPConv2d(torch.autograd.Function):
                                 stride[0], stride[1], padding[0], padding[1],
                                 dilation[0], dilation[1], groups)
         if ctx.needs_input_grad[1]:
-            if hasattr(e, "conv2d_wgrad") and (cin >= 8 or groups > 1):
-                gw = e.conv2d_wgrad(gy, x, w.shape[2], w.shape[3],
-                                    stride[0], stride[1], padding[0],
-                                    padding[1], dilation[0], dilation[1],
-                                    groups)
-            else:
-                gw = torch.nn.grad.conv2d_weight(
-                    x, w.shape, gy, stride, padding, dilation, groups)
-                gw = _cl(gw)
+            xw = x
+            if cin < 8 and groups == 1:  # stem: wgrad on the zero-padded input
+                xw = e.pad_channels(x, 8)
+            gw = e.conv2d_wgrad(gy, xw, w.shape[2], w.shape[3],
+                                stride[0], stride[1], padding[0],
+                                padding[1], dilation[0], dilation[1],
+                                groups)
+            if cin < 8 and groups == 1:
+                gw = _cl(gw[:, :cin])
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
         return gx, gw, gb, None, None, None, None
@@ -105,25 +105,20 @@ class _HIPBatchNormAct(torch.autograd.Function):
         x = _cl(x)
         e = ext()
         act_id = _ACTS[act]
-        c = x.shape[1]
-        if training:
-            s, ss = e.bn_sums(x)  # fp32 [C]
-            n = x.numel() / c
-            mean = s / n
-            var = (ss / n - mean * mean).clamp_(min=0)
-            if running_mean is not None:
-                with torch.no_grad():
-                    unbiased = var * (n / max(n - 1, 1))
-                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                    running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
-        else:
-            mean = running_mean.float()
-            var = running_var.float()
-        rstd = (var + eps).rsqrt()
-        gamma = weight.float()
-        beta = bias.float()
-        scale = gamma * rstd
-        shift = beta - mean * scale
+        gamma = weight.float().contiguous()
+        beta = bias.float().contiguous()
+        rm, rv = running_mean, running_var
+        copy_back = False
+        if rm is not None and rm.dtype != torch.float32:
+            rm, rv = rm.float(), rv.float()
+            copy_back = True
+        # one fused kernel: sums -> mean/rstd/scale/shift + running update
+        mean, rstd, scale, shift = e.bn_stats(
+            x, gamma, beta, rm, rv, momentum, eps, training)
+        if copy_back:
+            with torch.no_grad():
+                running_mean.copy_(rm)
+                running_var.copy_(rv)
         res = _cl(residual) if residual is not None else None
         y = e.bn_apply_act(x, scale, shift, act_id, res)
         ctx.save_for_backward(x, gamma, beta, mean, rstd, y,

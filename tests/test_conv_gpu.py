@@ -93,6 +93,24 @@ def test_conv2d_dgrad(case):
     assert err < 2e-2 * max(scale, 1.0), f"{case}: err {err} scale {scale}"
 
 
+@pytest.mark.parametrize("case", CONV_CASES)
+def test_conv2d_wgrad(case):
+    e = _ext()
+    n, c, h, w_, k, r, s, p, g = case
+    torch.manual_seed(5)
+    x = _cl(torch.randn(n, c, h, w_, device="cuda", dtype=torch.bfloat16))
+    ho = (h + 2 * p - r) // s + 1
+    wo = (w_ + 2 * p - r) // s + 1
+    gy = _cl(torch.randn(n, k, ho, wo, device="cuda", dtype=torch.bfloat16))
+    gw = e.conv2d_wgrad(gy, x, r, r, s, s, p, p, 1, 1, g)
+    wf = torch.zeros(k, c // g, r, r, device="cuda", requires_grad=True)
+    F.conv2d(x.float(), wf, None, s, p, 1, g).backward(gy.float())
+    err = (gw.float() - wf.grad).abs().max().item()
+    scale = wf.grad.abs().max().item()
+    # wgrad reduces over N*Ho*Wo in bf16 products: relative tolerance vs scale
+    assert err < 3e-2 * max(scale, 1.0), f"{case}: err {err} scale {scale}"
+
+
 def test_conv2d_stem_with_pad_channels():
     e = _ext()
     torch.manual_seed(3)
