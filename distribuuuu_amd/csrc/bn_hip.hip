@@ -107,19 +107,48 @@ __global__ void bn_sums_kernel(const T* __restrict__ x, float* __restrict__ s,
       __syncthreads();
     }
     if (active && rl == 0) {
+      // partials: [block][2C]  (deterministic; no atomic serialization)
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        atomicAdd(&s[cp * V + j], slot[j]);
-        atomicAdd(&ss[cp * V + j], slot[V + j]);
+        s[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
+        s[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
       }
     }
     __syncthreads();
   }
+  (void)ss;
+}
+
+// ---- stage 2: column-reduce the per-block partials -------------------------
+// 256 threads = 64 channels x 4 row-lanes; each lane strides the partial
+// rows 4-deep-unrolled, then an LDS reduce folds the 4 lanes.
+__global__ void reduce_partials_kernel(const float* __restrict__ part,
+                                       float* __restrict__ out, int width,
+                                       int nblocks) {
+  __shared__ float red[256];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rlane = threadIdx.x >> 6;
+  float acc = 0.f;
+  if (c < width) {
+    int b = rlane;
+    for (; b + 12 < nblocks; b += 16) {
+      acc += part[(int64_t)b * width + c] +
+             part[(int64_t)(b + 4) * width + c] +
+             part[(int64_t)(b + 8) * width + c] +
+             part[(int64_t)(b + 12) * width + c];
+    }
+    for (; b < nblocks; b += 4) acc += part[(int64_t)b * width + c];
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  if (rlane == 0 && c < width)
+    out[c] = red[threadIdx.x] + red[threadIdx.x + 64] +
+             red[threadIdx.x + 128] + red[threadIdx.x + 192];
 }
 
 // ---- finalize: mean/rstd/scale/shift + running-stat update ----------------
-__global__ void bn_finalize_kernel(const float* __restrict__ s,
-                                   const float* __restrict__ ss,
+__global__ void bn_finalize_kernel(const float* __restrict__ part,
+                                   int nblocks,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ running_mean,
@@ -130,10 +159,26 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s,
                                    float* __restrict__ shift, int C,
                                    float inv_cnt, float unbias, float momentum,
                                    float eps, int update_running) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float m = s[c] * inv_cnt;
-  float v = ss[c] * inv_cnt - m * m;
+  __shared__ float red[512];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rlane = threadIdx.x >> 6;
+  float sum = 0.f, sumsq = 0.f;
+  if (c < C) {
+    for (int b = rlane; b < nblocks; b += 4) {
+      sum += part[(int64_t)b * 2 * C + c];
+      sumsq += part[(int64_t)b * 2 * C + C + c];
+    }
+  }
+  red[threadIdx.x] = sum;
+  red[256 + threadIdx.x] = sumsq;
+  __syncthreads();
+  if (rlane != 0 || c >= C) return;
+  sum = red[threadIdx.x] + red[threadIdx.x + 64] + red[threadIdx.x + 128] +
+        red[threadIdx.x + 192];
+  sumsq = red[256 + threadIdx.x] + red[256 + threadIdx.x + 64] +
+          red[256 + threadIdx.x + 128] + red[256 + threadIdx.x + 192];
+  const float m = sum * inv_cnt;
+  float v = sumsq * inv_cnt - m * m;
   v = fmaxf(v, 0.f);
   const float r = rsqrtf(v + eps);
   mean[c] = m;
@@ -179,8 +224,28 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
   const P* xp = reinterpret_cast<const P*>(x);
   const P* rp = reinterpret_cast<const P*>(res);
   P* yp = reinterpret_cast<P*>(y);
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < npacks;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + 3 * stride < npacks; i += 4 * stride) {
+    P px4[4], pr4[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      px4[u] = xp[i + u * stride];
+      if (HAS_RES) pr4[u] = rp[i + u * stride];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int c0 = (int)((i + u * stride) % cpacks) * V;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float z = to_f32(px4[u].v[j]) * scale[c0 + j] + shift[c0 + j];
+        if (HAS_RES) z += to_f32(pr4[u].v[j]);
+        px4[u].v[j] = from_f32<T>(act_apply(z, act));
+      }
+      yp[i + u * stride] = px4[u];
+    }
+  }
+  for (; i < npacks; i += stride) {
     const int c0 = (int)(i % cpacks) * V;
     P px = xp[i];
     P pr;
@@ -304,12 +369,13 @@ __global__ void bn_bwd_reduce_kernel(
     if (active && rl == 0) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        atomicAdd(&sum_g[cp * V + j], slot[j]);
-        atomicAdd(&sum_gxh[cp * V + j], slot[V + j]);
+        sum_g[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
+        sum_g[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
       }
     }
     __syncthreads();
   }
+  (void)sum_gxh;
 }
 
 // ---- backward apply: gx (+ gres) -------------------------------------------
@@ -330,12 +396,9 @@ __global__ void bn_bwd_dx_kernel(
   const P* rp = reinterpret_cast<const P*>(res);
   P* oxp = reinterpret_cast<P*>(gx);
   P* orp = reinterpret_cast<P*>(gres);
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < npacks;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  auto body = [&](int64_t i, const P& pg, const P& px, const P& py,
+                  const P& pr) {
     const int c0 = (int)(i % cpacks) * V;
-    P pg = gp[i], px = xp[i], py, pr;
-    if (act != 0) py = ypk[i];
-    if (HAS_RES && act >= 2) pr = rp[i];
     P ox, orr;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -361,12 +424,34 @@ __global__ void bn_bwd_dx_kernel(
     }
     oxp[i] = ox;
     if (HAS_RES) orp[i] = orr;
+  };
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + 3 * stride < npacks; i += 4 * stride) {
+    P pg4[4], px4[4], py4[4], pr4[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int64_t k = i + u * stride;
+      pg4[u] = gp[k];
+      px4[u] = xp[k];
+      if (act != 0) py4[u] = ypk[k];
+      if (HAS_RES && act >= 2) pr4[u] = rp[k];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) body(i + u * stride, pg4[u], px4[u], py4[u], pr4[u]);
+  }
+  for (; i < npacks; i += stride) {
+    P pg = gp[i], px = xp[i], py, pr;
+    if (act != 0) py = ypk[i];
+    if (HAS_RES && act >= 2) pr = rp[i];
+    body(i, pg, px, py, pr);
   }
 }
 
 int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
-  // target ~2048 reduction blocks (8 per CU), slab a multiple of the per-iter row count
-  int64_t rpb = std::max<int64_t>(ceil_div(rows, 2048), rows_per_iter);
+  // 512 stage-1 blocks: BW-saturating with the 8-deep unroll while keeping
+  // the stage-2 partial reduction short
+  int64_t rpb = std::max<int64_t>(ceil_div(rows, 512), rows_per_iter);
   return ceil_div(rpb, rows_per_iter) * rows_per_iter;
 }
 
@@ -397,8 +482,8 @@ std::vector<at::Tensor> bn_sums(at::Tensor x) {
   const int C = x.size(1);
   const int64_t rows = x.numel() / C;
   auto opts = x.options().dtype(at::kFloat);
-  auto s = at::zeros({C}, opts);
-  auto ss = at::zeros({C}, opts);
+  auto s = at::empty({C}, opts);
+  auto ss = at::empty({C}, opts);
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_sums", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
@@ -406,11 +491,41 @@ std::vector<at::Tensor> bn_sums(at::Tensor x) {
     const int nrl = std::max(256 / cpacks, 1);
     const int64_t rpb = pick_rows_per_block(rows, nrl);
     const int grid = (int)ceil_div(rows, rpb);
+    auto part = at::empty({grid, 2 * C}, opts);
     hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(grid), dim3(256), 0,
                        cur_stream(), (const scalar_t*)x.data_ptr(),
-                       s.data_ptr<float>(), ss.data_ptr<float>(), rows, C, rpb);
+                       part.data_ptr<float>(), nullptr, rows, C, rpb);
+    auto both = at::empty({2 * C}, opts);
+    hipLaunchKernelGGL(reduce_partials_kernel, dim3(ceil_div(2 * C, 64)),
+                       dim3(256), 0, cur_stream(), part.data_ptr<float>(),
+                       both.data_ptr<float>(), 2 * C, grid);
+    s = both.narrow(0, 0, C);
+    ss = both.narrow(0, C, C);
   });
   return {s, ss};
+}
+
+// stage-1 only: per-block partials [grid, 2C] (shared by bn_stats)
+static std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
+  const int C = x.size(1);
+  const int64_t rows = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  at::Tensor part;
+  int grid_out = 0;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_partials", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int grid = (int)ceil_div(rows, rpb);
+    part = at::empty({grid, 2 * C}, opts);
+    hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(grid), dim3(256), 0,
+                       cur_stream(), (const scalar_t*)x.data_ptr(),
+                       part.data_ptr<float>(), nullptr, rows, C, rpb);
+    grid_out = grid;
+  });
+  return {part, grid_out};
 }
 
 // One-shot stats: sums -> (mean, rstd, scale, shift) + running update.
@@ -428,15 +543,15 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
   auto rstd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
-  const int grid = (int)ceil_div(C, 256);
+  const int grid = (int)ceil_div(C, 64);
   if (training) {
-    auto sums = bn_sums(x);
+    auto pg = bn_partials(x);
     const int64_t rows = x.numel() / C;
     const float inv_cnt = 1.f / (float)rows;
     const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(grid), dim3(256), 0,
-                       cur_stream(), sums[0].data_ptr<float>(),
-                       sums[1].data_ptr<float>(), gamma.data_ptr<float>(),
+                       cur_stream(), pg.first.data_ptr<float>(),
+                       pg.second, gamma.data_ptr<float>(),
                        beta.data_ptr<float>(),
                        running_mean.defined() ? running_mean.data_ptr<float>()
                                               : nullptr,
@@ -497,8 +612,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
   const int C = x.size(1);
   const int64_t rows = x.numel() / C;
   auto fopts = x.options().dtype(at::kFloat);
-  auto sum_g = at::zeros({C}, fopts);
-  auto sum_gxh = at::zeros({C}, fopts);
+  at::Tensor sum_g, sum_gxh;
   const bool has_res = res.has_value();
   auto gx = at::empty_like(x);
   auto gres = need_gres ? at::empty_like(x) : at::Tensor();
@@ -511,6 +625,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
     const int nrl = std::max(256 / cpacks, 1);
     const int64_t rpb = pick_rows_per_block(rows, nrl);
     const int rgrid = (int)ceil_div(rows, rpb);
+    auto part = at::empty({rgrid, 2 * C}, fopts);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
     if (has_res)
@@ -520,7 +635,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                          (const scalar_t*)y.data_ptr(), resp,
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                         sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
+                         part.data_ptr<float>(), nullptr,
                          rows, C, rpb, (int)act);
     else
       hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
@@ -529,8 +644,14 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                          (const scalar_t*)y.data_ptr(), resp,
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                         sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
+                         part.data_ptr<float>(), nullptr,
                          rows, C, rpb, (int)act);
+    auto both = at::empty({2 * C}, fopts);
+    hipLaunchKernelGGL(reduce_partials_kernel, dim3(ceil_div(2 * C, 64)),
+                       dim3(256), 0, stream, part.data_ptr<float>(),
+                       both.data_ptr<float>(), 2 * C, rgrid);
+    sum_g = both.narrow(0, 0, C);
+    sum_gxh = both.narrow(0, C, C);
     int64_t npacks = x.numel() / V;
     int grid = grid_1d(npacks, 256);
     float inv_cnt = 1.f / (float)rows;
