@@ -49,6 +49,13 @@ at::Tensor pad_channels(at::Tensor x, int64_t Cn);
 at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw,
                         int64_t dh, int64_t dw, int64_t groups);
+// depthwise.hip
+at::Tensor dwconv_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw);
+at::Tensor dwconv_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw);
+at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 // sgd.hip
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
@@ -87,4 +94,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("weight_flip_t", &weight_flip_t);
   m.def("dilate_nhwc", &dilate_nhwc);
   m.def("pad_channels", &pad_channels);
+  m.def("dwconv_fwd", &dwconv_fwd);
+  m.def("dwconv_dgrad", &dwconv_dgrad);
+  m.def("dwconv_wgrad", &dwconv_wgrad);
 }

@@ -37,6 +37,7 @@ struct ConvParams {
   int sh, sw, ph, pw, dh, dw;
   int Ho, Wo;               // logical output (M = N*Ho*Wo)
   int HoA, WoA;             // allocated output strides (>= Ho,Wo)
+  int osh, osw;             // output position multipliers (dgrad-s2 scatter)
   int M, nspan, ksteps;     // nspan = ceil(S*Cg/BK), ksteps = R*nspan
   int tiles_m;              // for XCD swizzle
 };
@@ -171,8 +172,8 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
       const int n = m / HoWo;
       const int rem = m % HoWo;
       const int64_t obase =
-          (((int64_t)n * p.HoA + rem / p.Wo) * p.WoA + rem % p.Wo) * p.Kt +
-          g * p.Kg;
+          (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh) * p.WoA +
+           (rem % p.Wo) * p.osw) * p.Kt + g * p.Kg;
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int k = tile_n * BN + wn * 64 + ni * 16 + il;
@@ -190,7 +191,8 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
 // ---------------------------------------------------------------------------
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
-                           int64_t pw, int64_t dh, int64_t dw, int64_t groups) {
+                           int64_t pw, int64_t dh, int64_t dw, int64_t groups,
+                           int64_t osh = 1, int64_t osw = 1) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16 only");
   check_nhwc(x, "x");
@@ -209,6 +211,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   p.sh = sh; p.sw = sw; p.ph = ph; p.pw = pw; p.dh = dh; p.dw = dw;
   p.Ho = Ho; p.Wo = Wo;
   p.HoA = y.size(2); p.WoA = y.size(3);
+  p.osh = osh; p.osw = osw;
   p.M = N * Ho * Wo;
   p.nspan = (S * Cg + BK - 1) / BK;
   p.ksteps = R * p.nspan;
@@ -369,6 +372,15 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
   const int Cg = w.size(1), R = w.size(2), S = w.size(3);
   const int Ct = Cg * groups;
   auto wt = weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
+  if (R == 1 && S == 1 && (sh > 1 || sw > 1) && ph == 0 && pw == 0) {
+    // strided-output GEMM: gx[ho*sh, wo*sw] = gy[ho, wo] @ w^T, rest zero
+    auto gx1 = at::empty({N, Ct, H, W}, gy.options().memory_format(
+                                            at::MemoryFormat::ChannelsLast));
+    gx1.zero_();
+    conv2d_fwd_into(gy, wt, gx1, gy.size(2), gy.size(3), 1, 1, 0, 0, 1, 1,
+                    groups, sh, sw);
+    return gx1;
+  }
   auto gyd = (sh == 1 && sw == 1) ? gy : dilate_nhwc(gy, sh, sw);
   const int Hd = gyd.size(2), Wd = gyd.size(3);
   const int pph = dh * (R - 1) - ph, ppw = dw * (S - 1) - pw;

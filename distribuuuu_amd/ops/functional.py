@@ -72,6 +72,42 @@ class _HIPConv2d(torch.autograd.Function):
         return gx, gw, gb, None, None, None, None
 
 
+class _HIPDepthwiseConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, padding):
+        x = _cl(x)
+        w = _cl(w)
+        y = ext().dwconv_fwd(x, w, stride[0], stride[1], padding[0], padding[1])
+        if bias is not None:
+            y = y + bias.reshape(1, -1, 1, 1)
+        ctx.save_for_backward(x, w)
+        ctx.conf = (stride, padding, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        stride, padding, has_bias = ctx.conf
+        gy = _cl(gy)
+        e = ext()
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = e.dwconv_dgrad(gy, w, x.shape[2], x.shape[3], stride[0],
+                                stride[1], padding[0], padding[1])
+        if ctx.needs_input_grad[1]:
+            gw = e.dwconv_wgrad(gy, x, w.shape[2], w.shape[3], stride[0],
+                                stride[1], padding[0], padding[1])
+        if has_bias and ctx.needs_input_grad[2]:
+            gb = gy.sum(dim=(0, 2, 3))
+        return gx, gw, gb, None, None
+
+
+def _is_depthwise(x, weight, groups, dilation):
+    return (groups > 1 and groups == x.shape[1] and weight.shape[0] == groups
+            and weight.shape[1] == 1 and dilation == (1, 1)
+            and x.shape[1] % 8 == 0)
+
+
 def _hip_conv_ok(x, weight, groups):
     if x.dtype != torch.bfloat16:
         return False
@@ -86,6 +122,10 @@ def _hip_conv_ok(x, weight, groups):
 
 def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
            groups=1):
+    stride, padding, dilation = tuple(stride), tuple(padding), tuple(dilation)
+    if (x.dtype == torch.bfloat16 and _is_depthwise(x, weight, groups, dilation)
+            and use_hip(x, "dwconv_fwd")):
+        return _HIPDepthwiseConv2d.apply(x, weight, bias, stride, padding)
     if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
         return _HIPConv2d.apply(x, weight, bias, stride, padding, dilation, groups)
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)

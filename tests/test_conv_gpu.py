@@ -146,6 +146,51 @@ def test_conv_autograd_function_end_to_end():
         assert err < 3e-2 * max(ref.abs().max().item(), 1.0), err
 
 
+@pytest.mark.parametrize("case", [(2, 32, 28, 28, 3, 1), (2, 96, 28, 28, 5, 2),
+                                  (2, 144, 14, 14, 3, 1)])
+def test_depthwise_conv(case):
+    """Depthwise fwd/dgrad/wgrad vs fp32 grouped ATen conv."""
+    e = _ext()
+    n, c, h, w_, k, s = case
+    p = k // 2
+    torch.manual_seed(6)
+    x = _cl(torch.randn(n, c, h, w_, device="cuda", dtype=torch.bfloat16))
+    w = _cl(torch.randn(c, 1, k, k, device="cuda", dtype=torch.bfloat16))
+    y = e.dwconv_fwd(x, w, s, s, p, p)
+    ref = F.conv2d(x.float(), w.float(), None, s, p, 1, c)
+    assert (y.float() - ref).abs().max() < 2e-2 * max(ref.abs().max().item(), 1)
+
+    gy = _cl(torch.randn_like(y))
+    gx = e.dwconv_dgrad(gy, w, h, w_, s, s, p, p)
+    gw = e.dwconv_wgrad(gy, x, k, k, s, s, p, p)
+    xr = x.float().detach().requires_grad_(True)
+    wr = w.float().detach().requires_grad_(True)
+    F.conv2d(xr, wr, None, s, p, 1, c).backward(gy.float())
+    assert (gx.float() - xr.grad).abs().max() < 2e-2 * max(
+        xr.grad.abs().max().item(), 1)
+    assert (gw.float() - wr.grad).abs().max() < 3e-2 * max(
+        wr.grad.abs().max().item(), 1)
+
+
+def test_efficientnet_forward_backward_gpu():
+    from distribuuuu_amd import models
+
+    torch.manual_seed(7)
+    m = models.build_model("efficientnet_b0", num_classes=10)
+    m = m.to("cuda").to(torch.bfloat16)
+    for mod in m.modules():
+        if hasattr(mod, "running_mean"):
+            mod.float()
+    m = m.to(memory_format=torch.channels_last)
+    x = torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    y = m(x)
+    assert y.shape == (4, 10)
+    y.float().sum().backward()
+    assert all(p.grad is not None for p in m.parameters())
+    torch.cuda.synchronize()
+
+
 def test_dilate_and_weight_flip():
     e = _ext()
     x = _cl(torch.arange(2 * 8 * 3 * 3, device="cuda", dtype=torch.bfloat16)
