@@ -342,7 +342,7 @@ class _HIPLinear(torch.autograd.Function):
 
 def linear(x, weight, bias=None):
     if (use_hip(x, "gemm_nt") and x.dtype == torch.bfloat16 and x.dim() == 2
-            and x.shape[1] % 8 == 0):
+            and x.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
         return _HIPLinear.apply(x, weight, bias)
     return F.linear(x, weight, bias)
 
