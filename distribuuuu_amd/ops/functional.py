@@ -51,12 +51,12 @@ class _HIPConv2d(torch.autograd.Function):
         e = ext()
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            if cin < 8 and groups == 1:
-                raise RuntimeError("dgrad through a <8-channel conv "
-                                   "(stem input never requires grad)")
-            gx = e.conv2d_dgrad(gy, w, x.shape[2], x.shape[3],
+            wd = e.pad_channels(w, 8) if (cin < 8 and groups == 1) else w
+            gx = e.conv2d_dgrad(gy, wd, x.shape[2], x.shape[3],
                                 stride[0], stride[1], padding[0], padding[1],
                                 dilation[0], dilation[1], groups)
+            if cin < 8 and groups == 1:
+                gx = _cl(gx[:, :cin])
         if ctx.needs_input_grad[1]:
             xw = x
             if cin < 8 and groups == 1:  # stem: wgrad on the zero-padded input
