@@ -17,8 +17,8 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
 std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
-                               at::Tensor beta, int64_t act, bool training,
-                               bool need_gres);
+                               at::Tensor scale, at::Tensor shift,
+                               int64_t act, bool training, bool need_gres);
 // pool.hip
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
                                     int64_t P);
@@ -75,8 +75,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
   m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"), py::arg("y"),
         py::arg("res"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
-        py::arg("beta"), py::arg("act"), py::arg("training"),
-        py::arg("need_gres"));
+        py::arg("scale"), py::arg("shift"), py::arg("act"),
+        py::arg("training"), py::arg("need_gres"));
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("gap_fwd", &gap_fwd);

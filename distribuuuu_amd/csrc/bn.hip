@@ -3,11 +3,16 @@
 // Stats and parameters are fp32 regardless of the compute dtype (bf16-accuracy
 // requirement, SURVEY.md §7 hard-part 4).
 //
-// Reduction layout: NHWC rows x C columns; every thread owns one 8-channel
-// pack column for a slab of rows (16-byte loads, coalesced across the whole
-// row; Guideline 13), accumulates in fp32 registers, then one atomicAdd per
-// channel. One finalize kernel computes mean/rstd/scale/shift AND updates the
-// running stats in place — replacing ~8 tiny ATen glue launches per BN layer.
+// Structure (all deterministic — no fp32 atomics):
+//   fwd:  bn_sums_kernel (per-block partials, 8-deep MLP unroll)
+//         -> bn_finalize_kernel (partial reduce + mean/rstd/scale/shift +
+//            running-stat update, one launch)
+//         -> bn_apply_kernel (y = act(x*scale + shift + res))
+//   bwd:  bn_bwd_reduce_kernel (RAW per-channel sums sum(g), sum(g*x);
+//         no per-channel parameter gathers on the hot path)
+//         -> bn_bwd_finalize_kernel (tiny: -> gw, gb and the dx coefficients
+//            P1, P2, P3 with gx = P1*g + P3*x + P2)
+//         -> bn_bwd_dx_kernel (3 coefficient gathers per pack)
 #include "common.h"
 
 namespace {
@@ -37,15 +42,14 @@ DEV_INLINE float act_grad(float y, float z, int act) {
   }
 }
 
-// ---- per-channel sum / sum-of-squares ------------------------------------
-// thread -> (pack column cp, row lane); block strides over a row slab.
-// Row lanes are LDS-tree-reduced per block so only ONE atomicAdd per channel
-// per block reaches HBM (per-address atomic serialization otherwise dominates).
 DEV_INLINE int p2_floor(int v) { return 1 << (31 - __builtin_clz(v)); }
 
+// ---- fwd stage 1: per-channel sum / sum-of-squares partials ----------------
+// thread -> (pack column cp, row lane rl); 8-deep unrolled row loop for MLP;
+// LDS tree over row lanes; block writes its partial row [2C].
 template <typename T>
-__global__ void bn_sums_kernel(const T* __restrict__ x, float* __restrict__ s,
-                               float* __restrict__ ss, int64_t rows, int C,
+__global__ void bn_sums_kernel(const T* __restrict__ x,
+                               float* __restrict__ part, int64_t rows, int C,
                                int64_t rows_per_block) {
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
@@ -62,8 +66,6 @@ __global__ void bn_sums_kernel(const T* __restrict__ x, float* __restrict__ s,
   for (int cp = cp0; cp < cpacks; cp += ncp) {
     float acc[V] = {}, acc2[V] = {};
     if (active) {
-      // 8-deep unroll: 8 independent 16-B loads in flight per thread hides
-      // HBM latency (1 outstanding load caps the chip at ~0.2 TB/s)
       int64_t r = row0 + rl;
       for (; r + 7 * (int64_t)nrl < row1; r += 8 * (int64_t)nrl) {
         P pk[8];
@@ -106,21 +108,18 @@ __global__ void bn_sums_kernel(const T* __restrict__ x, float* __restrict__ s,
       __syncthreads();
     }
     if (active && rl == 0) {
-      // partials: [block][2C]  (deterministic; no atomic serialization)
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        s[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
-        s[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
+        part[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
+        part[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
       }
     }
     __syncthreads();
   }
-  (void)ss;
 }
 
-// ---- stage 2: column-reduce the per-block partials -------------------------
-// 256 threads = 64 channels x 4 row-lanes; each lane strides the partial
-// rows 4-deep-unrolled, then an LDS reduce folds the 4 lanes.
+// ---- generic stage 2: column-reduce partials [nblocks, width] -> [width] ---
+// 256 threads = 64 channels x 4 row-lanes.
 __global__ void reduce_partials_kernel(const float* __restrict__ part,
                                        float* __restrict__ out, int width,
                                        int nblocks) {
@@ -145,9 +144,8 @@ __global__ void reduce_partials_kernel(const float* __restrict__ part,
              red[threadIdx.x + 128] + red[threadIdx.x + 192];
 }
 
-// ---- finalize: mean/rstd/scale/shift + running-stat update ----------------
-__global__ void bn_finalize_kernel(const float* __restrict__ part,
-                                   int nblocks,
+// ---- fwd finalize: partial reduce + mean/rstd/scale/shift + running update -
+__global__ void bn_finalize_kernel(const float* __restrict__ part, int nblocks,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ running_mean,
@@ -259,16 +257,15 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
   }
 }
 
-// ---- backward reduction: per-channel sum(g), sum(g * xhat) -----------------
-// g = gy * act'(...); xhat = (x - mean) * rstd. Same pack-column layout.
+// ---- bwd stage 1: RAW per-channel sums sum(g), sum(g*x) --------------------
+// g = gy * act'; act' needs only y (relu) or z = scale*x+shift(+res)
+// (silu/sigmoid). No mean/rstd/gamma gathers on the hot path.
 template <typename T, bool HAS_RES>
 __global__ void bn_bwd_reduce_kernel(
     const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
-    const T* __restrict__ res, const float* __restrict__ mean,
-    const float* __restrict__ rstd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, float* __restrict__ sum_g,
-    float* __restrict__ sum_gxh, int64_t rows, int C, int64_t rows_per_block,
-    int act) {
+    const T* __restrict__ res, const float* __restrict__ scale,
+    const float* __restrict__ shift, float* __restrict__ part, int64_t rows,
+    int C, int64_t rows_per_block, int act) {
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
@@ -285,18 +282,16 @@ __global__ void bn_bwd_reduce_kernel(
   const P* ypk = reinterpret_cast<const P*>(y);
   const P* rp = reinterpret_cast<const P*>(res);
   for (int cp = cp0; cp < cpacks; cp += ncp) {
-    float accg[V] = {}, accgx[V] = {};
-    if (active) {
-      float m[V], r[V], w[V], b[V];
+    float sc[V], sh[V];
+    if (act >= 2) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        const int c = cp * V + j;
-        m[j] = mean[c];
-        r[j] = rstd[c];
-        w[j] = gamma[c];
-        b[j] = beta[c];
+        sc[j] = scale[cp * V + j];
+        sh[j] = shift[cp * V + j];
       }
-      // 4-deep unroll x 2-3 tensors = 8-12 loads in flight per thread
+    }
+    float accg[V] = {}, accgx[V] = {};
+    if (active) {
       int64_t row = row0 + rl;
       for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
         P pg4[4], px4[4], py4[4], pr4[4];
@@ -312,18 +307,18 @@ __global__ void bn_bwd_reduce_kernel(
         for (int u = 0; u < 4; ++u)
 #pragma unroll
           for (int j = 0; j < V; ++j) {
-            float xh = (to_f32(px4[u].v[j]) - m[j]) * r[j];
+            float xv = to_f32(px4[u].v[j]);
             float g = to_f32(pg4[u].v[j]);
             if (act != 0) {
               float z = 0.f;
               if (act >= 2) {
-                z = xh * w[j] + b[j];
+                z = xv * sc[j] + sh[j];
                 if (HAS_RES) z += to_f32(pr4[u].v[j]);
               }
               g *= act_grad(to_f32(py4[u].v[j]), z, act);
             }
             accg[j] += g;
-            accgx[j] += g * xh;
+            accgx[j] += g * xv;
           }
       }
       for (; row < row1; row += nrl) {
@@ -333,18 +328,18 @@ __global__ void bn_bwd_reduce_kernel(
         if (HAS_RES && act >= 2) prr = rp[off];
 #pragma unroll
         for (int j = 0; j < V; ++j) {
-          float xh = (to_f32(px.v[j]) - m[j]) * r[j];
+          float xv = to_f32(px.v[j]);
           float g = to_f32(pg.v[j]);
           if (act != 0) {
             float z = 0.f;
             if (act >= 2) {
-              z = xh * w[j] + b[j];
+              z = xv * sc[j] + sh[j];
               if (HAS_RES) z += to_f32(prr.v[j]);
             }
             g *= act_grad(to_f32(py.v[j]), z, act);
           }
           accg[j] += g;
-          accgx[j] += g * xh;
+          accgx[j] += g * xv;
         }
       }
     }
@@ -368,26 +363,68 @@ __global__ void bn_bwd_reduce_kernel(
     if (active && rl == 0) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        sum_g[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
-        sum_g[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
+        part[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
+        part[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
       }
     }
     __syncthreads();
   }
-  (void)sum_gxh;
 }
 
-// ---- backward apply: gx (+ gres) -------------------------------------------
-// training: gx = gamma*rstd * (g - (sum_g + xhat*sum_gxh)/count)
-// eval:     gx = gamma*rstd * g
-template <typename T, int V, bool HAS_RES, bool TRAINING>
+// ---- bwd finalize: partials -> gw, gb and the dx coefficients --------------
+// sum_gxh = rstd*(sum(g*x) - mean*sum(g)); gw = sum_gxh, gb = sum(g)
+// training: gx = w*r*(g - (sum_g + xhat*sum_gxh)/cnt) = P1*g + P3*x + P2 with
+//   P1 = w*r; P3 = -w*r^2*sum_gxh/cnt; P2 = -P1*sum_g/cnt - P3*mean
+// eval: P1 = w*r; P2 = P3 = 0.
+__global__ void bn_bwd_finalize_kernel(
+    const float* __restrict__ part, int nblocks,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, float* __restrict__ gw,
+    float* __restrict__ gb, float* __restrict__ P1, float* __restrict__ P2,
+    float* __restrict__ P3, int C, float inv_cnt, int training) {
+  __shared__ float red[512];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rlane = threadIdx.x >> 6;
+  float sg = 0.f, sgx = 0.f;
+  if (c < C) {
+    for (int b = rlane; b < nblocks; b += 4) {
+      sg += part[(int64_t)b * 2 * C + c];
+      sgx += part[(int64_t)b * 2 * C + C + c];
+    }
+  }
+  red[threadIdx.x] = sg;
+  red[256 + threadIdx.x] = sgx;
+  __syncthreads();
+  if (rlane != 0 || c >= C) return;
+  sg = red[threadIdx.x] + red[threadIdx.x + 64] + red[threadIdx.x + 128] +
+       red[threadIdx.x + 192];
+  sgx = red[256 + threadIdx.x] + red[256 + threadIdx.x + 64] +
+        red[256 + threadIdx.x + 128] + red[256 + threadIdx.x + 192];
+  const float m = mean[c], r = rstd[c], w = gamma[c];
+  const float sum_gxh = r * (sgx - m * sg);
+  gw[c] = sum_gxh;
+  gb[c] = sg;
+  const float p1 = w * r;
+  P1[c] = p1;
+  if (training) {
+    const float p3 = -w * r * r * sum_gxh * inv_cnt;
+    P3[c] = p3;
+    P2[c] = -p1 * sg * inv_cnt - p3 * m;
+  } else {
+    P3[c] = 0.f;
+    P2[c] = 0.f;
+  }
+}
+
+// ---- bwd dx: gx = P1*g + P3*x + P2 (+ gres = g) ----------------------------
+template <typename T, int V, bool HAS_RES>
 __global__ void bn_bwd_dx_kernel(
     const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
-    const T* __restrict__ res, const float* __restrict__ mean,
-    const float* __restrict__ rstd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, const float* __restrict__ sum_g,
-    const float* __restrict__ sum_gxh, T* __restrict__ gx,
-    T* __restrict__ gres, int64_t npacks, int cpacks, int act, float inv_cnt) {
+    const T* __restrict__ res, const float* __restrict__ scale,
+    const float* __restrict__ shift, const float* __restrict__ P1c,
+    const float* __restrict__ P2c, const float* __restrict__ P3c,
+    T* __restrict__ gx, T* __restrict__ gres, int64_t npacks, int cpacks,
+    int act) {
   using P = Pack<T, V>;
   const P* gp = reinterpret_cast<const P*>(gy);
   const P* xp = reinterpret_cast<const P*>(x);
@@ -395,117 +432,61 @@ __global__ void bn_bwd_dx_kernel(
   const P* rp = reinterpret_cast<const P*>(res);
   P* oxp = reinterpret_cast<P*>(gx);
   P* orp = reinterpret_cast<P*>(gres);
-  auto body = [&](int64_t i, const P& pg, const P& px, const P& py,
-                  const P& pr) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < npacks;
+       i += stride) {
     const int c0 = (int)(i % cpacks) * V;
+    P pg = gp[i], px = xp[i], py, pr;
+    if (act != 0) py = ypk[i];
+    if (HAS_RES && act >= 2) pr = rp[i];
     P ox, orr;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
       const int c = c0 + j;
-      const float m = mean[c], r = rstd[c], w = gamma[c];
-      float xh = (to_f32(px.v[j]) - m) * r;
+      float xv = to_f32(px.v[j]);
       float g = to_f32(pg.v[j]);
       if (act != 0) {
         float z = 0.f;
         if (act >= 2) {
-          z = xh * w + beta[c];
+          z = xv * scale[c] + shift[c];
           if (HAS_RES) z += to_f32(pr.v[j]);
         }
         g *= act_grad(to_f32(py.v[j]), z, act);
       }
       if (HAS_RES) orr.v[j] = from_f32<T>(g);
-      float v;
-      if (TRAINING)
-        v = w * r * (g - (sum_g[c] + xh * sum_gxh[c]) * inv_cnt);
-      else
-        v = w * r * g;
-      ox.v[j] = from_f32<T>(v);
+      ox.v[j] = from_f32<T>(P1c[c] * g + P3c[c] * xv + P2c[c]);
     }
     oxp[i] = ox;
     if (HAS_RES) orp[i] = orr;
-  };
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  for (; i + 3 * stride < npacks; i += 4 * stride) {
-    P pg4[4], px4[4], py4[4], pr4[4];
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      const int64_t k = i + u * stride;
-      pg4[u] = gp[k];
-      px4[u] = xp[k];
-      if (act != 0) py4[u] = ypk[k];
-      if (HAS_RES && act >= 2) pr4[u] = rp[k];
-    }
-#pragma unroll
-    for (int u = 0; u < 4; ++u) body(i + u * stride, pg4[u], px4[u], py4[u], pr4[u]);
-  }
-  for (; i < npacks; i += stride) {
-    P pg = gp[i], px = xp[i], py, pr;
-    if (act != 0) py = ypk[i];
-    if (HAS_RES && act >= 2) pr = rp[i];
-    body(i, pg, px, py, pr);
   }
 }
 
 int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
-  // 512 stage-1 blocks: BW-saturating with the 8-deep unroll while keeping
-  // the stage-2 partial reduction short
+  // 512 stage-1 blocks: BW-saturating with the MLP unroll while keeping the
+  // stage-2 partial reduction short
   int64_t rpb = std::max<int64_t>(ceil_div(rows, 512), rows_per_iter);
   return ceil_div(rpb, rows_per_iter) * rows_per_iter;
 }
 
-template <typename scalar_t, int V, bool HR, bool TR>
+template <typename scalar_t, int V, bool HR>
 void launch_dx(const at::Tensor& gy, const at::Tensor& x, const at::Tensor& y,
-               const scalar_t* resp, const at::Tensor& mean,
-               const at::Tensor& rstd, const at::Tensor& gamma,
-               const at::Tensor& beta, const at::Tensor& sum_g,
-               const at::Tensor& sum_gxh, at::Tensor& gx, scalar_t* gresp,
-               int64_t npacks, int cpacks, int act, float inv_cnt, int grid,
+               const scalar_t* resp, const at::Tensor& scale,
+               const at::Tensor& shift, const at::Tensor& P1,
+               const at::Tensor& P2, const at::Tensor& P3, at::Tensor& gx,
+               scalar_t* gresp, int64_t npacks, int cpacks, int act, int grid,
                hipStream_t stream) {
-  hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, V, HR, TR>), dim3(grid),
+  hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, V, HR>), dim3(grid),
                      dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
                      (const scalar_t*)x.data_ptr(),
                      (const scalar_t*)y.data_ptr(), resp,
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     sum_g.data_ptr<float>(), sum_gxh.data_ptr<float>(),
-                     (scalar_t*)gx.data_ptr(), gresp, npacks, cpacks, act,
-                     inv_cnt);
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),
+                     P1.data_ptr<float>(), P2.data_ptr<float>(),
+                     P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,
+                     npacks, cpacks, act);
 }
 
-}  // namespace
-
-std::vector<at::Tensor> bn_sums(at::Tensor x) {
-  CHECK_GPU(x);
-  check_nhwc(x, "x");
-  const int C = x.size(1);
-  const int64_t rows = x.numel() / C;
-  auto opts = x.options().dtype(at::kFloat);
-  auto s = at::empty({C}, opts);
-  auto ss = at::empty({C}, opts);
-  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_sums", [&] {
-    constexpr int V = 16 / sizeof(scalar_t);
-    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
-    const int cpacks = C / V;
-    const int nrl = std::max(256 / cpacks, 1);
-    const int64_t rpb = pick_rows_per_block(rows, nrl);
-    const int grid = (int)ceil_div(rows, rpb);
-    auto part = at::empty({grid, 2 * C}, opts);
-    hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(grid), dim3(256), 0,
-                       cur_stream(), (const scalar_t*)x.data_ptr(),
-                       part.data_ptr<float>(), nullptr, rows, C, rpb);
-    auto both = at::empty({2 * C}, opts);
-    hipLaunchKernelGGL(reduce_partials_kernel, dim3(ceil_div(2 * C, 64)),
-                       dim3(256), 0, cur_stream(), part.data_ptr<float>(),
-                       both.data_ptr<float>(), 2 * C, grid);
-    s = both.narrow(0, 0, C);
-    ss = both.narrow(0, C, C);
-  });
-  return {s, ss};
-}
-
-// stage-1 only: per-block partials [grid, 2C] (shared by bn_stats)
-static std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
+// stage-1 only helper: per-block partials [grid, 2C]
+std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
   const int C = x.size(1);
   const int64_t rows = x.numel() / C;
   auto opts = x.options().dtype(at::kFloat);
@@ -521,14 +502,27 @@ static std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
     part = at::empty({grid, 2 * C}, opts);
     hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(grid), dim3(256), 0,
                        cur_stream(), (const scalar_t*)x.data_ptr(),
-                       part.data_ptr<float>(), nullptr, rows, C, rpb);
+                       part.data_ptr<float>(), rows, C, rpb);
     grid_out = grid;
   });
   return {part, grid_out};
 }
 
-// One-shot stats: sums -> (mean, rstd, scale, shift) + running update.
-// Returns {mean, rstd, scale, shift}.
+}  // namespace
+
+std::vector<at::Tensor> bn_sums(at::Tensor x) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int C = x.size(1);
+  auto pg = bn_partials(x);
+  auto both = at::empty({2 * C}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3((int)ceil_div(2 * C, 64)),
+                     dim3(256), 0, cur_stream(), pg.first.data_ptr<float>(),
+                     both.data_ptr<float>(), 2 * C, pg.second);
+  return {both.narrow(0, 0, C), both.narrow(0, C, C)};
+}
+
+// One-shot stats: partials -> (mean, rstd, scale, shift) + running update.
 std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
                                  at::Tensor beta,
                                  c10::optional<at::Tensor> rm_opt,
@@ -542,14 +536,13 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
   auto rstd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
-  const int grid = (int)ceil_div(C, 64);
   if (training) {
     auto pg = bn_partials(x);
     const int64_t rows = x.numel() / C;
     const float inv_cnt = 1.f / (float)rows;
     const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3(grid), dim3(256), 0,
-                       cur_stream(), pg.first.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((int)ceil_div(C, 64)),
+                       dim3(256), 0, cur_stream(), pg.first.data_ptr<float>(),
                        pg.second, gamma.data_ptr<float>(),
                        beta.data_ptr<float>(),
                        running_mean.defined() ? running_mean.data_ptr<float>()
@@ -561,8 +554,9 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
                        inv_cnt, unbias, (float)momentum, (float)eps,
                        running_mean.defined() ? 1 : 0);
   } else {
-    hipLaunchKernelGGL(bn_eval_prep_kernel, dim3(grid), dim3(256), 0,
-                       cur_stream(), running_mean.data_ptr<float>(),
+    hipLaunchKernelGGL(bn_eval_prep_kernel, dim3((int)ceil_div(C, 256)),
+                       dim3(256), 0, cur_stream(),
+                       running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(), gamma.data_ptr<float>(),
                        beta.data_ptr<float>(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), scale.data_ptr<float>(),
@@ -601,20 +595,26 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
   return y;
 }
 
+// Full backward. scale/shift are the forward's fused coefficients (needed to
+// recompute z for silu/sigmoid activation backward).
 std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
-                               at::Tensor beta, int64_t act, bool training,
-                               bool need_gres) {
+                               at::Tensor scale, at::Tensor shift,
+                               int64_t act, bool training, bool need_gres) {
   CHECK_GPU(gy);
   check_nhwc(gy, "gy");
   const int C = x.size(1);
   const int64_t rows = x.numel() / C;
   auto fopts = x.options().dtype(at::kFloat);
-  at::Tensor sum_g, sum_gxh;
   const bool has_res = res.has_value();
   auto gx = at::empty_like(x);
   auto gres = need_gres ? at::empty_like(x) : at::Tensor();
+  auto gw = at::empty({C}, fopts);
+  auto gb = at::empty({C}, fopts);
+  auto P1 = at::empty({C}, fopts);
+  auto P2 = at::empty({C}, fopts);
+  auto P3 = at::empty({C}, fopts);
 
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
@@ -632,52 +632,34 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                          dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
                          (const scalar_t*)x.data_ptr(),
                          (const scalar_t*)y.data_ptr(), resp,
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                         part.data_ptr<float>(), nullptr,
-                         rows, C, rpb, (int)act);
+                         scale.data_ptr<float>(), shift.data_ptr<float>(),
+                         part.data_ptr<float>(), rows, C, rpb, (int)act);
     else
       hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
                          dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
                          (const scalar_t*)x.data_ptr(),
                          (const scalar_t*)y.data_ptr(), resp,
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                         part.data_ptr<float>(), nullptr,
-                         rows, C, rpb, (int)act);
-    auto both = at::empty({2 * C}, fopts);
-    hipLaunchKernelGGL(reduce_partials_kernel, dim3(ceil_div(2 * C, 64)),
-                       dim3(256), 0, stream, part.data_ptr<float>(),
-                       both.data_ptr<float>(), 2 * C, rgrid);
-    sum_g = both.narrow(0, 0, C);
-    sum_gxh = both.narrow(0, C, C);
+                         scale.data_ptr<float>(), shift.data_ptr<float>(),
+                         part.data_ptr<float>(), rows, C, rpb, (int)act);
+    const float inv_cnt = 1.f / (float)rows;
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 64)),
+                       dim3(256), 0, stream, part.data_ptr<float>(), rgrid,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), gw.data_ptr<float>(),
+                       gb.data_ptr<float>(), P1.data_ptr<float>(),
+                       P2.data_ptr<float>(), P3.data_ptr<float>(), C, inv_cnt,
+                       training ? 1 : 0);
     int64_t npacks = x.numel() / V;
     int grid = grid_1d(npacks, 256);
-    float inv_cnt = 1.f / (float)rows;
     scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
-    if (training) {
-      if (has_res)
-        launch_dx<scalar_t, V, true, true>(gy, x, y, resp, mean, rstd, gamma,
-                                           beta, sum_g, sum_gxh, gx, gresp,
-                                           npacks, cpacks, (int)act, inv_cnt,
-                                           grid, stream);
-      else
-        launch_dx<scalar_t, V, false, true>(gy, x, y, resp, mean, rstd, gamma,
-                                            beta, sum_g, sum_gxh, gx, gresp,
-                                            npacks, cpacks, (int)act, inv_cnt,
-                                            grid, stream);
-    } else {
-      if (has_res)
-        launch_dx<scalar_t, V, true, false>(gy, x, y, resp, mean, rstd, gamma,
-                                            beta, sum_g, sum_gxh, gx, gresp,
-                                            npacks, cpacks, (int)act, inv_cnt,
-                                            grid, stream);
-      else
-        launch_dx<scalar_t, V, false, false>(gy, x, y, resp, mean, rstd, gamma,
-                                             beta, sum_g, sum_gxh, gx, gresp,
-                                             npacks, cpacks, (int)act, inv_cnt,
-                                             grid, stream);
-    }
+    if (has_res)
+      launch_dx<scalar_t, V, true>(gy, x, y, resp, scale, shift, P1, P2, P3,
+                                   gx, gresp, npacks, cpacks, (int)act, grid,
+                                   stream);
+    else
+      launch_dx<scalar_t, V, false>(gy, x, y, resp, scale, shift, P1, P2, P3,
+                                    gx, gresp, npacks, cpacks, (int)act, grid,
+                                    stream);
   });
-  return {gx, sum_gxh, sum_g, gres};
+  return {gx, gw, gb, gres};
 }

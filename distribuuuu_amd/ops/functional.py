@@ -161,7 +161,7 @@ class _HIPBatchNormAct(torch.autograd.Function):
                 running_var.copy_(rv)
         res = _cl(residual) if residual is not None else None
         y = e.bn_apply_act(x, scale, shift, act_id, res)
-        ctx.save_for_backward(x, gamma, beta, mean, rstd, y,
+        ctx.save_for_backward(x, gamma, scale, shift, mean, rstd, y,
                               res if res is not None else x.new_empty(0))
         ctx.act_id = act_id
         ctx.training = training
@@ -171,12 +171,12 @@ class _HIPBatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy):
-        x, gamma, beta, mean, rstd, y, res = ctx.saved_tensors
+        x, gamma, scale, shift, mean, rstd, y, res = ctx.saved_tensors
         e = ext()
         gy = _cl(gy)
         gx, gw, gb, gres = e.bn_bwd(
-            gy, x, y, res if ctx.has_res else None, mean, rstd, gamma, beta,
-            ctx.act_id, ctx.training, ctx.has_res)
+            gy, x, y, res if ctx.has_res else None, mean, rstd, gamma,
+            scale, shift, ctx.act_id, ctx.training, ctx.has_res)
         return (gx, gw.to(ctx.w_dtype), gb.to(ctx.w_dtype), None, None, None,
                 None, None, None, gres if ctx.has_res else None)
 
