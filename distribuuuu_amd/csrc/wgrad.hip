@@ -1,12 +1,18 @@
 // Weight-gradient implicit GEMM: gw[k][r][s][c] =
 // sum_m gy[m][k] * patch(x)[m][r][s][c], reduction over all output pixels m.
 //
-// Both operands are m-major in memory (gy NHWC rows, x NHWC gather), while the
-// MFMA wants the reduction (m) per-lane-contiguous — tiles are staged into LDS
-// TRANSPOSED ([k][m], [rsc][m]) by element scatter-writes, fragments then read
-// contiguously. Split-m: each block owns a pixel chunk and atomically
-// accumulates its fp32 partial tile into gw_accum[Kt][R*S*Cg]; a final cast
-// kernel produces the bf16 weight grad in channels_last layout.
+// Both operands are m-major in memory while the MFMA wants the reduction (m)
+// per-lane-contiguous. v2: tiles are staged in their NATURAL [m][col] order
+// with conflict-free 16-byte LDS writes into [32 m][16 col] subtiles, and the
+// MFMA fragments are read TRANSPOSED with ds_read_b64_tr_b16 (hardware
+// 4x4-word transpose within 16-lane groups; semantics verified by
+// tools/probes/tr_probe.hip on gfx950: out-lane o = q*4+e of a 16-lane group
+// receives element e of the 8-byte words addressed by source lanes
+// q, q+4, q+8, q+12). The v1 element-scatter staging was 8-way bank-conflict
+// bound (SQ_LDS_BANK_CONFLICT = 83% of LDS cycles).
+//
+// Split-m: each block owns a pixel chunk and atomically accumulates its fp32
+// partial tile into gw_accum[Kt][R*S*Cg]; a final cast kernel emits bf16.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4w;
@@ -17,8 +23,11 @@ namespace {
 constexpr int WBM = 64;   // k rows per tile
 constexpr int WBN = 64;   // rsc cols per tile
 constexpr int WBK = 64;   // pixels per k-step
-constexpr int WLDS = WBK + 8;  // m-stride with pad
 constexpr int CHUNK_STEPS = 64;  // k-steps per block chunk (4096 pixels)
+// [32 m][16 col] subtile: 32 rows x 32 B = 1024 B, padded to 1056 B so the
+// 8-lane b128 write groups land on distinct banks (264 dwords % 32 = 8).
+constexpr int SUBT = 1056;
+constexpr int TILE_BYTES = 8 * SUBT;  // 2 m-subtiles x 4 col-subtiles
 
 struct WgradParams {
   const __hip_bfloat16* x;   // [N,H,W,Ct]
@@ -32,14 +41,54 @@ struct WgradParams {
   int ktiles, ntiles, chunks;
 };
 
+// LDS byte offset of element (m, col) in a [64 m][64 col] tile stored as
+// [m/32][col/16] subtiles of [32][16].
+DEV_INLINE int lds_off(int m, int col) {
+  return ((m >> 5) * 4 + (col >> 4)) * SUBT + (m & 31) * 32 + (col & 15) * 2;
+}
+
+// per-lane tr_b16 source address for a fragment whose out-lane l wants
+// column (l&15) of a 16-col subtile and m-rows (l>>4)*8 + jj*4 + (0..3):
+// source lane s=l&15 contributes the word at row (l>>4)*8 + (s>>2) + jj*4,
+// byte (s&3)*8 of the subtile.
+DEV_INLINE int tr_addr(int subtile_base, int lane, int jj) {
+  const int s = lane & 15;
+  const int g = lane >> 4;
+  return subtile_base + (g * 8 + (s >> 2) + jj * 4) * 32 + (s & 3) * 8;
+}
+
+DEV_INLINE bf16x8w tr_read_frag(unsigned lds_base, int subtile_base,
+                                int lane) {
+  union {
+    uint2 h[2];
+    bf16x8w v;
+  } u;
+  const unsigned a0 = lds_base + tr_addr(subtile_base, lane, 0);
+  const unsigned a1 = lds_base + tr_addr(subtile_base, lane, 1);
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(u.h[0]), "=&v"(u.h[1])
+      : "v"(a0), "v"(a1)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);  // rule 18: fence MFMA below the wait
+  return u.v;
+}
+
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   const int g = blockIdx.z;
   const int ktile = blockIdx.x % p.ktiles;
   const int ntile = blockIdx.x / p.ktiles;
   const int chunk = blockIdx.y;
 
-  __shared__ __hip_bfloat16 ldsA[2][WBM][WLDS];  // [k][m]
-  __shared__ __hip_bfloat16 ldsB[2][WBN][WLDS];  // [rsc][m]
+  // ONE shared object (a second one forces vmcnt(0) drains before ds_reads)
+  __shared__ __align__(16) char smem[2 * 2 * TILE_BYTES];
+  auto ldsA = [&](int buf) -> char* { return smem + buf * 2 * TILE_BYTES; };
+  auto ldsB = [&](int buf) -> char* {
+    return smem + buf * 2 * TILE_BYTES + TILE_BYTES;
+  };
+  const unsigned smem_base = (unsigned)(unsigned long long)&smem[0];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -47,12 +96,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   const int wm = wid >> 1, wn = wid & 1;
   const int il = lane & 15, kq = lane >> 4;
 
-  // staging: thread t loads 8 contiguous k (or c) at pixel m_local:
+  // staging: thread t loads 8 contiguous cols at pixel m_local:
   //   m_local = t/8 (+32), col8 = (t%8)*8
   const int sml = tid >> 3;        // 0..31
   const int scol8 = (tid & 7) << 3;
   const int SCg = p.S * p.Cg;
-  // B-side decomposition of this thread's rsc range
   const int rsc = ntile * WBN + scol8;
   const bool rsc_ok = rsc < p.RSC;
   const int br = rsc_ok ? rsc / SCg : 0;
@@ -61,13 +109,14 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   const int bc = brem - bs * p.Cg;
   const int kcol = ktile * WBM + scol8;
   const bool k_ok = kcol < p.Kg;
+  const int wroff = lds_off(sml, scol8);  // same for both halves' subtile rows
+  const int wroff2 = lds_off(sml + 32, scol8);
 
   const int m0 = chunk * (CHUNK_STEPS * WBK);
   const int HoWo = p.Ho * p.Wo;
-  const int ksteps =
-      min(CHUNK_STEPS, (p.M - m0 + WBK - 1) / WBK);
+  const int ksteps = min(CHUNK_STEPS, (int)((p.M - m0 + WBK - 1) / WBK));
 
-  __hip_bfloat16 regA[2][8], regB[2][8];
+  uint4 regA[2], regB[2];
   auto stage_load = [&](int ks) {
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
@@ -77,40 +126,30 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
       const int n = mm / HoWo;
       const int rem = mm - n * HoWo;
       const int ho = rem / p.Wo, wo = rem - (rem / p.Wo) * p.Wo;
-      // gy[m][kcol..+8]
       if (m_ok && k_ok) {
-        *reinterpret_cast<uint4*>(&regA[half][0]) =
-            *reinterpret_cast<const uint4*>(
-                p.gy + ((int64_t)mm * p.Kt) + g * p.Kg + kcol);
+        regA[half] = *reinterpret_cast<const uint4*>(
+            p.gy + ((int64_t)mm * p.Kt) + g * p.Kg + kcol);
       } else {
-        *reinterpret_cast<uint4*>(&regA[half][0]) = uint4{0, 0, 0, 0};
+        regA[half] = uint4{0, 0, 0, 0};
       }
-      // x[n, ho*sh-ph+r*dh, wo*sw-pw+s*dw, gCg + c..+8]
       const int h = ho * p.sh - p.ph + br * p.dh;
       const int w_ = wo * p.sw - p.pw + bs * p.dw;
       if (m_ok && rsc_ok && h >= 0 && h < p.H && w_ >= 0 && w_ < p.W) {
-        *reinterpret_cast<uint4*>(&regB[half][0]) =
-            *reinterpret_cast<const uint4*>(
-                p.x + (((int64_t)n * p.H + h) * p.W + w_) * p.Ct + g * p.Cg +
-                bc);
+        regB[half] = *reinterpret_cast<const uint4*>(
+            p.x + (((int64_t)n * p.H + h) * p.W + w_) * p.Ct + g * p.Cg + bc);
       } else {
-        *reinterpret_cast<uint4*>(&regB[half][0]) = uint4{0, 0, 0, 0};
+        regB[half] = uint4{0, 0, 0, 0};
       }
     }
   };
   auto stage_write = [&](int buf) {
-#pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const int m = sml + half * 32;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        ldsA[buf][scol8 + i][m] = regA[half][i];
-        ldsB[buf][scol8 + i][m] = regB[half][i];
-      }
-    }
+    *reinterpret_cast<uint4*>(ldsA(buf) + wroff) = regA[0];
+    *reinterpret_cast<uint4*>(ldsA(buf) + wroff2) = regA[1];
+    *reinterpret_cast<uint4*>(ldsB(buf) + wroff) = regB[0];
+    *reinterpret_cast<uint4*>(ldsB(buf) + wroff2) = regB[1];
   };
 
-  f32x4w accv[2][2] = {};  // 2x2 fragments of 16x16 per wave
+  f32x4w accv[2][2] = {};
 
   stage_load(0);
   stage_write(0);
@@ -119,17 +158,19 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
 
   int cur = 0;
   for (int ks = 0; ks < ksteps; ++ks) {
+    const unsigned abase = smem_base + (unsigned)(cur * 2 * TILE_BYTES);
+    const unsigned bbase = abase + TILE_BYTES;
 #pragma unroll
     for (int mc = 0; mc < 2; ++mc) {  // two 32-deep m sub-steps
       bf16x8w afrag[2], bfrag[2];
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
-        afrag[mi] = *reinterpret_cast<const bf16x8w*>(
-            &ldsA[cur][wm * 32 + mi * 16 + il][mc * 32 + kq * 8]);
+        afrag[mi] =
+            tr_read_frag(abase, (mc * 4 + (wm * 2 + mi)) * SUBT, lane);
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        bfrag[ni] = *reinterpret_cast<const bf16x8w*>(
-            &ldsB[cur][wn * 32 + ni * 16 + il][mc * 32 + kq * 8]);
+        bfrag[ni] =
+            tr_read_frag(bbase, (mc * 4 + (wn * 2 + ni)) * SUBT, lane);
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -145,6 +186,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
     }
     cur ^= 1;
   }
+
   // epilogue: fp32 atomic accumulate (D: col=lane&15, row=(lane>>4)*4+rr)
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {

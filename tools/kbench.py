@@ -73,7 +73,10 @@ def bench_bn():
         rstd = torch.ones(c, device=DEV)
         g = torch.ones(c, device=DEV)
         b = torch.zeros(c, device=DEV)
-        t = timeit(lambda: e.bn_bwd(gy, x, y, res, mean, rstd, g, b, 1, True, True))
+        sc = torch.ones(c, device=DEV)
+        sh = torch.zeros(c, device=DEV)
+        t = timeit(lambda: e.bn_bwd(gy, x, y, res, mean, rstd, g, sc, sh, 1,
+                                    True, True))
         print(f"bn_bwd(all)  {str(shp):22s} {t*1e6:8.1f} us  {8*nbytes/t/1e9:7.0f} GB/s")
 
 
