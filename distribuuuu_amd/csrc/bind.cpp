@@ -56,6 +56,9 @@ at::Tensor dwconv_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw);
+// attention.hip
+at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
+                    at::Tensor rh, int64_t H, int64_t W);
 // sgd.hip
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
@@ -97,4 +100,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_fwd", &dwconv_fwd);
   m.def("dwconv_dgrad", &dwconv_dgrad);
   m.def("dwconv_wgrad", &dwconv_wgrad);
+  m.def("mhsa_fwd", &mhsa_fwd);
 }

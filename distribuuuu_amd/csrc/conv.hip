@@ -161,26 +161,49 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
     cur ^= 1;
   }
 
-  // ---- epilogue: scalar bf16 stores (C layout: col=lane&15, row=kg*4+reg) --
+  // ---- epilogue: LDS-restage then 16-wide bf16 stores ----------------------
+  // Scalar 2-byte stores (64/lane) are store-issue-bound (T21 diagnosis);
+  // instead each wave round-trips its 64x64 tile through a private fp32 LDS
+  // slab 16 rows at a time, then stores 16 consecutive channels per lane.
+  __syncthreads();  // everyone done reading the final A/B tiles
+  float* slab = reinterpret_cast<float*>(&ldsA[0][0][0]) + wid * (16 * 68);
   const int HoWo = p.Ho * p.Wo;
+  const int er = lane >> 2;          // 0..15 row within the 16-row stripe
+  const int ec = (lane & 3) << 4;    // 0,16,32,48 col
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
+    // scatter this mi-stripe's fragments to the slab (conflict-free b32)
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int m = tile_m * BM + wm * 64 + mi * 16 + kg * 4 + rr;
-      if (m >= p.M) continue;
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr)
+        slab[(kg * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+    __builtin_amdgcn_wave_barrier();
+    const int m = tile_m * BM + wm * 64 + mi * 16 + er;
+    if (m < p.M) {
       const int n = m / HoWo;
-      const int rem = m % HoWo;
+      const int rem = m - n * HoWo;
       const int64_t obase =
           (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh) * p.WoA +
            (rem % p.Wo) * p.osw) * p.Kt + g * p.Kg;
+      const int k0 = tile_n * BN + wn * 64 + ec;
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } u;
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const int k = tile_n * BN + wn * 64 + ni * 16 + il;
-        if (k < p.Kg)
-          p.y[obase + k] = from_f32<__hip_bfloat16>(acc[mi][ni][rr]);
+      for (int j = 0; j < 16; ++j)
+        u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (k0 + 16 <= p.Kg) {
+        *reinterpret_cast<uint4*>(&p.y[obase + k0]) = u.q[0];
+        *reinterpret_cast<uint4*>(&p.y[obase + k0 + 8]) = u.q[1];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          if (k0 + j < p.Kg) p.y[obase + k0 + j] = u.b[j];
       }
     }
+    __builtin_amdgcn_wave_barrier();
   }
 }
 

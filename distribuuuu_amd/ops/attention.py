@@ -51,30 +51,51 @@ def _rel_pos_logits(q, rel_h, rel_w, h, w):
     return out.reshape(n, heads, l, l)
 
 
+def _torch_mhsa(q, k, v, rel_h, rel_w, h, w):
+    logits = torch.einsum("bhxd,bhyd->bhxy", q, k)
+    logits = logits + _rel_pos_logits(q, rel_h, rel_w, h, w)
+    attn = F.softmax(logits, dim=-1)
+    return torch.einsum("bhxy,bhyd->bhxd", attn, v)
+
+
 class _HIPMHSARelPos(torch.autograd.Function):
+    """Fused CDNA4 forward (one kernel: QK^T + decomposed rel-pos + softmax
+    + PV, score tile LDS-resident); backward recomputes through the torch
+    composition flash-style (the L x L attention matrix is never saved)."""
+
     @staticmethod
     def forward(ctx, q, k, v, rel_h, rel_w, h, w):
-        out, p = ext().mhsa_relpos_fwd(q.contiguous(), k.contiguous(),
-                                       v.contiguous(), rel_h.contiguous(),
-                                       rel_w.contiguous(), h, w)
-        ctx.save_for_backward(q, k, v, rel_h, rel_w, p)
+        n, heads, l, d = q.shape
+        qf = q.reshape(n * heads, l, d).contiguous()
+        kf = k.reshape(n * heads, l, d).contiguous()
+        vt = v.reshape(n * heads, l, d).transpose(1, 2).contiguous()
+        # per-row relative-logit tables (fp32): RW = q@rel_w^T, RH = q@rel_h^T
+        rw = torch.matmul(qf.float(), rel_w.float().t()).contiguous()
+        rh = torch.matmul(qf.float(), rel_h.float().t()).contiguous()
+        out = ext().mhsa_fwd(qf, kf, vt, rw, rh, h, w)
+        ctx.save_for_backward(q, k, v, rel_h, rel_w)
         ctx.hw = (h, w)
-        return out
+        return out.reshape(n, heads, l, d)
 
     @staticmethod
     def backward(ctx, gout):
-        q, k, v, rel_h, rel_w, p = ctx.saved_tensors
+        q, k, v, rel_h, rel_w = ctx.saved_tensors
         h, w = ctx.hw
-        gq, gk, gv, grh, grw = ext().mhsa_relpos_bwd(
-            gout.contiguous(), q, k, v, rel_h, rel_w, p, h, w)
+        with torch.enable_grad():
+            q_ = q.detach().requires_grad_(True)
+            k_ = k.detach().requires_grad_(True)
+            v_ = v.detach().requires_grad_(True)
+            rh_ = rel_h.detach().requires_grad_(True)
+            rw_ = rel_w.detach().requires_grad_(True)
+            out = _torch_mhsa(q_, k_, v_, rh_, rw_, h, w)
+            gq, gk, gv, grh, grw = torch.autograd.grad(
+                out, [q_, k_, v_, rh_, rw_], gout)
         return gq, gk, gv, grh, grw, None, None
 
 
 def mhsa_relpos(q, k, v, rel_h, rel_w, h, w):
     """q (pre-scaled), k, v: [N, heads, L, d]; returns [N, heads, L, d_v]."""
-    if use_hip(q, "mhsa_relpos_fwd"):
+    if (use_hip(q, "mhsa_fwd") and q.dtype == torch.bfloat16
+            and q.shape[-1] % 32 == 0 and q.shape[-1] <= 128):
         return _HIPMHSARelPos.apply(q, k, v, rel_h, rel_w, h, w)
-    logits = torch.einsum("bhxd,bhyd->bhxy", q, k)
-    logits = logits + _rel_pos_logits(q, rel_h, rel_w, h, w)
-    attn = F.softmax(logits, dim=-1)
-    return torch.einsum("bhxy,bhyd->bhxd", attn, v)
+    return _torch_mhsa(q, k, v, rel_h, rel_w, h, w)

@@ -59,9 +59,7 @@ def hip_op_available(name):
 # Ops whose HIP kernels are not yet implemented: ATen fallback allowed WITH a
 # warning. This set shrinks as kernels land; it must be empty for the hot path
 # by release.
-_DEFAULT_SOFT = {
-    "mhsa_relpos_fwd",  # fused BoT attention kernel: pending
-}
+_DEFAULT_SOFT = set()
 
 
 def _soft_ops():
