@@ -117,7 +117,7 @@ def main():
 
     if rank == 0:
         print(json.dumps({
-            "metric": "images/sec ResNet-50 train (whole node)",
+            "metric": f"images/sec {args.arch} train (whole node)",
             "value": round(images_per_sec, 2),
             "unit": "images/sec",
             "n_gpus": world_size,

@@ -59,6 +59,10 @@ at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
 // attention.hip
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
                     at::Tensor rh, int64_t H, int64_t W);
+// augment.hip
+at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
+                              std::vector<double> mean,
+                              std::vector<double> std, at::ScalarType dtype);
 // sgd.hip
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
@@ -101,4 +105,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_dgrad", &dwconv_dgrad);
   m.def("dwconv_wgrad", &dwconv_wgrad);
   m.def("mhsa_fwd", &mhsa_fwd);
+  m.def("aug_crop_flip_norm", &aug_crop_flip_norm);
 }

@@ -121,13 +121,137 @@ class ImageFolderDataset(Dataset):
         return torch.from_numpy(arr.transpose(2, 0, 1).copy()).float(), label
 
 
+class RawImageDataset(ImageFolderDataset):
+    """Decode-only dataset: returns (uint8 HWC ndarray, label). Crop/flip/
+    normalize run on the GPU (K17) — CPU workers only decode."""
+
+    def __getitem__(self, idx):
+        from PIL import Image
+
+        path, label = self.samples[idx]
+        with Image.open(path) as img:
+            arr = np.asarray(img.convert("RGB"), dtype=np.uint8)
+        return arr, label
+
+
+def _draw_crop(h, w, train, im_size, resize):
+    """RandomResizedCrop parameters (train) or Resize+CenterCrop box (val),
+    matching the reference's torchvision transforms."""
+    if train:
+        area = h * w
+        for _ in range(10):
+            target_area = random.uniform(0.08, 1.0) * area
+            ar = np.exp(random.uniform(np.log(3 / 4), np.log(4 / 3)))
+            cw = int(round(np.sqrt(target_area * ar)))
+            ch = int(round(np.sqrt(target_area / ar)))
+            if cw <= w and ch <= h:
+                return (random.randint(0, h - ch), random.randint(0, w - cw),
+                        ch, cw, int(random.random() < 0.5))
+        s = min(h, w)
+        return ((h - s) // 2, (w - s) // 2, s, s, 0)
+    # val: shorter side -> resize, center crop im_size: equivalent crop box
+    s = min(h, w)
+    box = int(round(s * im_size / resize))
+    return ((h - box) // 2, (w - box) // 2, box, box, 0)
+
+
+class RawBatchCollate:
+    """Concatenate decoded uint8 images + per-image crop metadata for the GPU
+    augmentation kernel."""
+
+    def __init__(self, train, im_size, resize=256):
+        self.train = train
+        self.im_size = im_size
+        self.resize = resize
+
+    def __call__(self, batch):
+        metas = []
+        offset = 0
+        bufs = []
+        labels = []
+        for arr, label in batch:
+            h, w = arr.shape[0], arr.shape[1]
+            cy, cx, ch, cw, flip = _draw_crop(h, w, self.train, self.im_size,
+                                              self.resize)
+            metas.append([offset, h, w, cy, cx, ch, cw, flip])
+            bufs.append(arr.reshape(-1))
+            offset += arr.size
+            labels.append(label)
+        raw = torch.from_numpy(np.concatenate(bufs))
+        meta = torch.tensor(metas, dtype=torch.int32)
+        return raw, meta, torch.tensor(labels, dtype=torch.long)
+
+
+class GPUAugLoader:
+    """Wraps a raw-byte DataLoader: pinned hipMemcpyAsync H2D on a side
+    stream + the fused crop/flip/normalize kernel, one batch prefetched ahead
+    of compute (SURVEY.md K17 / BASELINE.json data pipeline)."""
+
+    def __init__(self, loader, im_size, dtype, device="cuda"):
+        self.loader = loader
+        self.im_size = im_size
+        self.dtype = dtype
+        self.device = device
+        self.stream = torch.cuda.Stream(device)
+        self.sampler = loader.sampler
+
+    def __len__(self):
+        return len(self.loader)
+
+    def _issue(self, cpu_batch):
+        from .ops.dispatch import require_ext
+
+        raw, meta, labels = cpu_batch
+        with torch.cuda.stream(self.stream):
+            raw_d = raw.to(self.device, non_blocking=True)
+            meta_d = meta.to(self.device, non_blocking=True)
+            labels_d = labels.to(self.device, non_blocking=True)
+            x = require_ext().aug_crop_flip_norm(
+                raw_d, meta_d, self.im_size, list(IMAGENET_MEAN),
+                list(IMAGENET_STD), self.dtype)
+        ev = torch.cuda.Event()
+        ev.record(self.stream)
+        return x, labels_d, ev
+
+    def __iter__(self):
+        it = iter(self.loader)
+        pending = None
+        for cpu_batch in it:
+            issued = self._issue(cpu_batch)
+            if pending is not None:
+                x, y, ev = pending
+                torch.cuda.current_stream().wait_event(ev)
+                yield x, y
+            pending = issued
+        if pending is not None:
+            x, y, ev = pending
+            torch.cuda.current_stream().wait_event(ev)
+            yield x, y
+
+
 def construct_train_loader():
     """Per-rank train loader (reference utils.py:121-152): per-rank batch size,
-    DistributedSampler(shuffle=True), drop_last=True, pinned memory."""
+    DistributedSampler(shuffle=True), drop_last=True, pinned memory. On a GPU
+    with the HIP extension, augmentation runs on-device (GPUAugLoader)."""
     if cfg.MODEL.DUMMY_INPUT:
         ds = DummyDataset(size=[3, cfg.TRAIN.IM_SIZE, cfg.TRAIN.IM_SIZE])
     else:
         root = os.path.join(cfg.TRAIN.DATASET, cfg.TRAIN.SPLIT)
+        if _gpu_aug_available():
+            ds = RawImageDataset(root, train=True, im_size=cfg.TRAIN.IM_SIZE)
+            loader = DataLoader(
+                ds,
+                batch_size=cfg.TRAIN.BATCH_SIZE,
+                shuffle=not _dist(),
+                sampler=DistributedSampler(ds, shuffle=True) if _dist() else None,
+                num_workers=cfg.TRAIN.WORKERS,
+                pin_memory=cfg.TRAIN.PIN_MEMORY,
+                drop_last=True,
+                collate_fn=RawBatchCollate(True, cfg.TRAIN.IM_SIZE),
+            )
+            dtype = (torch.bfloat16 if cfg.TRAIN.DTYPE == "bfloat16"
+                     else torch.float32)
+            return GPUAugLoader(loader, cfg.TRAIN.IM_SIZE, dtype)
         ds = ImageFolderDataset(root, train=True, im_size=cfg.TRAIN.IM_SIZE)
     sampler = DistributedSampler(ds, shuffle=True) if _dist() else None
     return DataLoader(
@@ -139,6 +263,14 @@ def construct_train_loader():
         pin_memory=cfg.TRAIN.PIN_MEMORY,
         drop_last=True,
     )
+
+
+def _gpu_aug_available():
+    if not torch.cuda.is_available():
+        return False
+    from .ops.dispatch import hip_op_available
+
+    return hip_op_available("aug_crop_flip_norm")
 
 
 def construct_val_loader():
