@@ -26,6 +26,9 @@ def parse_args():
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--syncbn", type=int, default=-1,
                    help="-1: auto (on when world>1), 0/1: force")
+    p.add_argument("--graph", type=int, default=-1,
+                   help="capture the step in a hipGraph: -1 auto "
+                        "(on for 1 GPU), 0/1 force")
     return p.parse_args()
 
 
@@ -91,13 +94,37 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph capture of the whole train step (single-GPU): the inner loop
+    # replays one graph instead of ~900 eager launches.
+    use_graph = (world_size == 1 and has_gpu) if args.graph == -1 else bool(
+        args.graph)
+    graph = None
+    if use_graph:
+        try:
+            torch.cuda.synchronize()
+            side = torch.cuda.Stream()
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    step()
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                step()
+        except Exception as exc:  # pragma: no cover - fall back to eager
+            print(f"[bench] graph capture failed ({exc}); eager mode",
+                  flush=True)
+            graph = None
+
     if world_size > 1:
         dist.barrier()
     if has_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        if graph is not None:
+            graph.replay()
+        else:
+            step()
     if has_gpu:
         torch.cuda.synchronize()
     if world_size > 1:
@@ -136,6 +163,7 @@ def main():
                 "im_size": 224,
                 "parallelism": f"dp{world_size}",
                 "syncbn": use_syncbn,
+                "hipgraph": graph is not None,
             },
         }))
 

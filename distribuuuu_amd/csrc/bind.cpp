@@ -64,10 +64,8 @@ at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
                               std::vector<double> mean,
                               std::vector<double> std, at::ScalarType dtype);
 // sgd.hip
-void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
-              std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
-              double lr, double momentum, double dampening,
-              double weight_decay, bool nesterov);
+void sgd_step(at::Tensor table, double lr, double momentum,
+              double dampening, double weight_decay, bool nesterov);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &relu_fwd);

@@ -1,10 +1,11 @@
 """Fused SGD-momentum optimizer.
 
-One multi-tensor HIP kernel per step consumes a device-resident pointer table
+One multi-tensor HIP kernel per step consumes a device-resident chunk table
 covering every parameter (the reference launches one ATen kernel per tensor —
-161 launches for ResNet-50). bf16 parameters carry fp32 master weights and fp32
-momentum so the update math never loses precision (the bf16-end-to-end accuracy
-requirement of SURVEY.md §7 hard-part 4).
+161 launches for ResNet-50). The table is built once and cached across steps
+(no per-step host work; the step is hipGraph-capturable). bf16 parameters
+carry fp32 master weights and fp32 momentum so the update math never loses
+precision (SURVEY.md §7 hard-part 4).
 
 CPU / no-extension path falls back to torch.optim.SGD semantics exactly
 (reference utils.py:187-196: momentum + nesterov + weight decay + dampening).
@@ -15,9 +16,40 @@ from torch.optim import SGD
 
 from .dispatch import hip_op_available, ext
 
+_CHUNK = 16384
+
 
 class HIPSGD(SGD):
     """torch.optim.SGD drop-in whose GPU step is one fused multi-tensor kernel."""
+
+    def _build_table(self, group):
+        recs = []
+        key = []
+        for p in group["params"]:
+            if p.grad is None:
+                continue
+            state = self.state[p]
+            if "momentum_buffer" not in state:
+                state["momentum_buffer"] = torch.zeros_like(
+                    p, dtype=torch.float32)
+                if p.dtype in (torch.bfloat16, torch.float16):
+                    state["master"] = p.detach().float().clone()
+            master = state.get("master", p)
+            flags = (1 if p.dtype == torch.bfloat16 else 0) | (
+                2 if p.grad.dtype == torch.bfloat16 else 0)
+            key.append((p.data_ptr(), p.grad.data_ptr()))
+            n = p.numel()
+            off = 0
+            while off < n:
+                cnt = min(_CHUNK, n - off)
+                recs.append([p.grad.data_ptr(), p.data_ptr(),
+                             state["momentum_buffer"].data_ptr(),
+                             master.data_ptr(), off,
+                             cnt | (flags << 32)])
+                off += _CHUNK
+        device = group["params"][0].device
+        table = torch.tensor(recs, dtype=torch.int64).to(device)
+        return table, tuple(key)
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -30,27 +62,26 @@ class HIPSGD(SGD):
             return self._fallback_step(loss)
 
         for group in self.param_groups:
-            params, grads, moms, masters = [], [], [], []
-            for p in group["params"]:
-                if p.grad is None:
-                    continue
-                state = self.state[p]
-                if "momentum_buffer" not in state:
-                    state["momentum_buffer"] = torch.zeros_like(
-                        p, dtype=torch.float32)
-                    if p.dtype in (torch.bfloat16, torch.float16):
-                        state["master"] = p.detach().float().clone()
-                params.append(p)
-                grads.append(p.grad)
-                moms.append(state["momentum_buffer"])
-                masters.append(state.get("master", p))  # fp32 view or self
-            if params:
-                ext().sgd_step(
-                    params, grads, moms, masters,
-                    group["lr"], group["momentum"], group["dampening"],
-                    group["weight_decay"], group["nesterov"],
-                )
+            cache = group.get("_hip_table")
+            key = tuple((p.data_ptr(),
+                         p.grad.data_ptr() if p.grad is not None else 0)
+                        for p in group["params"])
+            if cache is None or cache[1] != key:
+                table, _ = self._build_table(group)
+                group["_hip_table"] = (table, key)
+            else:
+                table = cache[0]
+            if table.numel():
+                ext().sgd_step(table, group["lr"], group["momentum"],
+                               group["dampening"], group["weight_decay"],
+                               group["nesterov"])
         return loss
+
+    def state_dict(self):
+        # drop the pointer cache from serialized state
+        for group in self.param_groups:
+            group.pop("_hip_table", None)
+        return super().state_dict()
 
     def _fallback_step(self, loss):
         """Plain SGD math, with fp32 master handling for bf16 params."""

@@ -187,20 +187,22 @@ def test_topk_acc():
 
 @pytest.mark.parametrize("param_dtype", [torch.float32, torch.bfloat16])
 def test_sgd_step_matches_torch(param_dtype):
-    e = _ext()
+    from distribuuuu_amd.ops.optim import HIPSGD
+
+    _ext()
     torch.manual_seed(0)
     shapes = [(64, 32, 3, 3), (128,), (1000, 512)]
     params = [torch.randn(s, device="cuda", dtype=param_dtype) for s in shapes]
     grads = [torch.randn(s, device="cuda", dtype=param_dtype) for s in shapes]
-    moms = [torch.zeros(s, device="cuda", dtype=torch.float32) for s in shapes]
-    masters = [p.float().clone() if param_dtype == torch.bfloat16 else p
-               for p in params]
+    for p, g in zip(params, grads):
+        p.grad = g
+    lr, mu, wd = 0.1, 0.9, 5e-5
+    opt = HIPSGD(params, lr=lr, momentum=mu, weight_decay=wd, nesterov=True)
     # fp32 reference
     ref_w = [p.float().clone() for p in params]
     ref_m = [torch.zeros_like(w) for w in ref_w]
-    lr, mu, damp, wd = 0.1, 0.9, 0.0, 5e-5
     for _ in range(3):
-        e.sgd_step(params, grads, moms, masters, lr, mu, damp, wd, True)
+        opt.step()
         for w, m, g in zip(ref_w, ref_m, grads):
             gf = g.float() + wd * w
             m.mul_(mu).add_(gf)
