@@ -63,7 +63,10 @@ class DistributedDataParallel(nn.Module):
         size = 0
         for p in reversed(params):
             nbytes = p.numel() * p.element_size()
-            if bucket.params and size + nbytes > self._bucket_cap:
+            # a bucket is one flat tensor: split on size cap AND dtype change
+            # (bf16 models keep BN parameters in fp32)
+            if bucket.params and (size + nbytes > self._bucket_cap
+                                  or p.dtype != bucket.params[0].dtype):
                 self._buckets.append(bucket)
                 bucket = _Bucket()
                 size = 0

@@ -159,3 +159,47 @@ def test_scaled_all_reduce_two_ranks():
         assert p.exitcode == 0
     assert vals[0] == pytest.approx(1.5)
     assert vals[1] == pytest.approx(15.0)
+
+
+class _TwoDtypeNet(nn.Module):
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(3)
+        self.a = nn.Parameter(torch.randn(64, dtype=torch.bfloat16))
+        self.b = nn.Parameter(torch.randn(64, dtype=torch.float32))
+
+    def forward(self, x):
+        return (x * self.b).sum() + (x.to(torch.bfloat16) * self.a).float().sum()
+
+
+def _mixed_dtype_worker(rank, world, port, q):
+    """bf16 + fp32 parameters in one module (the bf16 training layout where
+    BN stays fp32) must split into per-dtype buckets."""
+    _init(rank, world, port)
+    from distribuuuu_amd.parallel import DistributedDataParallel
+
+    ddp = DistributedDataParallel(_TwoDtypeNet(), bucket_cap_mb=1)
+    x = torch.randn(64)
+    loss = ddp(x)
+    ddp.zero_grad()
+    loss.backward()
+    if rank == 0:
+        dtypes = sorted({str(b.buffer.dtype) for b in ddp._buckets})
+        q.put(dtypes)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_ddp_mixed_dtype_buckets():
+    port = 29614
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_mixed_dtype_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+    assert all(p.exitcode == 0 for p in procs)
+    dtypes = q.get()
+    assert "torch.bfloat16" in dtypes and "torch.float32" in dtypes

@@ -35,3 +35,22 @@ def test_imagenet_tutorial_single_proc():
                        env=env, capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "checkpoint round-trip OK" in r.stdout
+
+
+def test_bench_2proc_gloo():
+    """bench.py under torchrun world 2 on CPU (the driver's multi-GPU launch
+    shape) — checks the DDP/SyncBN/metric-aggregation path end to end."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29631", "bench.py", "--gpus", "2", "--steps", "2",
+         "--warmup", "1"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-3000:]
+    import json
+
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["syncbn"] is True
