@@ -5,67 +5,101 @@
 
 namespace {
 
-// ---- max pool fwd: thread per (n,ho,wo,c), coalesced over c ---------------
+// ---- max pool fwd: thread per (n,ho,wo,c-pack of 8), vectorized -----------
 template <typename T>
 __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                    int* __restrict__ idx, int N, int H, int W,
                                    int C, int Ho, int Wo, int K, int S, int P) {
-  const int64_t total = (int64_t)N * Ho * Wo * C;
+  constexpr int V = 16 / sizeof(T);
+  using Pk = Pack<T, V>;
+  const int cpacks = C / V;
+  const int64_t total = (int64_t)N * Ho * Wo * cpacks;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const int c = i % C;
-    int64_t t = i / C;
+    const int cp = i % cpacks;
+    int64_t t = i / cpacks;
     const int wo = t % Wo;
     t /= Wo;
     const int ho = t % Ho;
     const int n = t / Ho;
     const int h0 = ho * S - P, w0 = wo * S - P;
-    float best = -INFINITY;
-    int best_idx = 0;
+    float best[V];
+    __align__(16) int bidx[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      best[j] = -INFINITY;
+      bidx[j] = 0;
+    }
     for (int kh = 0; kh < K; ++kh) {
       const int h = h0 + kh;
       if (h < 0 || h >= H) continue;
       for (int kw = 0; kw < K; ++kw) {
         const int w = w0 + kw;
         if (w < 0 || w >= W) continue;
-        float v = to_f32(x[(((int64_t)n * H + h) * W + w) * C + c]);
-        if (v > best) {
-          best = v;
-          best_idx = h * W + w;
+        Pk p = *reinterpret_cast<const Pk*>(
+            x + (((int64_t)n * H + h) * W + w) * C + cp * V);
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float v = to_f32(p.v[j]);
+          if (v > best[j]) {
+            best[j] = v;
+            bidx[j] = h * W + w;
+          }
         }
       }
     }
-    y[i] = from_f32<T>(best);
-    idx[i] = best_idx;
+    const int64_t out = (((int64_t)n * Ho + ho) * Wo + wo) * C + cp * V;
+    Pk py;
+#pragma unroll
+    for (int j = 0; j < V; ++j) py.v[j] = from_f32<T>(best[j]);
+    *reinterpret_cast<Pk*>(y + out) = py;
+#pragma unroll
+    for (int j = 0; j < V / 4; ++j)
+      *reinterpret_cast<int4*>(idx + out + j * 4) =
+          *reinterpret_cast<int4*>(&bidx[j * 4]);
   }
 }
 
-// ---- max pool bwd: gather per input element (no atomics) ------------------
+// ---- max pool bwd: gather per input c-pack (no atomics) -------------------
 template <typename T>
 __global__ void maxpool_bwd_kernel(const T* __restrict__ gy,
                                    const int* __restrict__ idx,
                                    T* __restrict__ gx, int N, int H, int W,
                                    int C, int Ho, int Wo, int K, int S, int P) {
-  const int64_t total = (int64_t)N * H * W * C;
+  constexpr int V = 16 / sizeof(T);
+  using Pk = Pack<T, V>;
+  const int cpacks = C / V;
+  const int64_t total = (int64_t)N * H * W * cpacks;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const int c = i % C;
-    int64_t t = i / C;
+    const int cp = i % cpacks;
+    int64_t t = i / cpacks;
     const int w = t % W;
     t /= W;
     const int h = t % H;
     const int n = t / H;
     const int flat = h * W + w;
-    float acc = 0.f;
-    // output windows that can cover (h, w)
+    float acc[V] = {};
     const int ho_lo = max(0, (h + P - K + S) / S), ho_hi = min(Ho - 1, (h + P) / S);
     const int wo_lo = max(0, (w + P - K + S) / S), wo_hi = min(Wo - 1, (w + P) / S);
     for (int ho = ho_lo; ho <= ho_hi; ++ho)
       for (int wo = wo_lo; wo <= wo_hi; ++wo) {
-        const int64_t o = (((int64_t)n * Ho + ho) * Wo + wo) * C + c;
-        if (idx[o] == flat) acc += to_f32(gy[o]);
+        const int64_t o = (((int64_t)n * Ho + ho) * Wo + wo) * C + cp * V;
+        __align__(16) int bi[V];
+#pragma unroll
+        for (int j = 0; j < V / 4; ++j)
+          *reinterpret_cast<int4*>(&bi[j * 4]) =
+              *reinterpret_cast<const int4*>(idx + o + j * 4);
+        Pk pg = *reinterpret_cast<const Pk*>(gy + o);
+#pragma unroll
+        for (int j = 0; j < V; ++j)
+          if (bi[j] == flat) acc[j] += to_f32(pg.v[j]);
       }
-    gx[i] = from_f32<T>(acc);
+    Pk out;
+#pragma unroll
+    for (int j = 0; j < V; ++j) out.v[j] = from_f32<T>(acc[j]);
+    *reinterpret_cast<Pk*>(gx + (((int64_t)n * H + h) * W + w) * C + cp * V) =
+        out;
   }
 }
 
@@ -163,8 +197,10 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
                        x.options()
                            .dtype(at::kInt)
                            .memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * Ho * Wo * C;
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "maxpool_fwd", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C % ", V, " != 0");
+    int64_t total = (int64_t)N * Ho * Wo * (C / V);
     hipLaunchKernelGGL((maxpool_fwd_kernel<scalar_t>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
                        (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
@@ -180,8 +216,9 @@ at::Tensor maxpool_bwd(at::Tensor gy, at::Tensor idx, int64_t H, int64_t W,
   const int N = gy.size(0), C = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
   auto gx = at::empty({N, C, H, W},
                       gy.options().memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * H * W * C;
   DISPATCH_FLOAT_AND_BF16(gy.scalar_type(), "maxpool_bwd", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    int64_t total = (int64_t)N * H * W * (C / V);
     hipLaunchKernelGGL((maxpool_bwd_kernel<scalar_t>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
                        (const scalar_t*)gy.data_ptr(), idx.data_ptr<int>(),
