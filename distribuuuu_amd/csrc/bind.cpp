@@ -43,6 +43,8 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw,
                         int64_t dh, int64_t dw, int64_t groups);
 at::Tensor gemm_nt(at::Tensor a, at::Tensor b);
+at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw);
 at::Tensor weight_flip_t(at::Tensor w, int64_t groups);
 at::Tensor dilate_nhwc(at::Tensor x, int64_t sh, int64_t sw);
 at::Tensor pad_channels(at::Tensor x, int64_t Cn);
@@ -96,6 +98,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("gemm_nt", &gemm_nt);
+  m.def("conv2d_fwd_v2", &conv2d_fwd_v2);
   m.def("weight_flip_t", &weight_flip_t);
   m.def("dilate_nhwc", &dilate_nhwc);
   m.def("pad_channels", &pad_channels);

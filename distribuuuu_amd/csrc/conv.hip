@@ -245,11 +245,24 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   return y;
 }
 
+at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw);
+
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                       int64_t ph, int64_t pw, int64_t dh, int64_t dw,
                       int64_t groups) {
   const int N = x.size(0), H = x.size(2), W = x.size(3);
   const int Kt = w.size(0), R = w.size(2), S = w.size(3);
+  // v2 (3-slot glds ring) wins when the reduction is deep enough to fill
+  // its pipeline; v1 handles shallow-K, small-K and grouped shapes.
+  static const bool v2_off = []() {
+    const char* e = getenv("DISTRIBUUUU_CONV_V2");
+    return e && e[0] == '0';
+  }();
+  const int C_ = x.size(1);
+  if (!v2_off && groups == 1 && Kt >= 128 && R * S * C_ >= 576 &&
+      x.scalar_type() == at::kBFloat16)
+    return conv2d_fwd_v2(x, w, sh, sw, ph, pw, dh, dw);
   const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
   auto y = at::empty({N, Kt, Ho, Wo},
@@ -404,6 +417,14 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
                     groups, sh, sw);
     return gx1;
   }
+  const int Kt_ = gy.size(1);
+  if (sh == 1 && sw == 1 && groups == 1 && dh * (R - 1) == 2 * ph &&
+      dw * (S - 1) == 2 * pw) {
+    // same-size conv: the plain fwd path (incl. the v2 ring kernel) applies
+    return conv2d_fwd(gy, wt, 1, 1, dh * (R - 1) - ph, dw * (S - 1) - pw,
+                      dh, dw, 1);
+  }
+  (void)Kt_;
   auto gyd = (sh == 1 && sw == 1) ? gy : dilate_nhwc(gy, sh, sw);
   const int Hd = gyd.size(2), Wd = gyd.size(3);
   const int pph = dh * (R - 1) - ph, ppw = dw * (S - 1) - pw;

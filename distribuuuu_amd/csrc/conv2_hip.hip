@@ -1,0 +1,322 @@
+#include "hip/hip_runtime.h"
+// Implicit-GEMM conv forward, v2 (large-K layers): 256x256x64 tile, 8 waves,
+// global_load_lds 16-byte direct staging with the st_16x32 LDS XOR swizzle
+// applied on the SOURCE address (glds writes lane-linear; platform guide
+// §5.4 rule 21), double-buffered, counted-barrier loop.
+//
+// Predication-free inner loop: the host pads the input image in HBM
+// (pad_image) and zero-pads the weight's reduction span to a multiple of 64
+// (pad_weight_span), so every gather address is in-bounds and span-tail
+// garbage in A multiplies a zero in B. M/K tile tails clamp their source rows
+// (garbage rows are computed but never stored).
+#include "common_hip.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4c;
+typedef __bf16 bf16x8c __attribute__((ext_vector_type(8)));
+
+namespace {
+
+constexpr int BM2 = 256, BN2 = 128, BK2 = 64;
+constexpr int ATILE_B = BM2 * BK2 * 2;          // 32 KB
+constexpr int BTILE_B = BN2 * BK2 * 2;          // 16 KB
+constexpr int BUF_B = ATILE_B + BTILE_B;        // 48 KB per ring slot
+constexpr int NBUF = 3;                         // 144 KB ring
+
+struct Conv2Params {
+  const __hip_bfloat16* x;  // padded [N, Hp, Wp, C] (+slack)
+  const __hip_bfloat16* w;  // span-padded [K, R, SPAN64]
+  __hip_bfloat16* y;        // [N, Ho, Wo, K]
+  int N, Hp, Wp, C, K;
+  int R, SPAN64, Cg, S;
+  int sh, sw, dh, dw;
+  int Ho, Wo;
+  int M, nspan, ksteps;
+  int tiles_m;
+};
+
+// st_16x32 swizzle on a byte offset within a tile (1024-B subtiles)
+DEV_INLINE int swz(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
+
+#define WAITVM(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
+
+__global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
+  int tile_m = blockIdx.x, tile_n = blockIdx.y;
+  {  // XCD-aware bijective remap over m-tiles (T1)
+    const int nwg = p.tiles_m;
+    const int q = nwg / 8, r8 = nwg % 8;
+    const int xcd = tile_m % 8, idx = tile_m / 8;
+    tile_m = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q) + idx;
+  }
+
+  __shared__ __align__(16) char smem[NBUF * BUF_B];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 8 waves: 4 (M) x 2 (N)
+  const int wm = wid >> 1, wn = wid & 1;
+  const int il = lane & 15, kq = lane >> 4;
+
+  // ---- staging source precompute ------------------------------------------
+  // A: 4 slots/thread (32 KB); B: 2 slots/thread (16 KB). Stored slot
+  // s = tid + it*512; logical slot sl = s ^ (((s>>5)&1)<<1);
+  // row = sl>>3, span-col16 = sl&7.
+  int a_n[4], a_hwbase[4], a_sl[4];
+  const int HoWo = p.Ho * p.Wo;
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int s = tid + it * 512;
+    const int sl = s ^ (((s >> 5) & 1) << 1);
+    a_sl[it] = sl;
+    int m = tile_m * BM2 + (sl >> 3);
+    if (m >= p.M) m = 0;  // clamp: garbage row, never stored
+    const int n = m / HoWo;
+    const int rem = m - n * HoWo;
+    a_n[it] = n;
+    a_hwbase[it] = ((rem / p.Wo) * p.sh) * p.Wp + (rem % p.Wo) * p.sw;
+  }
+  int b_base[2], b_sl[2];
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int s = tid + it * 512;
+    const int sl = s ^ (((s >> 5) & 1) << 1);
+    b_sl[it] = sl;
+    int k = tile_n * BN2 + (sl >> 3);
+    if (k >= p.K) k = 0;  // clamp
+    b_base[it] = k * p.R * p.SPAN64 + (sl & 7) * 8;
+  }
+
+  auto stage = [&](int buf, int ks) {
+    const int r = ks / p.nspan;
+    const int span0 = (ks % p.nspan) * BK2;
+    // glds dest: wave-uniform base + lane*16
+    char* base = smem + buf * BUF_B + (tid >> 6) * 1024;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int span = span0 + (a_sl[it] & 7) * 8;
+      const int s_ = span / p.Cg;   // may reach S at the padded tail
+      const int c = span - s_ * p.Cg;
+      const __hip_bfloat16* src =
+          p.x +
+          ((int64_t)a_n[it] * p.Hp * p.Wp +
+           (a_hwbase[it] + r * p.dh * p.Wp + s_ * p.dw)) * p.C + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)(base + it * 8192),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const __hip_bfloat16* src = p.w + b_base[it] + r * p.SPAN64 + span0;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)(base + ATILE_B +
+                                                        it * 8192),
+          16, 0, 0);
+    }
+  };
+
+  // fragment LDS read byte offsets (swizzled), relative to a ring slot
+  // A: row = wm*64 + mi*16 + il ; B: row = wn*64 + ni*16 + il
+  int a_off[4][2], b_off[4][2];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc)
+      a_off[mi][kc] =
+          swz((wm * 64 + mi * 16 + il) * 128 + (kc * 32 + kq * 8) * 2);
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc)
+      b_off[ni][kc] = ATILE_B +
+          swz((wn * 64 + ni * 16 + il) * 128 + (kc * 32 + kq * 8) * 2);
+
+  f32x4c acc[4][4] = {};
+
+  // ---- prologue: fill the ring --------------------------------------------
+  const int nt = p.ksteps;
+  stage(0, 0);
+  if (nt > 1) stage(1, 1);
+  if (nt > 2) stage(2, 2);
+
+  // ---- main loop: counted vmcnt, raw barriers, loads in flight ------------
+  int bufsel = 0;
+  for (int t = 0; t < nt; ++t) {
+    const int infl = (nt - 1 - t) >= 2 ? 2 : (nt - 1 - t);
+    if (infl == 2) {
+      WAITVM(12);
+    } else if (infl == 1) {
+      WAITVM(6);
+    } else {
+      WAITVM(0);
+    }
+    __builtin_amdgcn_s_barrier();  // everyone's tile-t loads landed
+    const char* ta = smem + bufsel * BUF_B;
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      bf16x8c afrag[4], bfrag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        afrag[mi] = *reinterpret_cast<const bf16x8c*>(ta + a_off[mi][kc]);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bfrag[ni] = *reinterpret_cast<const bf16x8c*>(ta + b_off[ni][kc]);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // everyone done reading ring slot t%3
+    if (t + 3 < nt) stage(bufsel, t + 3);
+    bufsel = bufsel == 2 ? 0 : bufsel + 1;
+  }
+  __builtin_amdgcn_s_barrier();
+
+  // ---- epilogue: per-wave LDS restage -> 16-wide bf16 stores ---------------
+  float* slab = reinterpret_cast<float*>(smem) + wid * (16 * 68);
+  const int er = lane >> 2;
+  const int ec = (lane & 3) << 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr)
+        slab[(kq * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+    __builtin_amdgcn_wave_barrier();
+    const int m = tile_m * BM2 + wm * 64 + mi * 16 + er;
+    if (m < p.M) {
+      const int n = m / HoWo;
+      const int rem = m - n * HoWo;
+      const int64_t obase = ((int64_t)n * HoWo + rem) * p.K + tile_n * BN2 +
+                            wn * 64;
+      const int k0 = tile_n * BN2 + wn * 64 + ec;
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } u;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (k0 + 16 <= p.K) {
+        *reinterpret_cast<uint4*>(&p.y[obase + ec]) = u.q[0];
+        *reinterpret_cast<uint4*>(&p.y[obase + ec + 8]) = u.q[1];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          if (k0 + j < p.K) p.y[obase + ec + j] = u.b[j];
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+// ---- host-side transforms --------------------------------------------------
+template <typename T>
+__global__ void pad_image_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                 int N, int H, int W, int C, int Hp, int Wp,
+                                 int ph, int pw) {
+  const int64_t total = (int64_t)N * Hp * Wp * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int wp_ = t % Wp;
+    t /= Wp;
+    const int hp_ = t % Hp;
+    const int n = t / Hp;
+    const int h = hp_ - ph, w = wp_ - pw;
+    y[i] = (h >= 0 && h < H && w >= 0 && w < W)
+               ? x[(((int64_t)n * H + h) * W + w) * C + c]
+               : from_f32<T>(0.f);
+  }
+}
+
+template <typename T>
+__global__ void pad_weight_span_kernel(const T* __restrict__ w,
+                                       T* __restrict__ o, int K, int R,
+                                       int SC, int SPAN64) {
+  const int64_t total = (int64_t)K * R * SPAN64;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int sp = i % SPAN64;
+    int64_t t = i / SPAN64;
+    const int r = t % R;
+    const int k = t / R;
+    o[i] = sp < SC ? w[((int64_t)k * R + r) * SC + sp] : from_f32<T>(0.f);
+  }
+}
+
+}  // namespace
+
+at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
+  check_nhwc(x, "x");
+  check_nhwc(w, "w");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(Cg == C, "v2 is groups==1 only");
+  const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
+  const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  const int SC = S * C;
+  const int SPAN64 = (SC + BK2 - 1) / BK2 * BK2;
+
+  // padded input (+ slack for span-tail over-read)
+  at::Tensor xin = x;
+  int Hp = H, Wp = W;
+  if (ph > 0 || pw > 0) {
+    Hp = H + 2 * ph;
+    Wp = W + 2 * pw;
+    auto xp = at::empty({(int64_t)N * Hp * Wp * C + SPAN64 + 64},
+                        x.options());
+    int64_t total = (int64_t)N * Hp * Wp * C;
+    hipLaunchKernelGGL((pad_image_kernel<__hip_bfloat16>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (__hip_bfloat16*)xp.data_ptr(), N, H, W, C, Hp, Wp, ph,
+                       pw);
+    xin = xp;
+  } else if (SPAN64 != SC) {
+    // slack for span-tail reads past the logical end
+    auto xp = at::empty({(int64_t)N * H * W * C + SPAN64 + 64}, x.options());
+    // copy in PHYSICAL (NHWC) order
+    xp.narrow(0, 0, (int64_t)N * H * W * C)
+        .copy_(x.permute({0, 2, 3, 1}).reshape({-1}));
+    xin = xp;
+  }  // else: no padding, no span tail -> use x directly
+  // span-padded weight
+  at::Tensor wp = w;
+  if (SPAN64 != SC) {
+    wp = at::empty({(int64_t)K * R * SPAN64}, w.options());
+    int64_t total = (int64_t)K * R * SPAN64;
+    hipLaunchKernelGGL((pad_weight_span_kernel<__hip_bfloat16>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const __hip_bfloat16*)w.data_ptr(),
+                       (__hip_bfloat16*)wp.data_ptr(), K, R, SC, SPAN64);
+  }
+
+  auto y = at::empty({N, K, Ho, Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  Conv2Params p;
+  p.x = (const __hip_bfloat16*)xin.data_ptr();
+  p.w = (const __hip_bfloat16*)wp.data_ptr();
+  p.y = (__hip_bfloat16*)y.data_ptr();
+  p.N = N; p.Hp = Hp; p.Wp = Wp; p.C = C; p.K = K;
+  p.R = R; p.SPAN64 = SPAN64; p.Cg = C; p.S = S;
+  p.sh = sh; p.sw = sw; p.dh = dh; p.dw = dw;
+  p.Ho = Ho; p.Wo = Wo;
+  p.M = N * Ho * Wo;
+  p.nspan = SPAN64 / BK2;
+  p.ksteps = R * p.nspan;
+  p.tiles_m = (p.M + BM2 - 1) / BM2;
+  dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2);
+  hipLaunchKernelGGL(conv_igemm_v2_kernel, grid, dim3(512), 0, cur_stream(),
+                     p);
+  return y;
+}
