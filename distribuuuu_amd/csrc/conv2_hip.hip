@@ -276,6 +276,9 @@ at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     auto xp = at::empty({(int64_t)N * Hp * Wp * C + SPAN64 + 64},
                         x.options());
     int64_t total = (int64_t)N * Hp * Wp * C;
+    // span-tail reads of the LAST pixel land in the slack: it must be finite
+    // (tail A values multiply zero-padded B weights, but Inf*0 = NaN)
+    xp.narrow(0, total, SPAN64 + 64).zero_();
     hipLaunchKernelGGL((pad_image_kernel<__hip_bfloat16>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
                        (const __hip_bfloat16*)x.data_ptr(),
@@ -285,9 +288,10 @@ at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   } else if (SPAN64 != SC) {
     // slack for span-tail reads past the logical end
     auto xp = at::empty({(int64_t)N * H * W * C + SPAN64 + 64}, x.options());
-    // copy in PHYSICAL (NHWC) order
+    // copy in PHYSICAL (NHWC) order; zero the slack (read by span tails)
     xp.narrow(0, 0, (int64_t)N * H * W * C)
         .copy_(x.permute({0, 2, 3, 1}).reshape({-1}));
+    xp.narrow(0, (int64_t)N * H * W * C, SPAN64 + 64).zero_();
     xin = xp;
   }  // else: no padding, no span tail -> use x directly
   // span-padded weight
