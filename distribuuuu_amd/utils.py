@@ -197,7 +197,19 @@ def construct_meters(num_batches, prefix, topk=5):
 # Accuracy (reference utils.py:265-277)
 # ---------------------------------------------------------------------------
 def accuracy(output, target, topk=(1,)):
-    """Top-k accuracy as percentages."""
+    """Top-k accuracy as percentages. On GPU the fused per-row rank kernel
+     replaces topk+eq+sum."""
+    from .ops.dispatch import hip_op_available
+
+    if (output.is_cuda and len(topk) == 2 and topk[0] == 1
+            and hip_op_available("topk_acc")):
+        from .ops.dispatch import ext
+
+        with torch.no_grad():
+            c1, ck = ext().topk_acc(output.float().contiguous(),
+                                    target.contiguous(), topk[1])
+            n = output.shape[0]
+            return [c1[0].float() * (100.0 / n), ck[0].float() * (100.0 / n)]
     with torch.no_grad():
         maxk = max(topk)
         batch_size = target.size(0)
