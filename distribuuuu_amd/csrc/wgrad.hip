@@ -204,6 +204,204 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   }
 }
 
+// ---------------------------------------------------------------------------
+// Ring-staged wgrad (the conv-v2 structure applied to wgrad): packed 1 KiB
+// [32 m][16 col] subtiles filled by global_load_lds (lane-linear), tr_b16
+// fragment reads, 3-slot LDS ring with counted vmcnt across raw barriers.
+// Requirements (host-checked): M%64==0, Kg%64==0, (S*Cg)%64==0, Cg%8==0 and
+// a pre-padded image when ph/pw>0 (no predication anywhere; rsc/k-tail
+// contributions land in discarded output rows/cols).
+// ---------------------------------------------------------------------------
+constexpr int RSUBT = 1024;               // packed subtile
+constexpr int RTILE = 8 * RSUBT;          // one operand tile (8 KB)
+constexpr int RSLOT = 2 * RTILE;          // A + B per ring slot (16 KB)
+
+struct WgradRingParams {
+  const __hip_bfloat16* x;   // padded [N, Hp, Wp, Ct] when ph/pw > 0
+  const __hip_bfloat16* gy;  // [M, Kt]
+  float* acc;                // [Kt, R*S*Cg] zeroed
+  int Hp, Wp, Ct, Kt;
+  int R, S, Cg, Kg;
+  int sh, sw, dh, dw;
+  int Ho, Wo;
+  int M, RSC;
+  int ktiles, ntiles;
+};
+
+#define RWAITVM(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
+
+__global__ __launch_bounds__(256) void conv_wgrad_ring_kernel(
+    WgradRingParams p) {
+  const int g = blockIdx.z;
+  const int ktile = blockIdx.x % p.ktiles;
+  const int ntile = blockIdx.x / p.ktiles;
+  const int chunk = blockIdx.y;
+
+  __shared__ __align__(16) char smem[3 * RSLOT];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int il = lane & 15, kq = lane >> 4;
+
+  // staging geometry: slot s = tid + it*256 (it<2); subtile = s/64,
+  // msub = subtile/4, colsub = subtile%4, inner = s%64:
+  //   mloc = msub*32 + inner/2 ; col = colsub*16 + (inner%2)*8
+  int s_mloc[2], s_col[2];
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int s = tid + it * 256;
+    const int sub = s >> 6, inner = s & 63;
+    s_mloc[it] = (sub >> 2) * 32 + (inner >> 1);
+    s_col[it] = (sub & 3) * 16 + (inner & 1) * 8;
+  }
+  // B (x) rsc decomposition per thread (fixed): clamp tail to 0 (its output
+  // column >= RSC is never written)
+  const int SCg = p.S * p.Cg;
+  int b_r[2], b_s[2], b_c[2];
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int rsc = ntile * WBN + s_col[it];
+    if (rsc >= p.RSC) rsc = 0;
+    b_r[it] = rsc / SCg;
+    const int rem = rsc - b_r[it] * SCg;
+    b_s[it] = rem / p.Cg;
+    b_c[it] = rem - b_s[it] * p.Cg;
+  }
+  const int m0 = chunk * (CHUNK_STEPS * WBK);
+  const int HoWo = p.Ho * p.Wo;
+  const int nt = min(CHUNK_STEPS, (int)((p.M - m0 + WBK - 1) / WBK));
+
+  auto stage = [&](int buf, int ks) {
+    char* base = smem + buf * RSLOT + wid * 1024;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int m = m0 + ks * WBK + s_mloc[it];  // < M (M%64==0)
+      // A: gy[m][ktile*64 + col]
+      const __hip_bfloat16* asrc =
+          p.gy + (int64_t)m * p.Kt + g * p.Kg + ktile * WBM + s_col[it];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)asrc,
+          (__attribute__((address_space(3))) uint32_t*)(base + it * 4096),
+          16, 0, 0);
+      // B: x[pixel(m) shifted by (r, s)][c]
+      const int n = m / HoWo;
+      const int rem = m - n * HoWo;
+      const int h = (rem / p.Wo) * p.sh + b_r[it] * p.dh;
+      const int w_ = (rem % p.Wo) * p.sw + b_s[it] * p.dw;
+      const __hip_bfloat16* bsrc =
+          p.x + (((int64_t)n * p.Hp + h) * p.Wp + w_) * p.Ct + g * p.Cg +
+          b_c[it];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)bsrc,
+          (__attribute__((address_space(3))) uint32_t*)(base + RTILE +
+                                                        it * 4096),
+          16, 0, 0);
+    }
+  };
+
+  // tr_b16 fragment read (packed subtiles, stride 1024)
+  auto tr_frag = [&](const char* tile, int colsub, int mc) {
+    const int s = lane & 15, gq_ = lane >> 4;
+    const int base_off = (mc * 4 + colsub) * RSUBT + (gq_ * 8 + (s >> 2)) * 32 +
+                         (s & 3) * 8;
+    const unsigned a0 = (unsigned)(unsigned long long)(tile + base_off);
+    const unsigned a1 = a0 + 128;  // +4 m-rows
+    union {
+      uint2 hh[2];
+      bf16x8w vv;
+    } uu;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2\n\t"
+        "ds_read_b64_tr_b16 %1, %3\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=&v"(uu.hh[0]), "=&v"(uu.hh[1])
+        : "v"(a0), "v"(a1)
+        : "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    return uu.vv;
+  };
+
+  f32x4w accv[2][2] = {};
+
+  stage(0, 0);
+  if (nt > 1) stage(1, 1);
+  if (nt > 2) stage(2, 2);
+
+  int bufsel = 0;
+  for (int t = 0; t < nt; ++t) {
+    const int infl = (nt - 1 - t) >= 2 ? 2 : (nt - 1 - t);
+    if (infl == 2) {
+      RWAITVM(8);
+    } else if (infl == 1) {
+      RWAITVM(4);
+    } else {
+      RWAITVM(0);
+    }
+    __builtin_amdgcn_s_barrier();
+    const char* ta = smem + bufsel * RSLOT;
+    const char* tb = ta + RTILE;
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {
+      bf16x8w afrag[2], bfrag[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        afrag[mi] = tr_frag(ta, wm * 2 + mi, mc);
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        bfrag[ni] = tr_frag(tb, wn * 2 + ni, mc);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], accv[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + 3 < nt) stage(bufsel, t + 3);
+    bufsel = bufsel == 2 ? 0 : bufsel + 1;
+  }
+
+  // epilogue: fp32 atomic accumulate (D: col=lane&15, row=(lane>>4)*4+rr)
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int k = ktile * WBM + wm * 32 + mi * 16 + kq * 4 + rr;
+      if (k >= p.Kg) continue;
+      const int64_t rowbase = (int64_t)(g * p.Kg + k) * p.RSC;
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int col = ntile * WBN + wn * 32 + ni * 16 + il;
+        if (col < p.RSC) atomicAdd(&p.acc[rowbase + col], accv[mi][ni][rr]);
+      }
+    }
+  }
+}
+
+template <typename T>
+__global__ void wg_pad_image_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                    int N, int H, int W, int C, int Hp, int Wp,
+                                    int ph, int pw) {
+  const int64_t total = (int64_t)N * Hp * Wp * C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    int64_t t = i / C;
+    const int wp_ = t % Wp;
+    t /= Wp;
+    const int hp_ = t % Hp;
+    const int n = t / Hp;
+    const int h = hp_ - ph, w = wp_ - pw;
+    y[i] = (h >= 0 && h < H && w >= 0 && w < W)
+               ? x[(((int64_t)n * H + h) * W + w) * C + c]
+               : from_f32<T>(0.f);
+  }
+}
+
 __global__ void cast_acc_kernel(const float* __restrict__ acc,
                                 __hip_bfloat16* __restrict__ gw,
                                 int64_t total) {
@@ -225,6 +423,58 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
   const int Kt = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
   const int Cg = Ct / groups, Kg = Kt / groups;
   TORCH_CHECK(Cg % 8 == 0 && Kg % 8 == 0, "wgrad: Cg/Kg must be multiples of 8");
+  const int64_t M64 = (int64_t)N * Ho * Wo;
+  const bool ring_ok = (M64 % 64 == 0) && (Kg % 64 == 0) &&
+                       ((S * Cg) % 64 == 0) && (Cg % 8 == 0) &&
+                       ((R * S * Cg) % 16 == 0);
+  // measured: the ring variant trails the tr-staged kernel on most shapes
+  // (its 8-MFMA k-steps don't cover the pipeline); keep it opt-in.
+  static const bool ring_on = []() {
+    const char* e = getenv("DISTRIBUUUU_WGRAD_RING");
+    return e && e[0] == '1';
+  }();
+  if (ring_ok && ring_on) {
+    at::Tensor xin = x;
+    int Hp = H, Wp = W;
+    if (ph > 0 || pw > 0) {
+      Hp = H + 2 * ph;
+      Wp = W + 2 * pw;
+      auto xp = at::empty({(int64_t)N * Hp * Wp * Ct}, x.options());
+      int64_t total = xp.numel();
+      hipLaunchKernelGGL((wg_pad_image_kernel<__hip_bfloat16>),
+                         dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         (__hip_bfloat16*)xp.data_ptr(), N, H, W, Ct, Hp, Wp,
+                         ph, pw);
+      xin = xp;
+    }
+    WgradRingParams q;
+    q.x = (const __hip_bfloat16*)xin.data_ptr();
+    q.gy = (const __hip_bfloat16*)gy.data_ptr();
+    q.Hp = Hp; q.Wp = Wp; q.Ct = Ct; q.Kt = Kt;
+    q.R = R; q.S = S; q.Cg = Cg; q.Kg = Kg;
+    q.sh = sh; q.sw = sw; q.dh = dh; q.dw = dw;
+    q.Ho = Ho; q.Wo = Wo;
+    q.M = (int)M64;
+    q.RSC = R * S * Cg;
+    q.ktiles = Kg / WBM;
+    q.ntiles = (q.RSC + WBN - 1) / WBN;
+    const int chunks = (int)((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK));
+    auto accbuf2 = at::empty({(int64_t)Kt, q.RSC}, x.options().dtype(at::kFloat));
+    accbuf2.zero_();
+    q.acc = accbuf2.data_ptr<float>();
+    dim3 grid2(q.ktiles * q.ntiles, chunks, groups);
+    hipLaunchKernelGGL(conv_wgrad_ring_kernel, grid2, dim3(256), 0,
+                       cur_stream(), q);
+    auto gw2 = at::empty({Kt, Cg, (int64_t)R, (int64_t)S},
+                         x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const int64_t total2 = (int64_t)Kt * q.RSC;
+    hipLaunchKernelGGL(cast_acc_kernel, dim3(grid_1d(total2, 256)), dim3(256),
+                       0, cur_stream(), q.acc,
+                       (__hip_bfloat16*)gw2.data_ptr(), total2);
+    return gw2;
+  }
+
   WgradParams p;
   p.x = (const __hip_bfloat16*)x.data_ptr();
   p.gy = (const __hip_bfloat16*)gy.data_ptr();
