@@ -23,22 +23,32 @@ from .resnet import resnet50
 
 class MHSA(nn.Module):
     """Multi-head self-attention over an HxW feature map with decomposed
-    relative position embeddings (reference botnet.py:163-215)."""
+    relative position embeddings (reference botnet.py:163-215), or absolute
+    position embeddings when rel_pos_emb=False (reference AbsPosEmb,
+    botnet.py:60-74)."""
 
-    def __init__(self, dim, fmap_size, heads=4, dim_qk=128, dim_v=128):
+    def __init__(self, dim, fmap_size, heads=4, dim_qk=128, dim_v=128,
+                 rel_pos_emb=True):
         super().__init__()
         self.heads = heads
         self.dim_qk = dim_qk
         self.dim_v = dim_v
         self.scale = dim_qk ** -0.5
         self.fmap_size = fmap_size
+        self.rel_pos_emb = rel_pos_emb
         out_qk = heads * dim_qk
         out_v = heads * dim_v
         self.to_qk = Conv2d(dim, 2 * out_qk, 1, bias=False)
         self.to_v = Conv2d(dim, out_v, 1, bias=False)
         h, w = fmap_size
-        self.rel_h = nn.Parameter(torch.randn(2 * h - 1, dim_qk) * dim_qk ** -0.5)
-        self.rel_w = nn.Parameter(torch.randn(2 * w - 1, dim_qk) * dim_qk ** -0.5)
+        if rel_pos_emb:
+            self.rel_h = nn.Parameter(
+                torch.randn(2 * h - 1, dim_qk) * dim_qk ** -0.5)
+            self.rel_w = nn.Parameter(
+                torch.randn(2 * w - 1, dim_qk) * dim_qk ** -0.5)
+        else:
+            self.abs_h = nn.Parameter(torch.randn(h, dim_qk) * dim_qk ** -0.5)
+            self.abs_w = nn.Parameter(torch.randn(w, dim_qk) * dim_qk ** -0.5)
 
     def forward(self, x):
         n, _, hh, ww = x.shape
@@ -49,7 +59,17 @@ class MHSA(nn.Module):
         q = q.reshape(n, self.heads, self.dim_qk, hh * ww).transpose(2, 3)
         k = k.reshape(n, self.heads, self.dim_qk, hh * ww).transpose(2, 3)
         v = v.reshape(n, self.heads, self.dim_v, hh * ww).transpose(2, 3)
-        out = mhsa_relpos(q * self.scale, k, v, self.rel_h, self.rel_w, hh, ww)
+        if self.rel_pos_emb:
+            out = mhsa_relpos(q * self.scale, k, v, self.rel_h, self.rel_w,
+                              hh, ww)
+        else:
+            qs = q * self.scale
+            emb = (self.abs_h.unsqueeze(1) + self.abs_w.unsqueeze(0)).reshape(
+                hh * ww, self.dim_qk)
+            logits = torch.einsum("bhxd,bhyd->bhxy", qs, k)
+            logits = logits + torch.einsum("bhxd,yd->bhxy", qs, emb)
+            attn = torch.softmax(logits, dim=-1)
+            out = torch.einsum("bhxy,bhyd->bhxd", attn, v)
         return out.transpose(2, 3).reshape(n, self.heads * self.dim_v, hh, ww)
 
 
@@ -58,7 +78,7 @@ class BoTBlock(nn.Module):
     1x1 up -> BN (zero-init gamma) + residual + ReLU (reference botnet.py:101-160)."""
 
     def __init__(self, dim, fmap_size, dim_out, stride=1, heads=4, proj_factor=4,
-                 dim_qk=128, dim_v=128):
+                 dim_qk=128, dim_v=128, rel_pos_emb=True):
         super().__init__()
         self.shortcut = None
         if dim != dim_out or stride != 1:
@@ -70,7 +90,8 @@ class BoTBlock(nn.Module):
         attn_out = heads * dim_v
         self.conv1 = Conv2d(dim, bottleneck_dim, 1, bias=False)
         self.bn1 = BatchNorm2d(bottleneck_dim, act="relu")
-        self.mhsa = MHSA(bottleneck_dim, fmap_size, heads, dim_qk, dim_v)
+        self.mhsa = MHSA(bottleneck_dim, fmap_size, heads, dim_qk, dim_v,
+                         rel_pos_emb)
         self.pool = AvgPool2d(2) if stride == 2 else None
         self.bn2 = BatchNorm2d(attn_out, act="relu")
         self.conv3 = Conv2d(attn_out, dim_out, 1, bias=False)
@@ -92,7 +113,7 @@ class BoTStack(nn.Module):
     """Stack of 3 BoTBlocks replacing ResNet c5 (reference botnet.py:218-290)."""
 
     def __init__(self, dim=1024, fmap_size=(14, 14), dim_out=2048, heads=4,
-                 proj_factor=4, num_layers=3, stride=2):
+                 proj_factor=4, num_layers=3, stride=2, rel_pos_emb=True):
         super().__init__()
         blocks = []
         fm = fmap_size
@@ -101,7 +122,8 @@ class BoTStack(nn.Module):
             s = stride if is_first else 1
             blocks.append(
                 BoTBlock(dim if is_first else dim_out, fm, dim_out, stride=s,
-                         heads=heads, proj_factor=proj_factor)
+                         heads=heads, proj_factor=proj_factor,
+                         rel_pos_emb=rel_pos_emb)
             )
             if is_first and stride == 2:
                 fm = (fm[0] // 2, fm[1] // 2)
