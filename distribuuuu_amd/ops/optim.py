@@ -48,7 +48,10 @@ class HIPSGD(SGD):
                              cnt | (flags << 32)])
                 off += _CHUNK
         device = group["params"][0].device
-        table = torch.tensor(recs, dtype=torch.int64).to(device)
+        # pinned staging + async copy: capturable inside a hipGraph
+        cpu = torch.tensor(recs, dtype=torch.int64).pin_memory()
+        table = cpu.to(device, non_blocking=True)
+        group["_hip_table_pin"] = cpu  # keep the pinned buffer alive
         return table, tuple(key)
 
     @torch.no_grad()
@@ -81,6 +84,7 @@ class HIPSGD(SGD):
         # drop the pointer cache from serialized state
         for group in self.param_groups:
             group.pop("_hip_table", None)
+            group.pop("_hip_table_pin", None)
         return super().state_dict()
 
     def _fallback_step(self, loss):
