@@ -48,10 +48,22 @@ class HIPSGD(SGD):
                              cnt | (flags << 32)])
                 off += _CHUNK
         device = group["params"][0].device
-        # pinned staging + async copy: capturable inside a hipGraph
-        cpu = torch.tensor(recs, dtype=torch.int64).pin_memory()
-        table = cpu.to(device, non_blocking=True)
-        group["_hip_table_pin"] = cpu  # keep the pinned buffer alive
+        # pinned staging + async copy. The pinned buffer and device table are
+        # allocated on the FIRST build (outside any hipGraph capture, during
+        # warmup); capture-time rebuilds (grad pointers change under
+        # zero_grad(set_to_none=True)) only do host writes + one capturable
+        # async H2D copy into the existing device tensor.
+        cpu = torch.tensor(recs, dtype=torch.int64)
+        pin = group.get("_hip_table_pin")
+        table = group.get("_hip_table_dev")
+        if pin is None or pin.numel() != cpu.numel():
+            pin = cpu.pin_memory()
+            table = pin.to(device, non_blocking=True)
+            group["_hip_table_pin"] = pin
+            group["_hip_table_dev"] = table
+        else:
+            pin.copy_(cpu)
+            table.copy_(pin, non_blocking=True)
         return table, tuple(key)
 
     @torch.no_grad()
@@ -85,6 +97,7 @@ class HIPSGD(SGD):
         for group in self.param_groups:
             group.pop("_hip_table", None)
             group.pop("_hip_table_pin", None)
+            group.pop("_hip_table_dev", None)
         return super().state_dict()
 
     def _fallback_step(self, loss):
