@@ -14,6 +14,16 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
                                  double momentum, double eps, bool training);
 at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
                         int64_t act, c10::optional<at::Tensor> res);
+at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
+                        c10::optional<at::Tensor> res, at::Tensor scale,
+                        at::Tensor shift, int64_t act);
+std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
+                                     c10::optional<at::Tensor> res,
+                                     at::Tensor mean, at::Tensor rstd,
+                                     at::Tensor gamma, at::Tensor scale,
+                                     at::Tensor shift, at::Tensor sums,
+                                     double total_count, int64_t act,
+                                     bool training, bool need_gres);
 std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
@@ -80,6 +90,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("training") = true);
   m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
+  m.def("bn_bwd_stats", &bn_bwd_stats, py::arg("gy"), py::arg("x"),
+        py::arg("y"), py::arg("res"), py::arg("scale"), py::arg("shift"),
+        py::arg("act"));
+  m.def("bn_bwd_apply", &bn_bwd_apply, py::arg("gy"), py::arg("x"),
+        py::arg("y"), py::arg("res"), py::arg("mean"), py::arg("rstd"),
+        py::arg("gamma"), py::arg("scale"), py::arg("shift"), py::arg("sums"),
+        py::arg("total_count"), py::arg("act"), py::arg("training"),
+        py::arg("need_gres"));
   m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"), py::arg("y"),
         py::arg("res"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
         py::arg("scale"), py::arg("shift"), py::arg("act"),

@@ -596,6 +596,99 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
   return y;
 }
 
+// Local raw grad-stat sums [2C] = [sum(g), sum(g*x)] (stage 1 + stage 2);
+// used standalone by SyncBN (which all-reduces the result across ranks).
+at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
+                        c10::optional<at::Tensor> res, at::Tensor scale,
+                        at::Tensor shift, int64_t act) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int C = x.size(1);
+  const int64_t rows = x.numel() / C;
+  auto fopts = x.options().dtype(at::kFloat);
+  const bool has_res = res.has_value();
+  at::Tensor both;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd_stats", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    auto stream = cur_stream();
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
+    auto part = at::empty({rgrid, 2 * C}, fopts);
+    const scalar_t* resp =
+        has_res ? (const scalar_t*)res->data_ptr() : nullptr;
+    if (has_res)
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true>), dim3(rgrid),
+                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), resp,
+                         scale.data_ptr<float>(), shift.data_ptr<float>(),
+                         part.data_ptr<float>(), rows, C, rpb, (int)act);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
+                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), resp,
+                         scale.data_ptr<float>(), shift.data_ptr<float>(),
+                         part.data_ptr<float>(), rows, C, rpb, (int)act);
+    both = at::empty({2 * C}, fopts);
+    hipLaunchKernelGGL(reduce_partials_kernel, dim3((int)ceil_div(2 * C, 64)),
+                       dim3(256), 0, stream, part.data_ptr<float>(),
+                       both.data_ptr<float>(), 2 * C, rgrid);
+  });
+  return both;
+}
+
+// dx/gw/gb given (possibly cross-rank-reduced) raw sums and the TOTAL count.
+std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
+                                     c10::optional<at::Tensor> res,
+                                     at::Tensor mean, at::Tensor rstd,
+                                     at::Tensor gamma, at::Tensor scale,
+                                     at::Tensor shift, at::Tensor sums,
+                                     double total_count, int64_t act,
+                                     bool training, bool need_gres) {
+  const int C = x.size(1);
+  auto fopts = x.options().dtype(at::kFloat);
+  const bool has_res = res.has_value();
+  auto gx = at::empty_like(x);
+  auto gres = need_gres ? at::empty_like(x) : at::Tensor();
+  auto gw = at::empty({C}, fopts);
+  auto gb = at::empty({C}, fopts);
+  auto P1 = at::empty({C}, fopts);
+  auto P2 = at::empty({C}, fopts);
+  auto P3 = at::empty({C}, fopts);
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd_apply", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    auto stream = cur_stream();
+    const int cpacks = C / V;
+    const float inv_cnt = 1.f / (float)total_count;
+    // sums is a flat [2C] row => treat as a 1-block partial
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 64)),
+                       dim3(256), 0, stream, sums.data_ptr<float>(), 1,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), gw.data_ptr<float>(),
+                       gb.data_ptr<float>(), P1.data_ptr<float>(),
+                       P2.data_ptr<float>(), P3.data_ptr<float>(), C, inv_cnt,
+                       training ? 1 : 0);
+    int64_t npacks = x.numel() / V;
+    int grid = grid_1d(npacks, 256);
+    const scalar_t* resp =
+        has_res ? (const scalar_t*)res->data_ptr() : nullptr;
+    scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
+    if (has_res)
+      launch_dx<scalar_t, V, true>(gy, x, y, resp, scale, shift, P1, P2, P3,
+                                   gx, gresp, npacks, cpacks, (int)act, grid,
+                                   stream);
+    else
+      launch_dx<scalar_t, V, false>(gy, x, y, resp, scale, shift, P1, P2, P3,
+                                    gx, gresp, npacks, cpacks, (int)act, grid,
+                                    stream);
+  });
+  return {gx, gw, gb, gres};
+}
+
 // Full backward. scale/shift are the forward's fused coefficients (needed to
 // recompute z for silu/sigmoid activation backward).
 std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,

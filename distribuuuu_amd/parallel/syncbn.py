@@ -2,17 +2,18 @@
 
 Replaces torch.nn.SyncBatchNorm (reference `trainer.py:131`,
 `nn.SyncBatchNorm.convert_sync_batchnorm`). Forward computes local per-channel
-sum / sum-of-squares (on GPU via the bn_stats HIP kernel), all-reduces ONE
-coalesced [2C+1] fp32 tensor (the reference's implementation all-reduces
-mean & var separately per layer), normalizes with the fused apply(+residual)
-(+act) kernel, and backward all-reduces the two grad-stat vectors in one
-message. Keeps the convert_sync_batchnorm(module) toggle semantics of
-cfg.MODEL.SYNCBN.
+sum / sum-of-squares with the bn_sums HIP kernel, all-reduces ONE coalesced
+[2C+1] fp32 tensor (the reference's implementation all-reduces mean & var
+separately per layer), and normalizes with the fused apply(+residual)(+act)
+kernel. Backward computes local RAW grad-stat sums with bn_bwd_stats,
+all-reduces the [2C] vector, and finishes with the same fused finalize+dx
+kernels the single-rank path uses (bn_bwd_apply) — no ATen fp32
+materializations on the hot path. A plain-torch composition backs the CPU /
+gloo test tier.
 """
 
 import torch
 import torch.distributed as dist
-import torch.nn.functional as F
 
 from ..ops.modules import BatchNorm2d
 from ..ops import functional as DF
@@ -23,56 +24,76 @@ class _SyncBNFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps,
                 act, residual, process_group):
-        world_size = dist.get_world_size(process_group)
         c = x.shape[1]
-        if use_hip(x, "bn_stats"):
+        gamma = weight.float().contiguous()
+        beta = bias.float().contiguous()
+        hip = use_hip(x, "bn_sums")
+        if hip:
             x = DF._cl(x)
-            s, ss = ext().bn_sums(x)  # fp32 [C] sum, sumsq
+            s, ss = ext().bn_sums(x)  # fp32 [C] views of one [2C] tensor
         else:
             xf = x.float()
             s = xf.sum(dim=(0, 2, 3))
             ss = (xf * xf).sum(dim=(0, 2, 3))
-        count = torch.full((1,), x.numel() / c, dtype=torch.float32,
-                           device=x.device)
-        packed = torch.cat([s, ss, count])          # ONE [2C+1] collective
+        # equal per-rank counts in training (DistributedSampler, drop_last):
+        # total is computed host-side so no .item() device sync per layer
+        world = dist.get_world_size(process_group)
+        total = (x.numel() / c) * world
+        packed = torch.cat([s, ss])                 # ONE [2C] collective
         dist.all_reduce(packed, group=process_group)
-        s, ss, total = packed[:c], packed[c:2 * c], packed[2 * c].item()
+        s, ss = packed[:c], packed[c:]
         mean = s / total
-        var = ss / total - mean * mean
+        var = (ss / total - mean * mean).clamp_(min=0)
         if running_mean is not None:
             with torch.no_grad():
                 unbiased = var * (total / max(total - 1, 1))
                 running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                 running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
         rstd = (var + eps).rsqrt()
-        if use_hip(x, "bn_apply_act"):
-            res = DF._cl(residual) if residual is not None else None
-            scale = weight.float() * rstd
-            shift = bias.float() - mean * scale
-            y = ext().bn_apply_act(x, scale, shift, DF._ACTS[act], res)
+        scale = gamma * rstd
+        shift = beta - mean * scale
+        res = None
+        if residual is not None:
+            res = DF._cl(residual) if hip else residual
+        if hip:
+            y = ext().bn_apply_act(x, scale.contiguous(), shift.contiguous(),
+                                   DF._ACTS[act], res)
         else:
-            y = (x.float() - mean.reshape(1, -1, 1, 1)) * rstd.reshape(1, -1, 1, 1)
-            y = y * weight.float().reshape(1, -1, 1, 1) + bias.float().reshape(1, -1, 1, 1)
-            if residual is not None:
-                y = y + residual.float()
+            y = x.float() * scale.reshape(1, -1, 1, 1) + shift.reshape(1, -1, 1, 1)
+            if res is not None:
+                y = y + res.float()
             y = DF._apply_act(y, act).to(x.dtype)
-        ctx.save_for_backward(x, weight, mean, rstd, y)
-        ctx.meta = (act, total, residual is not None, process_group)
+        ctx.save_for_backward(x, gamma, scale.contiguous(), shift.contiguous(),
+                              mean.contiguous(), rstd.contiguous(), y,
+                              res if res is not None else x.new_empty(0))
+        ctx.meta = (act, total, residual is not None, process_group, hip,
+                    weight.dtype)
         return y
 
     @staticmethod
     def backward(ctx, gy):
-        x, weight, mean, rstd, y = ctx.saved_tensors
-        act, total, has_res, group = ctx.meta
-        gy = gy.contiguous()
-        # activation backward through the saved post-act output
-        if act == "relu":
-            gy = gy * (y > 0).to(gy.dtype)
-        elif act != "none":
-            raise NotImplementedError(f"SyncBN act backward: {act}")
-        gres = gy if has_res else None
-        xf = x.float()
+        x, gamma, scale, shift, mean, rstd, y, res = ctx.saved_tensors
+        act, total, has_res, group, hip, w_dtype = ctx.meta
+        act_id = DF._ACTS[act]
+        if hip:
+            gy = DF._cl(gy)
+            e = ext()
+            sums = e.bn_bwd_stats(gy, x, y, res if has_res else None,
+                                  scale, shift, act_id)      # raw [2C]
+            dist.all_reduce(sums, group=group)
+            gx, gw, gb, gres = e.bn_bwd_apply(
+                gy, x, y, res if has_res else None, mean, rstd, gamma,
+                scale, shift, sums, total, act_id, True, has_res)
+            return (gx, gw.to(w_dtype), gb.to(w_dtype), None, None, None,
+                    None, None, gres if has_res else None, None)
+        # CPU / test-tier composition
         gyf = gy.float()
+        if act == "relu":
+            gyf = gyf * (y > 0).float()
+        elif act != "none":
+            raise NotImplementedError(f"SyncBN CPU act backward: {act}")
+        gres = gyf.to(gy.dtype) if has_res else None
+        xf = x.float()
         xhat = (xf - mean.reshape(1, -1, 1, 1)) * rstd.reshape(1, -1, 1, 1)
         sum_gy = gyf.sum(dim=(0, 2, 3))
         sum_gy_xhat = (gyf * xhat).sum(dim=(0, 2, 3))
@@ -80,12 +101,10 @@ class _SyncBNFunction(torch.autograd.Function):
         dist.all_reduce(packed, group=group)
         c = x.shape[1]
         sum_gy, sum_gy_xhat = packed[:c], packed[c:]
-        gw = sum_gy_xhat
-        gb = sum_gy
-        w_rstd = (weight.float() * rstd).reshape(1, -1, 1, 1)
+        w_rstd = (gamma * rstd).reshape(1, -1, 1, 1)
         gx = w_rstd * (gyf - (sum_gy.reshape(1, -1, 1, 1)
                               + xhat * sum_gy_xhat.reshape(1, -1, 1, 1)) / total)
-        return (gx.to(x.dtype), gw.to(weight.dtype), gb.to(weight.dtype),
+        return (gx.to(x.dtype), sum_gy_xhat.to(w_dtype), sum_gy.to(w_dtype),
                 None, None, None, None, None, gres, None)
 
 
@@ -102,9 +121,10 @@ class SyncBatchNorm(BatchNorm2d):
                 or dist.get_world_size(self.process_group) == 1):
             return super().forward(x, residual)
         self.num_batches_tracked += 1
+        rm, rv = self.running_mean, self.running_var
         return _SyncBNFunction.apply(
-            x, self.weight, self.bias, self.running_mean, self.running_var,
-            self.momentum, self.eps, self.act, residual, self.process_group)
+            x, self.weight, self.bias, rm, rv, self.momentum, self.eps,
+            self.act, residual, self.process_group)
 
     @classmethod
     def convert_sync_batchnorm(cls, module, process_group=None):
