@@ -68,10 +68,13 @@ __global__ void bn_sums_kernel(const T* __restrict__ x,
     float acc[V] = {}, acc2[V] = {};
     if (active) {
       int64_t r = row0 + rl;
+      const int64_t rstep = (int64_t)nrl * cpacks;
+      const P* xq = xp + r * cpacks + cp;
       for (; r + 7 * (int64_t)nrl < row1; r += 8 * (int64_t)nrl) {
         P pk[8];
 #pragma unroll
-        for (int u = 0; u < 8; ++u) pk[u] = xp[(r + u * nrl) * cpacks + cp];
+        for (int u = 0; u < 8; ++u) pk[u] = xq[u * rstep];
+        xq += 8 * rstep;
 #pragma unroll
         for (int u = 0; u < 8; ++u)
 #pragma unroll
@@ -82,7 +85,8 @@ __global__ void bn_sums_kernel(const T* __restrict__ x,
           }
       }
       for (; r < row1; r += nrl) {
-        P p = xp[r * cpacks + cp];
+        P p = xq[0];
+        xq += rstep;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float v = to_f32(p.v[j]);
@@ -157,12 +161,28 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part, int nblocks,
                                    float* __restrict__ shift, int C,
                                    float inv_cnt, float unbias, float momentum,
                                    float eps, int update_running) {
+  // 16 channels x 16 row-lanes, 4-tap unroll: the previous 64x4 mapping left
+  // one load-latency chain of nblocks/4 iterations per thread (~36us per
+  // launch at nblocks=512 — stage 2 cost MORE than the bandwidth-bound
+  // stage 1).
   __shared__ float red[512];
-  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  const int rlane = threadIdx.x >> 6;
+  const int ci = threadIdx.x & 15;
+  const int rlane = threadIdx.x >> 4;
+  const int c = blockIdx.x * 16 + ci;
   float sum = 0.f, sumsq = 0.f;
   if (c < C) {
-    for (int b = rlane; b < nblocks; b += 4) {
+    int b = rlane;
+    for (; b + 48 < nblocks; b += 64) {
+      sum += (part[(int64_t)b * 2 * C + c] +
+              part[(int64_t)(b + 16) * 2 * C + c]) +
+             (part[(int64_t)(b + 32) * 2 * C + c] +
+              part[(int64_t)(b + 48) * 2 * C + c]);
+      sumsq += (part[(int64_t)b * 2 * C + C + c] +
+                part[(int64_t)(b + 16) * 2 * C + C + c]) +
+               (part[(int64_t)(b + 32) * 2 * C + C + c] +
+                part[(int64_t)(b + 48) * 2 * C + C + c]);
+    }
+    for (; b < nblocks; b += 16) {
       sum += part[(int64_t)b * 2 * C + c];
       sumsq += part[(int64_t)b * 2 * C + C + c];
     }
@@ -171,10 +191,11 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part, int nblocks,
   red[256 + threadIdx.x] = sumsq;
   __syncthreads();
   if (rlane != 0 || c >= C) return;
-  sum = red[threadIdx.x] + red[threadIdx.x + 64] + red[threadIdx.x + 128] +
-        red[threadIdx.x + 192];
-  sumsq = red[256 + threadIdx.x] + red[256 + threadIdx.x + 64] +
-          red[256 + threadIdx.x + 128] + red[256 + threadIdx.x + 192];
+#pragma unroll
+  for (int t = 1; t < 16; ++t) {
+    sum += red[t * 16 + ci];
+    sumsq += red[256 + t * 16 + ci];
+  }
   const float m = sum * inv_cnt;
   float v = sumsq * inv_cnt - m * m;
   v = fmaxf(v, 0.f);
@@ -329,16 +350,27 @@ __global__ void bn_bwd_reduce_kernel(
     float accg[V] = {}, accgx[V] = {};
     if (active) {
       int64_t row = row0 + rl;
+      // pointer-increment addressing: a per-access (row*cpacks + cp) 64-bit
+      // multiply throttled this loop to ~1 TB/s (same diagnosis as
+      // bn_apply_kernel above)
+      const int64_t rstep = (int64_t)nrl * cpacks;
+      const P* gq = gp + row * cpacks + cp;
+      const P* xq = xp + row * cpacks + cp;
+      const P* yq = ypk + row * cpacks + cp;
+      const P* rq = rp + row * cpacks + cp;
       for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
         P pg4[4], px4[4], py4[4], pr4[4];
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
-          const int64_t off = (row + u * nrl) * cpacks + cp;
-          pg4[u] = gp[off];
-          px4[u] = xp[off];
-          if (act != 0) py4[u] = ypk[off];
-          if (HAS_RES && act >= 2) pr4[u] = rp[off];
+          pg4[u] = gq[u * rstep];
+          px4[u] = xq[u * rstep];
+          if (act != 0) py4[u] = yq[u * rstep];
+          if (HAS_RES && act >= 2) pr4[u] = rq[u * rstep];
         }
+        gq += 4 * rstep;
+        xq += 4 * rstep;
+        yq += 4 * rstep;
+        rq += 4 * rstep;
 #pragma unroll
         for (int u = 0; u < 4; ++u)
 #pragma unroll
@@ -358,10 +390,13 @@ __global__ void bn_bwd_reduce_kernel(
           }
       }
       for (; row < row1; row += nrl) {
-        const int64_t off = row * cpacks + cp;
-        P pg = gp[off], px = xp[off], py, prr;
-        if (act != 0) py = ypk[off];
-        if (HAS_RES && act >= 2) prr = rp[off];
+        P pg = gq[0], px = xq[0], py, prr;
+        if (act != 0) py = yq[0];
+        if (HAS_RES && act >= 2) prr = rq[0];
+        gq += rstep;
+        xq += rstep;
+        yq += rstep;
+        rq += rstep;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float xv = to_f32(px.v[j]);
@@ -419,11 +454,23 @@ __global__ void bn_bwd_finalize_kernel(
     float* __restrict__ gb, float* __restrict__ P1, float* __restrict__ P2,
     float* __restrict__ P3, int C, float inv_cnt, int training) {
   __shared__ float red[512];
-  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  const int rlane = threadIdx.x >> 6;
+  const int ci = threadIdx.x & 15;
+  const int rlane = threadIdx.x >> 4;
+  const int c = blockIdx.x * 16 + ci;
   float sg = 0.f, sgx = 0.f;
   if (c < C) {
-    for (int b = rlane; b < nblocks; b += 4) {
+    int b = rlane;
+    for (; b + 48 < nblocks; b += 64) {
+      sg += (part[(int64_t)b * 2 * C + c] +
+             part[(int64_t)(b + 16) * 2 * C + c]) +
+            (part[(int64_t)(b + 32) * 2 * C + c] +
+             part[(int64_t)(b + 48) * 2 * C + c]);
+      sgx += (part[(int64_t)b * 2 * C + C + c] +
+              part[(int64_t)(b + 16) * 2 * C + C + c]) +
+             (part[(int64_t)(b + 32) * 2 * C + C + c] +
+              part[(int64_t)(b + 48) * 2 * C + C + c]);
+    }
+    for (; b < nblocks; b += 16) {
       sg += part[(int64_t)b * 2 * C + c];
       sgx += part[(int64_t)b * 2 * C + C + c];
     }
@@ -432,10 +479,11 @@ __global__ void bn_bwd_finalize_kernel(
   red[256 + threadIdx.x] = sgx;
   __syncthreads();
   if (rlane != 0 || c >= C) return;
-  sg = red[threadIdx.x] + red[threadIdx.x + 64] + red[threadIdx.x + 128] +
-       red[threadIdx.x + 192];
-  sgx = red[256 + threadIdx.x] + red[256 + threadIdx.x + 64] +
-        red[256 + threadIdx.x + 128] + red[256 + threadIdx.x + 192];
+#pragma unroll
+  for (int t = 1; t < 16; ++t) {
+    sg += red[t * 16 + ci];
+    sgx += red[256 + t * 16 + ci];
+  }
   const float m = mean[c], r = rstd[c], w = gamma[c];
   const float sum_gxh = r * (sgx - m * sg);
   gw[c] = sum_gxh;
@@ -645,7 +693,7 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
     const int64_t rows = x.numel() / C;
     const float inv_cnt = 1.f / (float)rows;
     const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((int)ceil_div(C, 64)),
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((int)ceil_div(C, 16)),
                        dim3(256), 0, cur_stream(), pg.first.data_ptr<float>(),
                        pg.second, gamma.data_ptr<float>(),
                        beta.data_ptr<float>(),
@@ -768,7 +816,7 @@ std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
     const int cpacks = C / V;
     const float inv_cnt = 1.f / (float)total_count;
     // sums is a flat [2C] row => treat as a 1-block partial
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 64)),
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
                        dim3(256), 0, stream, sums.data_ptr<float>(), 1,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        gamma.data_ptr<float>(), gw.data_ptr<float>(),
@@ -839,7 +887,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
                          scale.data_ptr<float>(), shift.data_ptr<float>(),
                          part.data_ptr<float>(), rows, C, rpb, (int)act);
     const float inv_cnt = 1.f / (float)rows;
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 64)),
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
                        dim3(256), 0, stream, part.data_ptr<float>(), rgrid,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        gamma.data_ptr<float>(), gw.data_ptr<float>(),
