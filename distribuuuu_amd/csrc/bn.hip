@@ -316,12 +316,12 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 // ---- bwd stage 1: RAW per-channel sums sum(g), sum(g*x) --------------------
 // g = gy * act'; act' needs only y (relu) or z = scale*x+shift(+res)
 // (silu/sigmoid). No mean/rstd/gamma gathers on the hot path.
-template <typename T, bool HAS_RES>
-__global__ void bn_bwd_reduce_kernel(
+template <typename T, bool HAS_RES, int ACT>
+__global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
     const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, float* __restrict__ part, int64_t rows,
-    int C, int64_t rows_per_block, int act) {
+    int C, int64_t rows_per_block) {
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
@@ -339,7 +339,7 @@ __global__ void bn_bwd_reduce_kernel(
   const P* rp = reinterpret_cast<const P*>(res);
   for (int cp = cp0; cp < cpacks; cp += ncp) {
     float sc[V], sh[V];
-    if (act >= 2) {
+    if (ACT >= 2) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         sc[j] = scale[cp * V + j];
@@ -363,8 +363,8 @@ __global__ void bn_bwd_reduce_kernel(
         for (int u = 0; u < 4; ++u) {
           pg4[u] = gq[u * rstep];
           px4[u] = xq[u * rstep];
-          if (act != 0) py4[u] = yq[u * rstep];
-          if (HAS_RES && act >= 2) pr4[u] = rq[u * rstep];
+          if (ACT != 0) py4[u] = yq[u * rstep];
+          if (HAS_RES && ACT >= 2) pr4[u] = rq[u * rstep];
         }
         gq += 4 * rstep;
         xq += 4 * rstep;
@@ -376,13 +376,13 @@ __global__ void bn_bwd_reduce_kernel(
           for (int j = 0; j < V; ++j) {
             float xv = to_f32(px4[u].v[j]);
             float g = to_f32(pg4[u].v[j]);
-            if (act != 0) {
+            if (ACT != 0) {
               float z = 0.f;
-              if (act >= 2) {
+              if (ACT >= 2) {
                 z = xv * sc[j] + sh[j];
                 if (HAS_RES) z += to_f32(pr4[u].v[j]);
               }
-              g *= act_grad(to_f32(py4[u].v[j]), z, act);
+              g *= act_grad(to_f32(py4[u].v[j]), z, ACT);
             }
             accg[j] += g;
             accgx[j] += g * xv;
@@ -390,8 +390,8 @@ __global__ void bn_bwd_reduce_kernel(
       }
       for (; row < row1; row += nrl) {
         P pg = gq[0], px = xq[0], py, prr;
-        if (act != 0) py = yq[0];
-        if (HAS_RES && act >= 2) prr = rq[0];
+        if (ACT != 0) py = yq[0];
+        if (HAS_RES && ACT >= 2) prr = rq[0];
         gq += rstep;
         xq += rstep;
         yq += rstep;
@@ -400,13 +400,13 @@ __global__ void bn_bwd_reduce_kernel(
         for (int j = 0; j < V; ++j) {
           float xv = to_f32(px.v[j]);
           float g = to_f32(pg.v[j]);
-          if (act != 0) {
+          if (ACT != 0) {
             float z = 0.f;
-            if (act >= 2) {
+            if (ACT >= 2) {
               z = xv * sc[j] + sh[j];
               if (HAS_RES) z += to_f32(prr.v[j]);
             }
-            g *= act_grad(to_f32(py.v[j]), z, act);
+            g *= act_grad(to_f32(py.v[j]), z, ACT);
           }
           accg[j] += g;
           accgx[j] += g * xv;
@@ -613,11 +613,45 @@ __global__ void bn_bwd_dx_kernel(
 }
 
 int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
-  // 512 stage-1 blocks: BW-saturating with the MLP unroll while keeping the
-  // stage-2 partial reduction short
-  int64_t rpb = std::max<int64_t>(ceil_div(rows, 512), rows_per_iter);
+  // stage-1 block budget: BW-saturating with the MLP unroll while keeping the
+  // stage-2 partial reduction short (env override for tuning probes)
+  const char* s = getenv("DISTRIBUUUU_BN_S1GRID");  // re-read: probe sweeps
+  const int cap = s ? atoi(s) : 512;
+  int64_t rpb = std::max<int64_t>(ceil_div(rows, cap), rows_per_iter);
   return ceil_div(rpb, rows_per_iter) * rows_per_iter;
 }
+
+// act is runtime but templated into the kernel: a unified body kept the
+// silu/sigmoid scale/shift register arrays live on every path and spilled
+// 90-186 VGPRs (scratch traffic capped the reduce at ~1.5 TB/s).
+template <typename scalar_t>
+void launch_bwd_reduce(const scalar_t* gy, const scalar_t* x,
+                       const scalar_t* y, const scalar_t* resp,
+                       const float* scale, const float* shift, float* part,
+                       int64_t rows, int C, int64_t rpb, int act, int rgrid,
+                       hipStream_t stream) {
+#define BR_CASE(HR, A)                                                      \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, HR, A>), dim3(rgrid),  \
+                     dim3(256), 0, stream, gy, x, y, resp, scale, shift,    \
+                     part, rows, C, rpb)
+  if (resp != nullptr) {
+    switch (act) {
+      case 0: BR_CASE(true, 0); break;
+      case 1: BR_CASE(true, 1); break;
+      case 2: BR_CASE(true, 2); break;
+      default: BR_CASE(true, 3); break;
+    }
+  } else {
+    switch (act) {
+      case 0: BR_CASE(false, 0); break;
+      case 1: BR_CASE(false, 1); break;
+      case 2: BR_CASE(false, 2); break;
+      default: BR_CASE(false, 3); break;
+    }
+  }
+#undef BR_CASE
+}
+
 
 template <typename scalar_t, int V, bool HR>
 void launch_dx(const at::Tensor& gy, const at::Tensor& x, const at::Tensor& y,
@@ -769,20 +803,11 @@ at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
     auto part = at::empty({rgrid, 2 * C}, fopts);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
-    if (has_res)
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true>), dim3(rgrid),
-                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)y.data_ptr(), resp,
-                         scale.data_ptr<float>(), shift.data_ptr<float>(),
-                         part.data_ptr<float>(), rows, C, rpb, (int)act);
-    else
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
-                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)y.data_ptr(), resp,
-                         scale.data_ptr<float>(), shift.data_ptr<float>(),
-                         part.data_ptr<float>(), rows, C, rpb, (int)act);
+    launch_bwd_reduce<scalar_t>(
+        (const scalar_t*)gy.data_ptr(), (const scalar_t*)x.data_ptr(),
+        (const scalar_t*)y.data_ptr(), resp, scale.data_ptr<float>(),
+        shift.data_ptr<float>(), part.data_ptr<float>(), rows, C, rpb,
+        (int)act, rgrid, stream);
     both = at::empty({2 * C}, fopts);
     hipLaunchKernelGGL(reduce_partials_kernel, dim3((int)ceil_div(2 * C, 64)),
                        dim3(256), 0, stream, part.data_ptr<float>(),
@@ -871,20 +896,11 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
     auto part = at::empty({rgrid, 2 * C}, fopts);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
-    if (has_res)
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true>), dim3(rgrid),
-                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)y.data_ptr(), resp,
-                         scale.data_ptr<float>(), shift.data_ptr<float>(),
-                         part.data_ptr<float>(), rows, C, rpb, (int)act);
-    else
-      hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, false>), dim3(rgrid),
-                         dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                         (const scalar_t*)x.data_ptr(),
-                         (const scalar_t*)y.data_ptr(), resp,
-                         scale.data_ptr<float>(), shift.data_ptr<float>(),
-                         part.data_ptr<float>(), rows, C, rpb, (int)act);
+    launch_bwd_reduce<scalar_t>(
+        (const scalar_t*)gy.data_ptr(), (const scalar_t*)x.data_ptr(),
+        (const scalar_t*)y.data_ptr(), resp, scale.data_ptr<float>(),
+        shift.data_ptr<float>(), part.data_ptr<float>(), rows, C, rpb,
+        (int)act, rgrid, stream);
     const float inv_cnt = 1.f / (float)rows;
     hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
                        dim3(256), 0, stream, part.data_ptr<float>(), rgrid,
