@@ -616,6 +616,8 @@ int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
   // stage-1 block budget: BW-saturating with the MLP unroll while keeping the
   // stage-2 partial reduction short (env override for tuning probes)
   const char* s = getenv("DISTRIBUUUU_BN_S1GRID");  // re-read: probe sweeps
+  // measured (tools/probes/bnbwd_ab.py): 512 beats 1024/2048/4096 across the
+  // full ResNet-50 shape mix once the reduce kernel stopped spilling
   const int cap = s ? atoi(s) : 512;
   int64_t rpb = std::max<int64_t>(ceil_div(rows, cap), rows_per_iter);
   return ceil_div(rpb, rows_per_iter) * rows_per_iter;

@@ -40,7 +40,15 @@ struct WgradParams {
   int Ho, Wo;
   int M, RSC;
   int ktiles, ntiles, chunks;
+  // magic-multiply division: q = (m * magic) >> 47 == m / d exactly for
+  // m*d < 2^47 (here m < 2^22, d < 2^15). Runtime u32 division in the
+  // per-k-step staging path costs ~30 VALU cycles each.
+  unsigned long long magicHoWo, magicWo;
 };
+
+DEV_INLINE int magic_div(int m, unsigned long long magic) {
+  return (int)(((unsigned long long)(unsigned)m * magic) >> 47);
+}
 
 // LDS byte offset of element (m, col) in a [64 m][64 col] tile stored as
 // [m/32][col/16] subtiles of [32][16].
@@ -75,6 +83,24 @@ DEV_INLINE bf16x8w tr_read_frag(unsigned lds_base, int subtile_base,
       : "memory");
   __builtin_amdgcn_sched_barrier(0);  // rule 18: fence MFMA below the wait
   return u.v;
+}
+
+// issue-only variant: the reads are NOT awaited here — callers batch many
+// issues and place one counted s_waitcnt before consumption. (The original
+// wait-per-fragment tr_read_frag serialized 8 full LDS latencies per k-step
+// against idle MFMAs — measured ~12% MFMA utilization.)
+union TrFrag {
+  uint2 h[2];
+  bf16x8w v;
+};
+
+DEV_INLINE void tr_read_issue(unsigned a0, unsigned a1, TrFrag& f) {
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3"
+      : "=&v"(f.h[0]), "=&v"(f.h[1])
+      : "v"(a0), "v"(a1)
+      : "memory");
 }
 
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
@@ -124,9 +150,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
       const int m = m0 + ks * WBK + sml + half * 32;
       const bool m_ok = m < p.M;
       const int mm = m_ok ? m : 0;
-      const int n = mm / HoWo;
+      const int n = magic_div(mm, p.magicHoWo);
       const int rem = mm - n * HoWo;
-      const int ho = rem / p.Wo, wo = rem - (rem / p.Wo) * p.Wo;
+      const int ho = magic_div(rem, p.magicWo);
+      const int wo = rem - ho * p.Wo;
       if (m_ok && k_ok) {
         regA[half] = *reinterpret_cast<const uint4*>(
             p.gy + ((int64_t)mm * p.Kt) + g * p.Kg + kcol);
@@ -161,24 +188,40 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   for (int ks = 0; ks < ksteps; ++ks) {
     const unsigned abase = smem_base + (unsigned)(cur * 2 * TILE_BYTES);
     const unsigned bbase = abase + TILE_BYTES;
+    // issue ALL 16 tr reads for both 32-deep m sub-steps, then consume with
+    // counted waits: mc=0's MFMAs run while mc=1's reads are still in flight
+    TrFrag fa[2][2], fb[2][2];
 #pragma unroll
-    for (int mc = 0; mc < 2; ++mc) {  // two 32-deep m sub-steps
-      bf16x8w afrag[2], bfrag[2];
+    for (int mc = 0; mc < 2; ++mc) {
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
-        afrag[mi] =
-            tr_read_frag(abase, (mc * 4 + (wm * 2 + mi)) * SUBT, lane);
+      for (int mi = 0; mi < 2; ++mi) {
+        const int sb = (mc * 4 + (wm * 2 + mi)) * SUBT;
+        tr_read_issue(abase + tr_addr(sb, lane, 0),
+                      abase + tr_addr(sb, lane, 1), fa[mc][mi]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int sb = (mc * 4 + (wn * 2 + ni)) * SUBT;
+        tr_read_issue(bbase + tr_addr(sb, lane, 0),
+                      bbase + tr_addr(sb, lane, 1), fb[mc][ni]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");  // mc=0's 8 reads done
+    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni)
-        bfrag[ni] =
-            tr_read_frag(bbase, (mc * 4 + (wn * 2 + ni)) * SUBT, lane);
+        accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            fa[0][mi].v, fb[0][ni].v, accv[mi][ni], 0, 0, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+    for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
-          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[mi], bfrag[ni], accv[mi][ni], 0, 0, 0);
-    }
+      for (int ni = 0; ni < 2; ++ni)
+        accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            fa[1][mi].v, fb[1][ni].v, accv[mi][ni], 0, 0, 0);
     __syncthreads();
     if (ks + 1 < ksteps) {
       stage_write(cur ^ 1);
@@ -383,6 +426,189 @@ __global__ __launch_bounds__(256) void conv_wgrad_ring_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 128x128x64 ring wgrad: the 64x64 tile runs out of math to cover its
+// staging (8 MFMAs per wave per k-step vs 16 tr-reads + 8 glds + 2 barriers
+// — measured ~330 TF ceiling). This tile gives each of the 4 waves a 64x64
+// output (32 MFMAs per k-step vs 16 tr-read pairs), staged by
+// global_load_lds into packed [32 m][16 col] subtiles in a 3-slot 96 KB
+// dynamic-LDS ring with counted vmcnt waits (the conv2.hip idiom).
+// Host-guaranteed: Kg%128==0, M%64==0, Cg%8==0, image pre-padded when ph>0;
+// rsc column tails clamp their load to col 0 and are dropped in the epilogue.
+// ---------------------------------------------------------------------------
+constexpr int R128_BM = 128;            // k rows per tile
+constexpr int R128_BN = 128;            // rsc cols per tile
+constexpr int R128_TILE = 16 * RSUBT;   // one operand tile (16 KB)
+constexpr int R128_SLOT = 2 * R128_TILE;
+
+struct Wgrad128Params {
+  const __hip_bfloat16* x;   // padded [N, Hp, Wp, Ct] when ph/pw > 0
+  const __hip_bfloat16* gy;  // [M, Kt]
+  float* acc;                // [Kt, R*S*Cg] zeroed
+  int Hp, Wp, Ct, Kt;
+  int R, S, Cg, Kg;
+  int sh, sw, dh, dw;
+  int Ho, Wo;
+  int M, RSC;
+  int ktiles, ntiles, csteps;
+  unsigned long long magicHoWo, magicWo;
+};
+
+__global__ __launch_bounds__(256) void conv_wgrad_ring128_kernel(
+    Wgrad128Params p) {
+  extern __shared__ __align__(16) char smem[];  // 3 * R128_SLOT
+  const int g = blockIdx.z;
+  const int ktile = blockIdx.x % p.ktiles;
+  const int ntile = blockIdx.x / p.ktiles;
+  const int chunk = blockIdx.y;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int il = lane & 15, kq = lane >> 4;
+
+  // staging: subtile s = it*4 + wid (it<4) per operand; within a subtile
+  // lane l fills bytes l*16 = element (m = l>>1, col8 = (l&1)*8) — the glds
+  // lane-linear image IS the packed [32 m][16 col] subtile layout.
+  int s_mloc[4], s_col[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int sub = it * 4 + wid;
+    s_mloc[it] = (sub >> 3) * 32 + (lane >> 1);
+    s_col[it] = (sub & 7) * 16 + (lane & 1) * 8;
+  }
+  const int SCg = p.S * p.Cg;
+  int b_r[4], b_s[4], b_c[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    int rsc = ntile * R128_BN + s_col[it];
+    if (rsc >= p.RSC) rsc = 0;  // clamped tail: its output col is discarded
+    b_r[it] = rsc / SCg;
+    const int rem = rsc - b_r[it] * SCg;
+    b_s[it] = rem / p.Cg;
+    b_c[it] = rem - b_s[it] * p.Cg;
+  }
+  const int m0 = chunk * (p.csteps * WBK);
+  const int HoWo = p.Ho * p.Wo;
+  const int nt = min(p.csteps, (int)((p.M - m0 + WBK - 1) / WBK));
+
+  auto stage = [&](int buf, int ks) {
+    char* base = smem + buf * R128_SLOT + wid * 1024;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int m = m0 + ks * WBK + s_mloc[it];  // < M (M%64==0)
+      const __hip_bfloat16* asrc =
+          p.gy + (int64_t)m * p.Kt + g * p.Kg + ktile * R128_BM + s_col[it];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)asrc,
+          (__attribute__((address_space(3))) uint32_t*)(base + it * 4096), 16,
+          0, 0);
+      const int n = magic_div(m, p.magicHoWo);
+      const int rem = m - n * HoWo;
+      const int ho = magic_div(rem, p.magicWo);
+      const int wo = rem - ho * p.Wo;
+      const int h = ho * p.sh + b_r[it] * p.dh;
+      const int w_ = wo * p.sw + b_s[it] * p.dw;
+      const __hip_bfloat16* bsrc =
+          p.x + (((int64_t)n * p.Hp + h) * p.Wp + w_) * p.Ct + g * p.Cg +
+          b_c[it];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)bsrc,
+          (__attribute__((address_space(3))) uint32_t*)(base + R128_TILE +
+                                                        it * 4096),
+          16, 0, 0);
+    }
+  };
+
+  f32x4w accv[4][4] = {};
+
+  stage(0, 0);
+  if (nt > 1) stage(1, 1);
+  if (nt > 2) stage(2, 2);
+
+  int bufsel = 0;
+  for (int t = 0; t < nt; ++t) {
+    const int infl = (nt - 1 - t) >= 2 ? 2 : (nt - 1 - t);
+    if (infl == 2) {
+      RWAITVM(16);
+    } else if (infl == 1) {
+      RWAITVM(8);
+    } else {
+      RWAITVM(0);
+    }
+    __builtin_amdgcn_s_barrier();
+    const char* ta = smem + bufsel * R128_SLOT;
+    const char* tb = ta + R128_TILE;
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {
+      // issue all 16 tr-read pairs of this m-half, then consume under one
+      // counted wait so the MFMAs overlap the remaining reads
+      TrFrag fa[4], fb[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int sb = (mc * 8 + wm * 4 + mi) * RSUBT;
+        const unsigned a0 = (unsigned)(unsigned long long)(ta + sb) +
+                            (unsigned)((kq * 8 + ((lane & 15) >> 2)) * 32 +
+                                       (lane & 3) * 8);
+        tr_read_issue(a0, a0 + 128, fa[mi]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int sb = (mc * 8 + wn * 4 + ni) * RSUBT;
+        const unsigned b0 = (unsigned)(unsigned long long)(tb + sb) +
+                            (unsigned)((kq * 8 + ((lane & 15) >> 2)) * 32 +
+                                       (lane & 3) * 8);
+        tr_read_issue(b0, b0 + 128, fb[ni]);
+      }
+      // LDS FIFO is in-order: 16 reads issued as fa0..3, fb0..3 (2 each);
+      // lgkmcnt(6) => all fa + fb[0] complete, then drop 2 per ni step
+      asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        if (ni == 1) {
+          asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+        if (ni == 2) {
+          asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+        if (ni == 3) {
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[mi].v, fb[ni].v, accv[mi][ni], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + 3 < nt) stage(bufsel, t + 3);
+    bufsel = bufsel == 2 ? 0 : bufsel + 1;
+  }
+
+  // epilogue: fp32 atomic accumulate (D: col=lane&15, row=(lane>>4)*4+rr)
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int k = ktile * R128_BM + wm * 64 + mi * 16 + kq * 4 + rr;
+      const int64_t rowbase = (int64_t)(g * p.Kg + k) * p.RSC;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int col = ntile * R128_BN + wn * 64 + ni * 16 + il;
+        if (col < p.RSC) atomicAdd(&p.acc[rowbase + col], accv[mi][ni][rr]);
+      }
+    }
+  }
+}
+
 template <typename T>
 __global__ void wg_pad_image_kernel(const T* __restrict__ x, T* __restrict__ y,
                                     int N, int H, int W, int C, int Hp, int Wp,
@@ -476,6 +702,74 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
     return gw2;
   }
 
+  // 128x128 ring kernel where it measured faster (tools/probes/convmap.py):
+  // 1x1 convolutions (no pad pass, B columns are plain channels) whose
+  // launch has enough tiles+chunks to fill the 256 CUs. 3x3 shapes keep the
+  // 64x64 kernel: the pad pass plus tap-spanning B columns cost more than
+  // the bigger tile saves.
+  const int64_t blocks64 = (Kg / 128) *
+                           (((int64_t)R * S * Cg + 127) / 128) *
+                           ((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK));
+  if (R == 1 && S == 1 && Kg % 128 == 0 && Cg >= 128 && M64 % 64 == 0 &&
+      Cg % 8 == 0 && blocks64 * groups >= 320) {
+    at::Tensor xin = x;
+    int Hp = H, Wp = W;
+    if (ph > 0 || pw > 0) {
+      Hp = H + 2 * ph;
+      Wp = W + 2 * pw;
+      auto xp = at::empty({(int64_t)N * Hp * Wp * Ct}, x.options());
+      int64_t total = xp.numel();
+      hipLaunchKernelGGL((wg_pad_image_kernel<__hip_bfloat16>),
+                         dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         (__hip_bfloat16*)xp.data_ptr(), N, H, W, Ct, Hp, Wp,
+                         ph, pw);
+      xin = xp;
+    }
+    Wgrad128Params q;
+    q.x = (const __hip_bfloat16*)xin.data_ptr();
+    q.gy = (const __hip_bfloat16*)gy.data_ptr();
+    q.Hp = Hp; q.Wp = Wp; q.Ct = Ct; q.Kt = Kt;
+    q.R = R; q.S = S; q.Cg = Cg; q.Kg = Kg;
+    q.sh = sh; q.sw = sw; q.dh = dh; q.dw = dw;
+    q.Ho = Ho; q.Wo = Wo;
+    q.M = (int)M64;
+    q.RSC = R * S * Cg;
+    q.ktiles = Kg / R128_BM;
+    q.ntiles = (q.RSC + R128_BN - 1) / R128_BN;
+    q.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
+    q.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
+    // shrink the split-m chunk until the launch fills the 256 CUs
+    int csteps = CHUNK_STEPS;
+    int chunks = (int)((M64 + (int64_t)csteps * WBK - 1) / (csteps * WBK));
+    while (csteps > 8 && (int64_t)q.ktiles * q.ntiles * chunks * groups < 512) {
+      csteps /= 2;
+      chunks = (int)((M64 + (int64_t)csteps * WBK - 1) / (csteps * WBK));
+    }
+    q.csteps = csteps;
+    auto accbuf = at::empty({(int64_t)Kt, q.RSC},
+                            x.options().dtype(at::kFloat));
+    accbuf.zero_();
+    q.acc = accbuf.data_ptr<float>();
+    static bool attr_done = false;
+    if (!attr_done) {
+      hipFuncSetAttribute((const void*)conv_wgrad_ring128_kernel,
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          3 * R128_SLOT);
+      attr_done = true;
+    }
+    dim3 grid(q.ktiles * q.ntiles, chunks, groups);
+    hipLaunchKernelGGL(conv_wgrad_ring128_kernel, grid, dim3(256),
+                       3 * R128_SLOT, cur_stream(), q);
+    auto gw = at::empty({Kt, Cg, (int64_t)R, (int64_t)S},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    const int64_t total = (int64_t)Kt * q.RSC;
+    hipLaunchKernelGGL(cast_acc_kernel, dim3(grid_1d(total, 256)), dim3(256),
+                       0, cur_stream(), q.acc, (__hip_bfloat16*)gw.data_ptr(),
+                       total);
+    return gw;
+  }
+
   WgradParams p;
   p.x = (const __hip_bfloat16*)x.data_ptr();
   p.gy = (const __hip_bfloat16*)gy.data_ptr();
@@ -488,6 +782,8 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
   p.ktiles = (Kg + WBM - 1) / WBM;
   p.ntiles = (p.RSC + WBN - 1) / WBN;
   p.chunks = (p.M + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK);
+  p.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
+  p.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
 
   auto accbuf = at::empty({(int64_t)Kt, p.RSC}, x.options().dtype(at::kFloat));
   accbuf.zero_();
