@@ -39,6 +39,7 @@ struct ConvParams {
   int Ho, Wo;               // logical output (M = N*Ho*Wo)
   int HoA, WoA;             // allocated output strides (>= Ho,Wo)
   int osh, osw;             // output position multipliers (dgrad-s2 scatter)
+  int oh0, ow0;             // output position base offsets (parity classes)
   int M, nspan, ksteps;     // nspan = ceil(S*Cg/BK), ksteps = R*nspan
   int tiles_m;              // for XCD swizzle
 };
@@ -185,8 +186,8 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
       const int n = m / HoWo;
       const int rem = m - n * HoWo;
       const int64_t obase =
-          (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh) * p.WoA +
-           (rem % p.Wo) * p.osw) * p.Kt + g * p.Kg;
+          (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh + p.oh0) * p.WoA +
+           (rem % p.Wo) * p.osw + p.ow0) * p.Kt + g * p.Kg;
       const int k0 = tile_n * BN + wn * 64 + ec;
       union {
         __hip_bfloat16 b[16];
@@ -216,7 +217,8 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
-                           int64_t osh = 1, int64_t osw = 1) {
+                           int64_t osh = 1, int64_t osw = 1, int64_t oh0 = 0,
+                           int64_t ow0 = 0) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16 only");
   check_nhwc(x, "x");
@@ -236,6 +238,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   p.Ho = Ho; p.Wo = Wo;
   p.HoA = y.size(2); p.WoA = y.size(3);
   p.osh = osh; p.osw = osw;
+  p.oh0 = oh0; p.ow0 = ow0;
   p.M = N * Ho * Wo;
   p.nspan = (S * Cg + BK - 1) / BK;
   p.ksteps = R * p.nspan;
@@ -416,6 +419,29 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
     conv2d_fwd_into(gy, wt, gx1, gy.size(2), gy.size(3), 1, 1, 0, 0, 1, 1,
                     groups, sh, sw);
     return gx1;
+  }
+  if (R == 3 && S == 3 && sh == 2 && sw == 2 && dh == 1 && dw == 1 &&
+      ph == 1 && pw == 1) {
+    // parity decomposition: for stride-2 with pad 1, each output-parity
+    // class (h%2, w%2) of gx receives only the taps of matching parity —
+    // four small {1,2}x{1,2}-tap convs over gy scattered with stride 2
+    // instead of a 3x3 conv over 2x-dilated (3/4-zero) gy. Derivation: for
+    // h = 2u + h0, ho = u + e with e = (h0 + 1 - r)/2, so the e-ordered tap
+    // weights are flipT(w) rows (1 - h0)::2.
+    auto gx = at::empty({N, Ct, H, W}, gy.options().memory_format(
+                                           at::MemoryFormat::ChannelsLast));
+    for (int h0 = 0; h0 < 2; ++h0) {
+      const int64_t hu = (H - h0 + 1) >> 1;
+      auto wr = (h0 == 0) ? wt.slice(2, 1, 2) : wt.slice(2, 0, 3, 2);
+      for (int w0 = 0; w0 < 2; ++w0) {
+        const int64_t wu = (W - w0 + 1) >> 1;
+        auto wsub = ((w0 == 0) ? wr.slice(3, 1, 2) : wr.slice(3, 0, 3, 2))
+                        .contiguous(at::MemoryFormat::ChannelsLast);
+        conv2d_fwd_into(gy, wsub, gx, hu, wu, 1, 1, 0, 0, 1, 1, groups, 2, 2,
+                        h0, w0);
+      }
+    }
+    return gx;
   }
   const int Kt_ = gy.size(1);
   if (sh == 1 && sw == 1 && groups == 1 && dh * (R - 1) == 2 * ph &&
