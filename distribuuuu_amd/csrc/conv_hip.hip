@@ -209,6 +209,156 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Small 1x1 GEMM: y[M, Kt] = x[M, C] @ w[Kt, C]^T for the shallow shapes
+// (C <= 256, Kt <= 512) where the 128x128x32 tile kernel spends most of its
+// time on pipeline fill/drain (2-8 k-steps) and half-empty B tiles (K = 64).
+// Both MFMA operands load DIRECT from global: w (<= 256 KB) is L2-resident
+// across the whole launch, and each x row is L1-reused by the wave's four
+// B fragments. Each wave owns 64 m-rows and loops over 64-wide k chunks.
+// ---------------------------------------------------------------------------
+struct SmallGemmParams {
+  const __hip_bfloat16* x;
+  const __hip_bfloat16* w;
+  __hip_bfloat16* y;
+  int C, Kt;
+  int64_t M;
+};
+
+__global__ __launch_bounds__(256) void conv_gemm_small_kernel(
+    SmallGemmParams p) {
+  __shared__ float slab4[4][16 * 68];
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int il = lane & 15, kq = lane >> 4;
+  const int64_t mbase = (int64_t)blockIdx.x * 256 + wid * 64;
+  const __hip_bfloat16* xrow[4];
+  bool m_ok[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    const int64_t m = mbase + mi * 16 + il;
+    m_ok[mi] = m < p.M;
+    xrow[mi] = p.x + (m_ok[mi] ? m : 0) * p.C + kq * 8;
+  }
+  float* slab = slab4[wid];
+  const int er = lane >> 2, ec = (lane & 3) << 4;
+  for (int k0 = 0; k0 < p.Kt; k0 += 64) {
+    const __hip_bfloat16* wrow[4];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      wrow[ni] = p.w + (int64_t)(k0 + ni * 16 + il) * p.C + kq * 8;
+    f32x4 acc[4][4] = {};
+    for (int c0 = 0; c0 < p.C; c0 += 32) {
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        afrag[mi] = m_ok[mi] ? *reinterpret_cast<const bf16x8*>(xrow[mi] + c0)
+                             : bf16x8{};
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bfrag[ni] = *reinterpret_cast<const bf16x8*>(wrow[ni] + c0);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    // per-wave LDS-restage epilogue (16-wide bf16 stores; T21)
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr)
+          slab[(kq * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+      __builtin_amdgcn_wave_barrier();
+      const int64_t m = mbase + mi * 16 + er;
+      if (m < p.M) {
+        union {
+          __hip_bfloat16 b[16];
+          uint4 q[2];
+        } u;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+        __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
+        *reinterpret_cast<uint4*>(yp) = u.q[0];
+        *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
+
+// C <= 64 variant: the whole per-wave A stripe (64 rows x C) is register-
+// cached across k chunks, so x is read exactly once regardless of Kt (the
+// generic kernel re-reads A per 64-wide k chunk, which loses for Kt > 64).
+template <int CS>  // C / 32
+__global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
+    SmallGemmParams p) {
+  __shared__ float slab4[4][16 * 68];
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int il = lane & 15, kq = lane >> 4;
+  const int64_t mbase = (int64_t)blockIdx.x * 256 + wid * 64;
+  float* slab = slab4[wid];
+  const int er = lane >> 2, ec = (lane & 3) << 4;
+  bf16x8 afr[CS][4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    const int64_t m = mbase + mi * 16 + il;
+    const __hip_bfloat16* xr = p.x + (m < p.M ? m : 0) * p.C + kq * 8;
+#pragma unroll
+    for (int cs = 0; cs < CS; ++cs)
+      afr[cs][mi] = m < p.M ? *reinterpret_cast<const bf16x8*>(xr + cs * 32)
+                            : bf16x8{};
+  }
+  for (int k0 = 0; k0 < p.Kt; k0 += 64) {
+    const __hip_bfloat16* wrow[4];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      wrow[ni] = p.w + (int64_t)(k0 + ni * 16 + il) * p.C + kq * 8;
+    f32x4 acc[4][4] = {};
+#pragma unroll
+    for (int cs = 0; cs < CS; ++cs) {
+      bf16x8 bfrag[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        bfrag[ni] = *reinterpret_cast<const bf16x8*>(wrow[ni] + cs * 32);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[cs][mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr)
+          slab[(kq * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+      __builtin_amdgcn_wave_barrier();
+      const int64_t m = mbase + mi * 16 + er;
+      if (m < p.M) {
+        union {
+          __hip_bfloat16 b[16];
+          uint4 q[2];
+        } u;
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+        __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
+        *reinterpret_cast<uint4*>(yp) = u.q[0];
+        *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -267,6 +417,33 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   if (!v2_off && groups == 1 && Kt >= 192 && R * S * C_ >= 576 &&
       x.scalar_type() == at::kBFloat16)
     return conv2d_fwd_v2(x, w, sh, sw, ph, pw, dh, dw);
+  static const bool small_off = []() {
+    const char* e = getenv("DISTRIBUUUU_CONV_SMALL");
+    return e && e[0] == '0';
+  }();
+  if (!small_off && groups == 1 && R == 1 && S == 1 && sh == 1 && sw == 1 &&
+      C_ % 32 == 0 && Kt % 64 == 0 && x.scalar_type() == at::kBFloat16 &&
+      ((C_ <= 64 && Kt <= 512) || (C_ <= 256 && Kt <= 64))) {
+    const int64_t M = (int64_t)N * H * W;
+    auto y = at::empty({N, Kt, H, W},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    SmallGemmParams sp;
+    sp.x = (const __hip_bfloat16*)x.data_ptr();
+    sp.w = (const __hip_bfloat16*)w.data_ptr();
+    sp.y = (__hip_bfloat16*)y.data_ptr();
+    sp.C = C_; sp.Kt = Kt; sp.M = M;
+    const dim3 sg((int)((M + 255) / 256));
+    if (C_ == 64)
+      hipLaunchKernelGGL(conv_gemm_smallc_kernel<2>, sg, dim3(256), 0,
+                         cur_stream(), sp);
+    else if (C_ == 32)
+      hipLaunchKernelGGL(conv_gemm_smallc_kernel<1>, sg, dim3(256), 0,
+                         cur_stream(), sp);
+    else
+      hipLaunchKernelGGL(conv_gemm_small_kernel, sg, dim3(256), 0,
+                         cur_stream(), sp);
+    return y;
+  }
   const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
   auto y = at::empty({N, Kt, Ho, Wo},
