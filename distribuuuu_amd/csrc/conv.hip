@@ -358,6 +358,113 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// K=64 small-C patch conv (stem 7x7 s2 C=8, 3x3 64->64): one 64-col B tile
+// means half of the 128-wide tile kernel idles. Here both operands load
+// direct from global with full per-lane patch predication: each lane's 16 B
+// A fragment is one (r, s, c-pack) patch element (magic-div span
+// decomposition), B is the span-padded [64, RSCp] weight (L2-resident).
+// Block = 4 waves x 64 m-rows, each wave owns all 64 output channels.
+// ---------------------------------------------------------------------------
+struct SmallConvParams {
+  const __hip_bfloat16* x;  // [N,H,W,C]
+  const __hip_bfloat16* w;  // [64, RSCp] span-padded flat
+  __hip_bfloat16* y;        // [N,Ho,Wo,64]
+  int C, H, W;
+  int R, S, SC;
+  int sh, sw, ph, pw;
+  int Ho, Wo;
+  int RSC, RSCp;
+  int64_t M;
+  unsigned long long magicHoWo, magicWo, magicSC, magicC;
+};
+
+DEV_INLINE int magic_div2(int m, unsigned long long magic) {
+  return (int)(((unsigned long long)(unsigned)m * magic) >> 47);
+}
+
+__global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
+  __shared__ float slab4[4][16 * 68];
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int il = lane & 15, kq = lane >> 4;
+  const int64_t mbase = (int64_t)blockIdx.x * 256 + wid * 64;
+  int pn[4], prow[4], pcol[4];
+  bool mok[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    const int64_t m = mbase + mi * 16 + il;
+    mok[mi] = m < p.M;
+    const int mm = (int)(mok[mi] ? m : 0);
+    const int n = magic_div2(mm, p.magicHoWo);
+    const int rem = mm - n * (p.Ho * p.Wo);
+    const int ho = magic_div2(rem, p.magicWo);
+    const int wo = rem - ho * p.Wo;
+    pn[mi] = n;
+    prow[mi] = ho * p.sh - p.ph;
+    pcol[mi] = wo * p.sw - p.pw;
+  }
+  const __hip_bfloat16* wr[4];
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni)
+    wr[ni] = p.w + (int64_t)(ni * 16 + il) * p.RSCp + kq * 8;
+  const int ksteps = p.RSCp / 32;
+  f32x4 acc[4][4] = {};
+  for (int t = 0; t < ksteps; ++t) {
+    const int e = t * 32 + kq * 8;
+    const int r = magic_div2(e, p.magicSC);
+    const int rem2 = e - r * p.SC;
+    const int sidx = magic_div2(rem2, p.magicC);
+    const int c = rem2 - sidx * p.C;
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int h = prow[mi] + r;
+      const int w_ = pcol[mi] + sidx;
+      const bool ok = mok[mi] && e < p.RSC && h >= 0 && h < p.H && w_ >= 0 &&
+                      w_ < p.W;
+      afrag[mi] =
+          ok ? *reinterpret_cast<const bf16x8*>(
+                   p.x + (((int64_t)pn[mi] * p.H + h) * p.W + w_) * p.C + c)
+             : bf16x8{};
+    }
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      bfrag[ni] = *reinterpret_cast<const bf16x8*>(wr[ni] + t * 32);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+  }
+  float* slab = slab4[wid];
+  const int er = lane >> 2, ec = (lane & 3) << 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr)
+        slab[(kq * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t m = mbase + mi * 16 + er;
+    if (m < p.M) {
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } u;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      __hip_bfloat16* yp = p.y + m * 64 + ec;
+      *reinterpret_cast<uint4*>(yp) = u.q[0];
+      *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -445,6 +552,38 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   }
   const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  if (!small_off && groups == 1 && Kt == 64 && C_ <= 64 && C_ % 8 == 0 &&
+      dh == 1 && dw == 1 && R * S > 1 && R * S * C_ <= 1024 &&
+      x.scalar_type() == at::kBFloat16) {
+    const int RSC = R * S * C_;
+    const int RSCp = (RSC + 31) / 32 * 32;
+    at::Tensor wsp = w;
+    if (RSCp != RSC) {
+      wsp = at::zeros({(int64_t)Kt, RSCp}, w.options());
+      wsp.narrow(1, 0, RSC).copy_(
+          w.permute({0, 2, 3, 1}).reshape({(int64_t)Kt, RSC}));
+    }
+    auto y = at::empty({N, Kt, Ho, Wo},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    SmallConvParams sp;
+    sp.x = (const __hip_bfloat16*)x.data_ptr();
+    sp.w = (const __hip_bfloat16*)wsp.data_ptr();
+    sp.y = (__hip_bfloat16*)y.data_ptr();
+    sp.C = C_; sp.H = H; sp.W = W;
+    sp.R = R; sp.S = S; sp.SC = S * C_;
+    sp.sh = sh; sp.sw = sw; sp.ph = ph; sp.pw = pw;
+    sp.Ho = Ho; sp.Wo = Wo;
+    sp.RSC = RSC; sp.RSCp = RSCp;
+    sp.M = (int64_t)N * Ho * Wo;
+    sp.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
+    sp.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
+    sp.magicSC = ((1ULL << 47) / (unsigned long long)(S * C_)) + 1;
+    sp.magicC = ((1ULL << 47) / (unsigned long long)C_) + 1;
+    hipLaunchKernelGGL(conv_smallk_kernel,
+                       dim3((int)((sp.M + 255) / 256)), dim3(256), 0,
+                       cur_stream(), sp);
+    return y;
+  }
   auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups);
