@@ -409,35 +409,77 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
 #pragma unroll
   for (int ni = 0; ni < 4; ++ni)
     wr[ni] = p.w + (int64_t)(ni * 16 + il) * p.RSCp + kq * 8;
-  const int ksteps = p.RSCp / 32;
   f32x4 acc[4][4] = {};
-  for (int t = 0; t < ksteps; ++t) {
-    const int e = t * 32 + kq * 8;
-    const int r = magic_div2(e, p.magicSC);
-    const int rem2 = e - r * p.SC;
-    const int sidx = magic_div2(rem2, p.magicC);
-    const int c = rem2 - sidx * p.C;
-    bf16x8 afrag[4], bfrag[4];
+  if (p.C % 32 == 0) {
+    // tap-ordered loop (C a multiple of 32): the per-tap bounds check and
+    // base address are computed once per (r, s) instead of per k-step —
+    // the k-step-ordered form below was VALU-bound on its per-step
+    // magic-div + predication chain (~80 VALU per 16 MFMAs).
+    const int csteps = p.C / 32;
+    for (int r = 0; r < p.R; ++r) {
+#pragma unroll 1
+      for (int sidx = 0; sidx < p.S; ++sidx) {
+        const __hip_bfloat16* ab[4];
+        bool ok[4];
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
-      const int h = prow[mi] + r;
-      const int w_ = pcol[mi] + sidx;
-      const bool ok = mok[mi] && e < p.RSC && h >= 0 && h < p.H && w_ >= 0 &&
-                      w_ < p.W;
-      afrag[mi] =
-          ok ? *reinterpret_cast<const bf16x8*>(
-                   p.x + (((int64_t)pn[mi] * p.H + h) * p.W + w_) * p.C + c)
-             : bf16x8{};
+        for (int mi = 0; mi < 4; ++mi) {
+          const int h = prow[mi] + r;
+          const int w_ = pcol[mi] + sidx;
+          ok[mi] = mok[mi] && h >= 0 && h < p.H && w_ >= 0 && w_ < p.W;
+          ab[mi] = p.x +
+                   (((int64_t)pn[mi] * p.H + h) * p.W + w_) * p.C + kq * 8;
+        }
+        const int e0 = (r * p.S + sidx) * p.C;
+        for (int cc = 0; cc < csteps; ++cc) {
+          bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi)
+            afrag[mi] = ok[mi] ? *reinterpret_cast<const bf16x8*>(ab[mi] +
+                                                                  cc * 32)
+                               : bf16x8{};
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            bfrag[ni] =
+                *reinterpret_cast<const bf16x8*>(wr[ni] + e0 + cc * 32);
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+              acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+        }
+      }
     }
+  } else {
+    const int ksteps = p.RSCp / 32;
+    for (int t = 0; t < ksteps; ++t) {
+      const int e = t * 32 + kq * 8;
+      const int r = magic_div2(e, p.magicSC);
+      const int rem2 = e - r * p.SC;
+      const int sidx = magic_div2(rem2, p.magicC);
+      const int c = rem2 - sidx * p.C;
+      bf16x8 afrag[4], bfrag[4];
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
-      bfrag[ni] = *reinterpret_cast<const bf16x8*>(wr[ni] + t * 32);
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 4; ++mi) {
+        const int h = prow[mi] + r;
+        const int w_ = pcol[mi] + sidx;
+        const bool ok = mok[mi] && e < p.RSC && h >= 0 && h < p.H &&
+                        w_ >= 0 && w_ < p.W;
+        afrag[mi] =
+            ok ? *reinterpret_cast<const bf16x8*>(
+                     p.x + (((int64_t)pn[mi] * p.H + h) * p.W + w_) * p.C + c)
+               : bf16x8{};
+      }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+        bfrag[ni] = *reinterpret_cast<const bf16x8*>(wr[ni] + t * 32);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
   }
   float* slab = slab4[wid];
   const int er = lane >> 2, ec = (lane & 3) << 4;

@@ -104,9 +104,25 @@ DEV_INLINE void tr_read_issue(unsigned a0, unsigned a1, TrFrag& f) {
 
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   const int g = blockIdx.z;
-  const int ktile = blockIdx.x % p.ktiles;
-  const int ntile = blockIdx.x / p.ktiles;
-  const int chunk = blockIdx.y;
+  int ktile, ntile, chunk;
+  const int KN = p.ktiles * p.ntiles;
+  if (KN <= 12) {
+    // XCD grouping for small tile counts (stem, 3x3 64ch): one chunk's
+    // <=12 blocks all land on ONE XCD (block b runs on XCD b%8) so the
+    // chunk's x/gy windows are read into one L2 instead of 7-8. (Grouping
+    // every shape this way measured slower — big-KN chunks overflow the
+    // 4 MiB XCD L2; host pads chunks to a multiple of 8 for the 1-D grid.)
+    const int xcd = blockIdx.x & 7, seq = blockIdx.x >> 3;
+    chunk = (seq / KN) * 8 + xcd;
+    if (chunk >= p.chunks) return;
+    const int kn = seq % KN;
+    ktile = kn % p.ktiles;
+    ntile = kn / p.ktiles;
+  } else {
+    ktile = blockIdx.x % p.ktiles;
+    ntile = blockIdx.x / p.ktiles;
+    chunk = blockIdx.y;
+  }
 
   // ONE shared object (a second one forces vmcnt(0) drains before ds_reads)
   __shared__ __align__(16) char smem[2 * 2 * TILE_BYTES];
@@ -788,6 +804,8 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
   accbuf.zero_();
   p.acc = accbuf.data_ptr<float>();
   dim3 grid(p.ktiles * p.ntiles, p.chunks, groups);
+  if (p.ktiles * p.ntiles <= 12)  // XCD-grouped 1-D grid (see kernel)
+    grid = dim3(p.ktiles * p.ntiles * ((p.chunks + 7) / 8 * 8), 1, groups);
   hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(256), 0, cur_stream(), p);
 
   auto gw = at::empty({Kt, Cg, (int64_t)R, (int64_t)S},
