@@ -370,6 +370,7 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
 struct SmallConvParams {
   const __hip_bfloat16* x;  // [N,H,W,C]
   const __hip_bfloat16* w;  // [64, RSCp] span-padded flat
+  const __hip_bfloat16* zbuf;  // 16 zero bytes (clamp target for OOB loads)
   __hip_bfloat16* y;        // [N,Ho,Wo,64]
   int C, H, W;
   int R, S, SC;
@@ -478,6 +479,119 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+  }
+  float* slab = slab4[wid];
+  const int er = lane >> 2, ec = (lane & 3) << 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr)
+        slab[(kq * 4 + rr) * 68 + ni * 16 + il] = acc[mi][ni][rr];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t m = mbase + mi * 16 + er;
+    if (m < p.M) {
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } u;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      __hip_bfloat16* yp = p.y + m * 64 + ec;
+      *reinterpret_cast<uint4*>(yp) = u.q[0];
+      *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
+    }
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+
+// Pipelined variant (C % 32 == 0): PMC showed the plain kernel waits on
+// memory ~85% of its cycles (SQ busy 12%) — each k-step's 8 loads are
+// consumed by the very next MFMAs. This one (a) selects the ADDRESS
+// (OOB -> a 16-byte zero buffer) instead of the loaded value, so no
+// post-load cndmask forces an early waitcnt, and (b) prefetches step i+1's
+// fragments while step i's MFMAs run.
+template <int CSTEPS>  // C / 32
+__global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
+    SmallConvParams p) {
+  __shared__ float slab4[4][16 * 68];
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int il = lane & 15, kq = lane >> 4;
+  const int64_t mbase = (int64_t)blockIdx.x * 256 + wid * 64;
+  int pn[4], prow[4], pcol[4];
+  bool mok[4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+    const int64_t m = mbase + mi * 16 + il;
+    mok[mi] = m < p.M;
+    const int mm = (int)(mok[mi] ? m : 0);
+    const int n = magic_div2(mm, p.magicHoWo);
+    const int rem = mm - n * (p.Ho * p.Wo);
+    const int ho = magic_div2(rem, p.magicWo);
+    const int wo = rem - ho * p.Wo;
+    pn[mi] = n;
+    prow[mi] = ho * p.sh - p.ph;
+    pcol[mi] = wo * p.sw - p.pw;
+  }
+  const __hip_bfloat16* wr[4];
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni)
+    wr[ni] = p.w + (int64_t)(ni * 16 + il) * p.RSCp + kq * 8;
+  const int S = p.S;
+  const int steps = p.R * S * CSTEPS;
+  int r = 0, si = 0, cc = 0;
+  bf16x8 Af[2][4], Bf[2][4];
+  auto issue = [&](int buf) {
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int h = prow[mi] + r;
+      const int w_ = pcol[mi] + si;
+      const bool ok = mok[mi] && h >= 0 && h < p.H && w_ >= 0 && w_ < p.W;
+      const __hip_bfloat16* ap =
+          ok ? p.x + (((int64_t)pn[mi] * p.H + h) * p.W + w_) * p.C +
+                   cc * 32 + kq * 8
+             : p.zbuf;
+      Af[buf][mi] = *reinterpret_cast<const bf16x8*>(ap);
+    }
+    const int e0 = (r * S + si) * p.C + cc * 32;
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      Bf[buf][ni] = *reinterpret_cast<const bf16x8*>(wr[ni] + e0);
+    ++cc;
+    if (cc == CSTEPS) {
+      cc = 0;
+      ++si;
+      if (si == S) {
+        si = 0;
+        ++r;
+      }
+    }
+  };
+  f32x4 acc[4][4] = {};
+  // 2x-unrolled so the double-buffer index is a compile-time constant
+  // (a runtime `i & 1` index on the fragment arrays spills them to
+  // scratch — rule 20 — measured 60x slower)
+  issue(0);
+  for (int i = 0; i < steps; i += 2) {
+    if (i + 1 < steps) issue(1);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            Af[0][mi], Bf[0][ni], acc[mi][ni], 0, 0, 0);
+    if (i + 1 < steps) {
+      if (i + 2 < steps) issue(0);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              Af[1][mi], Bf[1][ni], acc[mi][ni], 0, 0, 0);
     }
   }
   float* slab = slab4[wid];
@@ -621,9 +735,20 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     sp.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
     sp.magicSC = ((1ULL << 47) / (unsigned long long)(S * C_)) + 1;
     sp.magicC = ((1ULL << 47) / (unsigned long long)C_) + 1;
-    hipLaunchKernelGGL(conv_smallk_kernel,
-                       dim3((int)((sp.M + 255) / 256)), dim3(256), 0,
-                       cur_stream(), sp);
+    static at::Tensor zbuf;  // 16 zero bytes; allocated outside any capture
+    if (!zbuf.defined() || zbuf.device() != x.device())
+      zbuf = at::zeros({8}, x.options());
+    sp.zbuf = (const __hip_bfloat16*)zbuf.data_ptr();
+    const dim3 skg((int)((sp.M + 255) / 256));
+    if (C_ == 64)
+      hipLaunchKernelGGL(conv_smallk_pipe_kernel<2>, skg, dim3(256), 0,
+                         cur_stream(), sp);
+    else if (C_ == 32)
+      hipLaunchKernelGGL(conv_smallk_pipe_kernel<1>, skg, dim3(256), 0,
+                         cur_stream(), sp);
+    else
+      hipLaunchKernelGGL(conv_smallk_kernel, skg, dim3(256), 0, cur_stream(),
+                         sp);
     return y;
   }
   auto y = at::empty({N, Kt, Ho, Wo},

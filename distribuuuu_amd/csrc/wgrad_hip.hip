@@ -109,7 +109,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
   const int KN = p.ktiles * p.ntiles;
   if (KN <= 12) {
     // XCD grouping for small tile counts (stem, 3x3 64ch): one chunk's
-    // <=8 blocks all land on ONE XCD (block b runs on XCD b%8) so the
+    // <=12 blocks all land on ONE XCD (block b runs on XCD b%8) so the
     // chunk's x/gy windows are read into one L2 instead of 7-8. (Grouping
     // every shape this way measured slower — big-KN chunks overflow the
     // 4 MiB XCD L2; host pads chunks to a multiple of 8 for the 1-D grid.)
