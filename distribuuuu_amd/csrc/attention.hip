@@ -9,8 +9,10 @@
 // RH = q@rel_h^T [B, L, 2H-1] (fp32, computed by two small gemm_nt calls).
 // rel logits: S[qi, kj] += RW[qi][wj-wi+W-1] + RH[qi][hj-hi+H-1].
 //
-// Output: O [B, L, D] bf16 and P [B, L, L] is NOT materialized — backward
-// recomputes (flash-style) in the Python Function.
+// Output: O [B, L, D] bf16; when a pout tensor is passed, the softmax
+// probs P [B, L, L] are also written (bf16) so backward is a short
+// analytic bmm chain instead of a full torch recompute (at L = 196 the
+// probs are 39 MB/layer — "flash-style never materialize" buys nothing).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4a;
@@ -33,6 +35,7 @@ struct MhsaParams {
   const float* rw;           // [B, L, 2W-1]
   const float* rh;           // [B, L, 2H-1]
   __hip_bfloat16* o;         // [B, L, D]
+  __hip_bfloat16* pout;      // optional [B, L, L] softmax probs (backward)
   int B, L, D, H, W;
   int ltiles16;  // ceil(L/16)
   int lpad;      // LDS row width for S (multiple of 16 + pad)
@@ -118,11 +121,18 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
     s = __shfl(s, 0);
     const float inv = 1.f / s;
     __hip_bfloat16* prow = &P[(wid * 16 + r) * p.lpad];
+    const int qr = qrow_base + r;
+    __hip_bfloat16* gp = (p.pout != nullptr && qr < p.L)
+                             ? p.pout + ((int64_t)b * p.L + qr) * p.L
+                             : nullptr;
     for (int jr = 0; jr < lrounds; ++jr) {
       const int j = jr * 64 + lane;
-      if (j < p.lpad)
-        prow[j] = from_f32<__hip_bfloat16>(
+      if (j < p.lpad) {
+        const __hip_bfloat16 pv = from_f32<__hip_bfloat16>(
             j < p.L ? __expf(row[j] - m) * inv : 0.f);
+        prow[j] = pv;
+        if (gp && j < p.L) gp[j] = pv;
+      }
     }
   }
   __builtin_amdgcn_wave_barrier();
@@ -166,7 +176,8 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
 
 // q pre-scaled; returns O [B, L, D]
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
-                    at::Tensor rh, int64_t H, int64_t W) {
+                    at::Tensor rh, int64_t H, int64_t W,
+                    c10::optional<at::Tensor> pout) {
   CHECK_GPU(q);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "mhsa: bf16 only");
   const int B = q.size(0), L = q.size(1), D = q.size(2);
@@ -180,6 +191,7 @@ at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
   p.rw = rw.data_ptr<float>();
   p.rh = rh.data_ptr<float>();
   p.o = (__hip_bfloat16*)o.data_ptr();
+  p.pout = pout.has_value() ? (__hip_bfloat16*)pout->data_ptr() : nullptr;
   p.B = B; p.L = L; p.D = D; p.H = H; p.W = W;
   p.ltiles16 = (L + 15) / 16;
   p.lpad = ((L + 31) / 32) * 32;  // multiple of 32 for the P fragment chunks

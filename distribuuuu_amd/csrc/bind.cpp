@@ -70,7 +70,8 @@ at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 // attention.hip
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
-                    at::Tensor rh, int64_t H, int64_t W);
+                    at::Tensor rh, int64_t H, int64_t W,
+                    c10::optional<at::Tensor> pout);
 // augment.hip
 at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
                               std::vector<double> mean,
