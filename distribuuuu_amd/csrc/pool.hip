@@ -5,10 +5,14 @@
 namespace {
 
 // ---- max pool fwd: thread per (n,ho,wo,c-pack of 8), vectorized -----------
+// argmax saved as a WINDOW-LOCAL byte offset (kh*K + kw): the earlier int32
+// flat index cost 4x the idx bytes on both sides (the backward gather reads
+// idx for up to 4 windows per input element)
 template <typename T>
 __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                   int* __restrict__ idx, int N, int H, int W,
-                                   int C, int Ho, int Wo, int K, int S, int P) {
+                                   unsigned char* __restrict__ idx, int N,
+                                   int H, int W, int C, int Ho, int Wo, int K,
+                                   int S, int P) {
   constexpr int V = 16 / sizeof(T);
   using Pk = Pack<T, V>;
   const int cpacks = C / V;
@@ -23,7 +27,7 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     const int n = t / Ho;
     const int h0 = ho * S - P, w0 = wo * S - P;
     float best[V];
-    __align__(16) int bidx[V];
+    __align__(8) unsigned char bidx[V];
 #pragma unroll
     for (int j = 0; j < V; ++j) {
       best[j] = -INFINITY;
@@ -42,7 +46,7 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
           float v = to_f32(p.v[j]);
           if (v > best[j]) {
             best[j] = v;
-            bidx[j] = h * W + w;
+            bidx[j] = (unsigned char)(kh * K + kw);
           }
         }
       }
@@ -52,17 +56,19 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < V; ++j) py.v[j] = from_f32<T>(best[j]);
     *reinterpret_cast<Pk*>(y + out) = py;
-#pragma unroll
-    for (int j = 0; j < V / 4; ++j)
-      *reinterpret_cast<int4*>(idx + out + j * 4) =
-          *reinterpret_cast<int4*>(&bidx[j * 4]);
+    if (V == 8)
+      *reinterpret_cast<uint2*>(idx + out) =
+          *reinterpret_cast<uint2*>(&bidx[0]);
+    else
+      *reinterpret_cast<unsigned int*>(idx + out) =
+          *reinterpret_cast<unsigned int*>(&bidx[0]);
   }
 }
 
 // ---- max pool bwd: gather per input c-pack (no atomics) -------------------
 template <typename T>
 __global__ void maxpool_bwd_kernel(const T* __restrict__ gy,
-                                   const int* __restrict__ idx,
+                                   const unsigned char* __restrict__ idx,
                                    T* __restrict__ gx, int N, int H, int W,
                                    int C, int Ho, int Wo, int K, int S, int P) {
   constexpr int V = 16 / sizeof(T);
@@ -77,22 +83,25 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ gy,
     t /= W;
     const int h = t % H;
     const int n = t / H;
-    const int flat = h * W + w;
     float acc[V] = {};
     const int ho_lo = max(0, (h + P - K + S) / S), ho_hi = min(Ho - 1, (h + P) / S);
     const int wo_lo = max(0, (w + P - K + S) / S), wo_hi = min(Wo - 1, (w + P) / S);
     for (int ho = ho_lo; ho <= ho_hi; ++ho)
       for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        // window-local offset this input element would need to have won
+        const int want = (h - (ho * S - P)) * K + (w - (wo * S - P));
         const int64_t o = (((int64_t)n * Ho + ho) * Wo + wo) * C + cp * V;
-        __align__(16) int bi[V];
-#pragma unroll
-        for (int j = 0; j < V / 4; ++j)
-          *reinterpret_cast<int4*>(&bi[j * 4]) =
-              *reinterpret_cast<const int4*>(idx + o + j * 4);
+        __align__(8) unsigned char bi[V];
+        if (V == 8)
+          *reinterpret_cast<uint2*>(&bi[0]) =
+              *reinterpret_cast<const uint2*>(idx + o);
+        else
+          *reinterpret_cast<unsigned int*>(&bi[0]) =
+              *reinterpret_cast<const unsigned int*>(idx + o);
         Pk pg = *reinterpret_cast<const Pk*>(gy + o);
 #pragma unroll
         for (int j = 0; j < V; ++j)
-          if (bi[j] == flat) acc[j] += to_f32(pg.v[j]);
+          if (bi[j] == want) acc[j] += to_f32(pg.v[j]);
       }
     Pk out;
 #pragma unroll
@@ -194,7 +203,7 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto idx = at::empty({N, C, Ho, Wo},
                        x.options()
-                           .dtype(at::kInt)
+                           .dtype(at::kByte)
                            .memory_format(at::MemoryFormat::ChannelsLast));
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "maxpool_fwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
@@ -203,7 +212,7 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
     hipLaunchKernelGGL((maxpool_fwd_kernel<scalar_t>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
                        (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
-                       idx.data_ptr<int>(), N, H, W, C, Ho, Wo, K, S, P);
+                       idx.data_ptr<unsigned char>(), N, H, W, C, Ho, Wo, K, S, P);
   });
   return {y, idx};
 }
@@ -220,7 +229,7 @@ at::Tensor maxpool_bwd(at::Tensor gy, at::Tensor idx, int64_t H, int64_t W,
     int64_t total = (int64_t)N * H * W * (C / V);
     hipLaunchKernelGGL((maxpool_bwd_kernel<scalar_t>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
-                       (const scalar_t*)gy.data_ptr(), idx.data_ptr<int>(),
+                       (const scalar_t*)gy.data_ptr(), idx.data_ptr<unsigned char>(),
                        (scalar_t*)gx.data_ptr(), N, H, W, C, Ho, Wo, K, S, P);
   });
   return gx;
