@@ -676,7 +676,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     return e && e[0] == '0';
   }();
   const int C_ = x.size(1);
-  if (!v2_off && groups == 1 && Kt >= 192 && R * S * C_ >= 576 &&
+  if (!v2_off && groups == 1 && Kt >= 192 && R * S * C_ >= 512 &&
       x.scalar_type() == at::kBFloat16)
     return conv2d_fwd_v2(x, w, sh, sw, ph, pw, dh, dw);
   static const bool small_off = []() {
