@@ -24,49 +24,82 @@ def _cl(x):
 # ---------------------------------------------------------------------------
 # Convolution (K1-K5): implicit-GEMM MFMA kernels on GPU
 # ---------------------------------------------------------------------------
+def _pad8(n):
+    return (n + 7) // 8 * 8
+
+
+def _pad_k(w, k8):
+    """Zero-pad output channels (dim 0) of a channels_last weight."""
+    k, c, r, s = w.shape
+    wp = w.new_zeros((k8, c, r, s)).contiguous(
+        memory_format=torch.channels_last)
+    wp[:k] = w
+    return wp
+
+
 class _HIPConv2d(torch.autograd.Function):
+    """groups==1 convs whose C or K is not a multiple of 8 (stem C=3, SE
+    squeeze widths like 58) run on zero-padded channels — padding C cannot
+    change the output, and padded-K rows are sliced off. Without this, such
+    shapes fell back to ATen/MIOpen naive kernels (~230 us per tiny SE conv
+    on RegNetY)."""
+
     @staticmethod
     def forward(ctx, x, w, bias, stride, padding, dilation, groups):
         x = _cl(x)
         w = _cl(w)
         e = ext()
         cin = x.shape[1]
+        kout = w.shape[0]
+        cpad = groups == 1 and cin % 8 != 0
+        kpad = groups == 1 and kout % 8 != 0
         xp, wp = x, w
-        if cin < 8 and groups == 1:  # stem C=3: zero-pad channels to 8
-            xp = e.pad_channels(x, 8)
-            wp = e.pad_channels(w, 8)
+        if cpad:
+            xp = e.pad_channels(x, _pad8(cin))
+            wp = e.pad_channels(w, _pad8(cin))
+        if kpad:
+            wp = _pad_k(wp, _pad8(kout))
         y = e.conv2d_fwd(xp, wp, stride[0], stride[1], padding[0], padding[1],
                          dilation[0], dilation[1], groups)
+        if kpad:
+            y = _cl(y[:, :kout])
         if bias is not None:
             y = y + bias.reshape(1, -1, 1, 1)
         ctx.save_for_backward(x, w)
-        ctx.conf = (stride, padding, dilation, groups, bias is not None, cin)
+        ctx.conf = (stride, padding, dilation, groups, bias is not None, cin,
+                    kout, cpad, kpad)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         x, w = ctx.saved_tensors
-        stride, padding, dilation, groups, has_bias, cin = ctx.conf
+        (stride, padding, dilation, groups, has_bias, cin, kout, cpad,
+         kpad) = ctx.conf
         gy = _cl(gy)
         e = ext()
         gx = gw = gb = None
+        gyp = e.pad_channels(gy, _pad8(kout)) if kpad else gy
         if ctx.needs_input_grad[0]:
-            wd = e.pad_channels(w, 8) if (cin < 8 and groups == 1) else w
-            gx = e.conv2d_dgrad(gy, wd, x.shape[2], x.shape[3],
+            wd = e.pad_channels(w, _pad8(cin)) if cpad else w
+            if kpad:
+                wd = _pad_k(wd, _pad8(kout))
+            gx = e.conv2d_dgrad(gyp, wd, x.shape[2], x.shape[3],
                                 stride[0], stride[1], padding[0], padding[1],
                                 dilation[0], dilation[1], groups)
-            if cin < 8 and groups == 1:
+            if cpad:
                 gx = _cl(gx[:, :cin])
         if ctx.needs_input_grad[1]:
-            xw = x
-            if cin < 8 and groups == 1:  # stem: wgrad on the zero-padded input
-                xw = e.pad_channels(x, 8)
-            gw = e.conv2d_wgrad(gy, xw, w.shape[2], w.shape[3],
+            xw = e.pad_channels(x, _pad8(cin)) if cpad else x
+            gw = e.conv2d_wgrad(gyp, xw, w.shape[2], w.shape[3],
                                 stride[0], stride[1], padding[0],
                                 padding[1], dilation[0], dilation[1],
                                 groups)
-            if cin < 8 and groups == 1:
-                gw = _cl(gw[:, :cin])
+            if kpad:
+                gw = gw[:kout]
+            if cpad:
+                gw = gw[:, :cin]
+            if kpad or cpad:
+                gw = _cl(gw)
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
         return gx, gw, gb, None, None, None, None
@@ -111,12 +144,12 @@ def _is_depthwise(x, weight, groups, dilation):
 def _hip_conv_ok(x, weight, groups):
     if x.dtype != torch.bfloat16:
         return False
+    if groups == 1:
+        return True  # any C/K: the pad-channels path covers %8 misfits
     cg = weight.shape[1]
     kg = weight.shape[0] // groups
-    if cg % 8 != 0 and not (cg < 8 and groups == 1):
-        return False  # depthwise & odd group widths: dedicated kernels pending
-    if kg % 8 != 0:
-        return False
+    if cg % 8 != 0 or kg % 8 != 0:
+        return False  # odd group widths: dedicated kernels pending
     return True
 
 

@@ -255,3 +255,31 @@ def test_mhsa_fused_backward():
                      (rel_h.grad, rhf.grad), (rel_w.grad, rwf.grad)]:
         err = (got.float() - ref).abs().max().item()
         assert err < 6e-2 * max(ref.abs().max().item(), 1.0), err
+
+
+@pytest.mark.parametrize("cin,kout,r", [(58, 232, 1), (232, 58, 1), (30, 58, 3)])
+def test_conv_odd_channels_pad_path(cin, kout, r):
+    """groups==1 convs with C or K not %8 (SE squeeze widths) run on
+    zero-padded channels; forward and all grads must match fp32 ATen."""
+    import distribuuuu_amd.ops.functional as DF
+
+    torch.manual_seed(0)
+    x = torch.randn(8, cin, 7, 7, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True).to(memory_format=torch.channels_last)
+    w = torch.randn(kout, cin, r, r, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True).to(memory_format=torch.channels_last)
+    b = torch.randn(kout, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = DF.conv2d(x, w, b, (1, 1), (r // 2, r // 2), (1, 1), 1)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    yf = F.conv2d(xf, wf, bf, 1, r // 2)
+    yf.backward(g.float())
+    assert torch.allclose(y.float(), yf, atol=5e-1, rtol=5e-2)
+    assert torch.allclose(x.grad.float(), xf.grad, atol=5e-1, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=2.0, rtol=5e-2)
+    assert torch.allclose(b.grad.float(), bf.grad, atol=5e-1, rtol=5e-2)
