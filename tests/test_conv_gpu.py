@@ -264,10 +264,10 @@ def test_conv_odd_channels_pad_path(cin, kout, r):
     import distribuuuu_amd.ops.functional as DF
 
     torch.manual_seed(0)
-    x = torch.randn(8, cin, 7, 7, device="cuda", dtype=torch.bfloat16,
-                    requires_grad=True).to(memory_format=torch.channels_last)
-    w = torch.randn(kout, cin, r, r, device="cuda", dtype=torch.bfloat16,
-                    requires_grad=True).to(memory_format=torch.channels_last)
+    x = torch.randn(8, cin, 7, 7, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last).requires_grad_(True)
+    w = torch.randn(kout, cin, r, r, device="cuda", dtype=torch.bfloat16).to(
+        memory_format=torch.channels_last).requires_grad_(True)
     b = torch.randn(kout, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
     y = DF.conv2d(x, w, b, (1, 1), (r // 2, r // 2), (1, 1), 1)
