@@ -51,7 +51,9 @@ class DistributedDataParallel(nn.Module):
 
     # -- setup -------------------------------------------------------------
     def _broadcast_state(self):
-        """One-shot rank-0 -> all param+buffer broadcast, coalesced per dtype."""
+        """One-shot rank-0 -> all param+buffer broadcast at construction
+        (per tensor; init-only, so latency is irrelevant — the per-step
+        buffer sync below is the coalesced one)."""
         tensors = [p.data for p in self.module.parameters()]
         tensors += [b.data for b in self.module.buffers()]
         for t in tensors:
