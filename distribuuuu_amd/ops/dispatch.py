@@ -31,8 +31,11 @@ def ext():
         from distribuuuu_amd import _hip_ops  # built in-tree by setup.py
 
         _EXT = _hip_ops
-    except ImportError:
+    except ImportError as exc:
         _EXT = None
+        if torch.cuda.is_available():
+            # surface the real cause (a bad dlopen looks like "not built")
+            warnings.warn(f"_hip_ops import failed on a GPU machine: {exc!r}")
     return _EXT
 
 
