@@ -227,13 +227,18 @@ def accuracy(output, target, topk=(1,)):
 # LR schedules (reference utils.py:280-316) — pure functions of epoch
 # ---------------------------------------------------------------------------
 def lr_fun_steps(cur_epoch):
-    ind = sum(1 for s in cfg.OPTIM.STEPS if cur_epoch >= s)
-    return cfg.OPTIM.BASE_LR * (cfg.OPTIM.LR_MULT ** ind)
+    """Multiplier LR_MULT**ind; reference semantics (utils.py:280-284):
+    STEPS lists epoch boundaries STARTING WITH 0, and ind is the index of
+    the last boundary passed (so STEPS=[0, 30, 60] halves twice)."""
+    ind = [i for i, s in enumerate(cfg.OPTIM.STEPS) if cur_epoch >= s][-1]
+    return cfg.OPTIM.LR_MULT ** ind
 
 
 def lr_fun_cos(cur_epoch):
+    """Half-period cosine multiplier with MIN_LR as a RELATIVE floor
+    (reference utils.py:287-291: BASE_LR * [(1-MIN_LR)*cos + MIN_LR])."""
     lr = 0.5 * (1.0 + math.cos(math.pi * cur_epoch / cfg.OPTIM.MAX_EPOCH))
-    return (cfg.OPTIM.BASE_LR - cfg.OPTIM.MIN_LR) * lr + cfg.OPTIM.MIN_LR
+    return (1.0 - cfg.OPTIM.MIN_LR) * lr + cfg.OPTIM.MIN_LR
 
 
 def get_lr_fun():
@@ -245,7 +250,7 @@ def get_lr_fun():
 
 def get_epoch_lr(cur_epoch):
     """Schedule LR with linear warmup over WARMUP_EPOCHS from WARMUP_FACTOR."""
-    lr = get_lr_fun()(cur_epoch)
+    lr = get_lr_fun()(cur_epoch) * cfg.OPTIM.BASE_LR
     if cur_epoch < cfg.OPTIM.WARMUP_EPOCHS:
         alpha = cur_epoch / cfg.OPTIM.WARMUP_EPOCHS
         warmup_factor = cfg.OPTIM.WARMUP_FACTOR * (1.0 - alpha) + alpha

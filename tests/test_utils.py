@@ -28,7 +28,9 @@ def test_lr_cos_schedule():
 def test_lr_steps_schedule():
     cfg.OPTIM.LR_POLICY = "steps"
     cfg.OPTIM.BASE_LR = 0.1
-    cfg.OPTIM.STEPS = [30, 60]
+    # reference convention: STEPS starts with 0 (the [-1] indexing of
+    # lr_fun_steps requires it); each later boundary multiplies by LR_MULT
+    cfg.OPTIM.STEPS = [0, 30, 60]
     cfg.OPTIM.LR_MULT = 0.1
     cfg.OPTIM.WARMUP_EPOCHS = 0
     assert utils.get_epoch_lr(10) == pytest.approx(0.1)
