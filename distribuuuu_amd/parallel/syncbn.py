@@ -3,7 +3,7 @@
 Replaces torch.nn.SyncBatchNorm (reference `trainer.py:131`,
 `nn.SyncBatchNorm.convert_sync_batchnorm`). Forward computes local per-channel
 sum / sum-of-squares with the bn_sums HIP kernel, all-reduces ONE coalesced
-[2C+1] fp32 tensor (the reference's implementation all-reduces mean & var
+[2C] fp32 tensor (the reference's implementation all-reduces mean & var
 separately per layer), and normalizes with the fused apply(+residual)(+act)
 kernel. Backward computes local RAW grad-stat sums with bn_bwd_stats,
 all-reduces the [2C] vector, and finishes with the same fused finalize+dx
