@@ -32,8 +32,29 @@ def parse_args():
     return p.parse_args()
 
 
+def _self_launch(args):
+    """`python bench.py --gpus N` without torchrun must NOT silently measure
+    1 GPU: spawn torch.distributed.run ourselves, one rank per GPU."""
+    import socket
+    import subprocess
+    import sys
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", f"--nproc-per-node={args.gpus}",
+           "--master-addr=127.0.0.1", f"--master-port={port}",
+           os.path.abspath(__file__)] + sys.argv[1:]
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    raise SystemExit(subprocess.call(cmd, env=env))
+
+
 def main():
     args = parse_args()
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        _self_launch(args)
     world_size = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))

@@ -213,20 +213,29 @@ class GPUAugLoader:
         ev.record(self.stream)
         return x, labels_d, ev
 
+    @staticmethod
+    def _hand_off(pending):
+        x, y, ev = pending
+        cur = torch.cuda.current_stream()
+        cur.wait_event(ev)
+        # the consumer stream now owns these side-stream allocations: without
+        # record_stream the caching allocator could recycle the block for a
+        # later side-stream aug batch while main-stream backward still reads
+        # the saved inputs (wgrad) — intermittent silent batch corruption
+        x.record_stream(cur)
+        y.record_stream(cur)
+        return x, y
+
     def __iter__(self):
         it = iter(self.loader)
         pending = None
         for cpu_batch in it:
             issued = self._issue(cpu_batch)
             if pending is not None:
-                x, y, ev = pending
-                torch.cuda.current_stream().wait_event(ev)
-                yield x, y
+                yield self._hand_off(pending)
             pending = issued
         if pending is not None:
-            x, y, ev = pending
-            torch.cuda.current_stream().wait_event(ev)
-            yield x, y
+            yield self._hand_off(pending)
 
 
 def construct_train_loader():
@@ -300,10 +309,11 @@ def _dist():
 
 
 class DeviceSyntheticLoader:
-    """Device-resident synthetic batches for benchmarking: tensors are created
-    once in HBM (288 GB per GPU — keep them resident) and
-    lightly permuted per step on a side stream; zero H2D traffic, zero CPU
-    worker cost. Yields (inputs, targets) like a DataLoader."""
+    """Device-resident synthetic batches for benchmarking: one fixed batch is
+    created in HBM at init (288 GB per GPU — keep it resident)
+    and yielded every step; zero H2D traffic, zero CPU worker cost. The
+    compute per step is identical to a fresh batch (bench.py reports
+    ``data: synthetic``). Yields (inputs, targets) like a DataLoader."""
 
     def __init__(self, batch_size, im_size=224, num_classes=1000, length=10 ** 9,
                  device="cuda", dtype=torch.float32, channels_last=False):

@@ -9,7 +9,7 @@ needed.
 import torch
 import torch.nn.functional as F
 
-from .dispatch import use_hip, ext
+from .dispatch import use_hip, ext, fallback_warn
 
 
 def rel_to_abs(x):
@@ -139,4 +139,7 @@ def mhsa_relpos(q, k, v, rel_h, rel_w, h, w):
     if (use_hip(q, "mhsa_fwd") and q.dtype == torch.bfloat16
             and q.shape[-1] % 32 == 0 and q.shape[-1] <= 128):
         return _HIPMHSARelPos.apply(q, k, v, rel_h, rel_w, h, w)
+    if q.is_cuda:
+        fallback_warn("mhsa_relpos",
+                      f"dtype {q.dtype} head_dim {q.shape[-1]}")
     return _torch_mhsa(q, k, v, rel_h, rel_w, h, w)

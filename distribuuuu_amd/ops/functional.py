@@ -9,7 +9,7 @@ weights [K,C,R,S] logical → [K,R,S,C] physical.
 import torch
 import torch.nn.functional as F
 
-from .dispatch import ext, use_hip
+from .dispatch import ext, fallback_warn, use_hip
 
 _ACTS = {"none": 0, "relu": 1, "silu": 2, "sigmoid": 3}
 
@@ -143,12 +143,15 @@ def _is_depthwise(x, weight, groups, dilation):
 
 def _hip_conv_ok(x, weight, groups):
     if x.dtype != torch.bfloat16:
+        fallback_warn("conv2d", f"dtype {x.dtype} (only bf16 has MFMA kernels)")
         return False
     if groups == 1:
         return True  # any C/K: the pad-channels path covers %8 misfits
     cg = weight.shape[1]
     kg = weight.shape[0] // groups
     if cg % 8 != 0 or kg % 8 != 0:
+        fallback_warn("conv2d",
+                      f"grouped conv width C/g={cg} K/g={kg} not %8")
         return False  # odd group widths: dedicated kernels pending
     return True
 
@@ -377,6 +380,9 @@ def linear(x, weight, bias=None):
     if (use_hip(x, "gemm_nt") and x.dtype == torch.bfloat16 and x.dim() == 2
             and x.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
         return _HIPLinear.apply(x, weight, bias)
+    if x.is_cuda:
+        fallback_warn("linear",
+                      f"dtype {x.dtype} dim {x.dim()} shape {tuple(x.shape)}")
     return F.linear(x, weight, bias)
 
 

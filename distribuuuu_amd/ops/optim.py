@@ -92,13 +92,33 @@ class HIPSGD(SGD):
                                group["nesterov"])
         return loss
 
-    def state_dict(self):
-        # drop the pointer cache from serialized state
+    def _drop_table_cache(self):
         for group in self.param_groups:
             group.pop("_hip_table", None)
             group.pop("_hip_table_pin", None)
             group.pop("_hip_table_dev", None)
+
+    def state_dict(self):
+        # drop the pointer cache from serialized state
+        self._drop_table_cache()
         return super().state_dict()
+
+    def load_state_dict(self, state_dict):
+        # loading replaces momentum/master tensors; the cached device table
+        # embeds their raw pointers, so it must be rebuilt on the next step
+        super().load_state_dict(state_dict)
+        self._drop_table_cache()
+        # loaded state tensors may arrive in the wrong dtype/device; the
+        # table kernel requires fp32 momentum/master on the param's device
+        for group in self.param_groups:
+            for p in group["params"]:
+                state = self.state.get(p)
+                if not state:
+                    continue
+                for k in ("momentum_buffer", "master"):
+                    if k in state and state[k] is not None:
+                        state[k] = state[k].to(device=p.device,
+                                               dtype=torch.float32)
 
     def _fallback_step(self, loss):
         """Plain SGD math, with fp32 master handling for bf16 params."""

@@ -71,13 +71,16 @@ def train_epoch(train_loader, net, criterion, optimizer, epoch, device, dtype):
                                                     "set_epoch"):
         train_loader.sampler.set_epoch(epoch)
     net.train()
+    # world==1: set_to_none skips the per-step grad memsets (161 for ResNet-50);
+    # under DDP the bucket grad-views must stay installed, so keep memsets there
+    zero_none = utils.get_world_size() == 1
     end = time.time()
     for idx, (inputs, targets) in enumerate(train_loader):
         data_time.update(time.time() - end)
         inputs, targets = _prepare_batch(inputs, targets, device, dtype)
         outputs = net(inputs)
         loss = criterion(outputs.float(), targets)
-        optimizer.zero_grad(set_to_none=False)
+        optimizer.zero_grad(set_to_none=zero_none)
         loss.backward()
         optimizer.step()
         acc1, acck = utils.accuracy(outputs, targets, topk=(1, topk))

@@ -209,7 +209,9 @@ def accuracy(output, target, topk=(1,)):
             c1, ck = ext().topk_acc(output.float().contiguous(),
                                     target.contiguous(), topk[1])
             n = output.shape[0]
-            return [c1[0].float() * (100.0 / n), ck[0].float() * (100.0 / n)]
+            # shape-[1] tensors to match the fallback (callers index res[0])
+            return [c1.reshape(1).float() * (100.0 / n),
+                    ck.reshape(1).float() * (100.0 / n)]
     with torch.no_grad():
         maxk = max(topk)
         batch_size = target.size(0)
@@ -343,13 +345,18 @@ def save_checkpoint(model, optimizer, epoch, best_acc1, best=False):
 
 def load_checkpoint(checkpoint_file, model, optimizer=None):
     """CPU-mapped load tolerant of bare state_dicts and missing optimizer state.
-    Returns the next epoch to run (0 for bare state_dicts)."""
+    Returns the next epoch to run. Matching the reference (utils.py:390-410):
+    start_epoch/best_acc1 advance ONLY when optimizer state was actually
+    restored — a weights-only load restarts at epoch 0 with fresh optimizer."""
     ckpt = torch.load(checkpoint_file, map_location="cpu", weights_only=False)
     target = unwrap_model(model)
+    start_epoch, best_acc1 = 0, 0.0
     if isinstance(ckpt, dict) and "state_dict" in ckpt:
         target.load_state_dict(ckpt["state_dict"])
         if optimizer is not None and "optimizer" in ckpt and cfg.TRAIN.LOAD_OPT:
             optimizer.load_state_dict(ckpt["optimizer"])
-        return ckpt.get("epoch", -1) + 1, ckpt.get("best_acc1", 0.0)
-    target.load_state_dict(ckpt)
-    return 0, 0.0
+            start_epoch = ckpt.get("epoch", -1) + 1
+            best_acc1 = ckpt.get("best_acc1", 0.0)
+    else:
+        target.load_state_dict(ckpt)
+    return start_epoch, best_acc1

@@ -18,8 +18,11 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "distribuuuu_amd", "csrc")
 
+# *_hip.hip are torch-hipify build fallout (byte-copies of the hand-written
+# sources with one #include swapped) — regenerated every build, never sources
 sources = sorted(glob.glob(os.path.join(CSRC, "*.cpp"))) + sorted(
-    glob.glob(os.path.join(CSRC, "*.hip"))
+    f for f in glob.glob(os.path.join(CSRC, "*.hip"))
+    if not f.endswith("_hip.hip")
 )
 
 setup(

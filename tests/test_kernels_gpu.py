@@ -173,6 +173,8 @@ def test_cross_entropy():
 
 
 def test_topk_acc():
+    # oracle = the pure-torch fallback on CPU tensors (utils.accuracy on CUDA
+    # would dispatch back to the same kernel — circular, ADVICE round-1)
     e = _ext()
     torch.manual_seed(0)
     logits = torch.randn(128, 100, device="cuda")
@@ -180,9 +182,15 @@ def test_topk_acc():
     c1, ck = e.topk_acc(logits, target, 5)
     from distribuuuu_amd import utils
 
-    a1, a5 = utils.accuracy(logits, target, topk=(1, 5))
+    a1, a5 = utils.accuracy(logits.cpu(), target.cpu(), topk=(1, 5))
     assert c1.item() / 128 * 100 == pytest.approx(a1.item(), abs=1e-3)
     assert ck.item() / 128 * 100 == pytest.approx(a5.item(), abs=1e-3)
+    # utils.accuracy on the GPU path must return shape-[1] tensors so
+    # trainer.py's acc1[0] indexing works (ADVICE round-1 high)
+    g1, g5 = utils.accuracy(logits, target, topk=(1, 5))
+    assert g1.shape == (1,) and g5.shape == (1,)
+    assert g1.item() == pytest.approx(a1.item(), abs=1e-3)
+    assert g5.item() == pytest.approx(a5.item(), abs=1e-3)
 
 
 @pytest.mark.parametrize("param_dtype", [torch.float32, torch.bfloat16])
