@@ -7,24 +7,26 @@ at::Tensor relu_bwd(at::Tensor gy, at::Tensor y);
 at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b);
 // bn.hip
 std::vector<at::Tensor> bn_sums(at::Tensor x);
+at::Tensor bn_reduce_partials(at::Tensor part);
 std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
                                  at::Tensor beta,
                                  c10::optional<at::Tensor> rm_opt,
                                  c10::optional<at::Tensor> rv_opt,
-                                 double momentum, double eps, bool training);
+                                 double momentum, double eps, bool training,
+                                 c10::optional<at::Tensor> part_opt);
 at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
                         int64_t act, c10::optional<at::Tensor> res);
-at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
+at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x,
                         c10::optional<at::Tensor> res, at::Tensor scale,
                         at::Tensor shift, int64_t act);
-std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
+std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x,
                                      c10::optional<at::Tensor> res,
                                      at::Tensor mean, at::Tensor rstd,
                                      at::Tensor gamma, at::Tensor scale,
                                      at::Tensor shift, at::Tensor sums,
                                      double total_count, int64_t act,
                                      bool training, bool need_gres);
-std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
+std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
                                at::Tensor scale, at::Tensor shift,
@@ -49,6 +51,9 @@ std::vector<at::Tensor> topk_acc(at::Tensor logits, at::Tensor target,
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                       int64_t ph, int64_t pw, int64_t dh, int64_t dw,
                       int64_t groups);
+std::vector<at::Tensor> conv2d_fwd_bn(at::Tensor x, at::Tensor w, int64_t sh,
+                                      int64_t sw, int64_t ph, int64_t pw,
+                                      int64_t dh, int64_t dw, int64_t groups);
 at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw,
                         int64_t dh, int64_t dw, int64_t groups);
@@ -85,21 +90,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("bn_sums", &bn_sums);
+  m.def("bn_reduce_partials", &bn_reduce_partials);
   m.def("bn_stats", &bn_stats, py::arg("x"), py::arg("gamma"), py::arg("beta"),
         py::arg("rm") = py::none(), py::arg("rv") = py::none(),
         py::arg("momentum") = 0.1, py::arg("eps") = 1e-5,
-        py::arg("training") = true);
+        py::arg("training") = true, py::arg("part") = py::none());
   m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
   m.def("bn_bwd_stats", &bn_bwd_stats, py::arg("gy"), py::arg("x"),
-        py::arg("y"), py::arg("res"), py::arg("scale"), py::arg("shift"),
+        py::arg("res"), py::arg("scale"), py::arg("shift"),
         py::arg("act"));
   m.def("bn_bwd_apply", &bn_bwd_apply, py::arg("gy"), py::arg("x"),
-        py::arg("y"), py::arg("res"), py::arg("mean"), py::arg("rstd"),
+        py::arg("res"), py::arg("mean"), py::arg("rstd"),
         py::arg("gamma"), py::arg("scale"), py::arg("shift"), py::arg("sums"),
         py::arg("total_count"), py::arg("act"), py::arg("training"),
         py::arg("need_gres"));
-  m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"), py::arg("y"),
+  m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"),
         py::arg("res"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
         py::arg("scale"), py::arg("shift"), py::arg("act"),
         py::arg("training"), py::arg("need_gres"));
@@ -114,6 +120,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_acc", &topk_acc);
   m.def("sgd_step", &sgd_step);
   m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd_bn", &conv2d_fwd_bn);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("gemm_nt", &gemm_nt);

@@ -27,9 +27,14 @@ DEV_INLINE float act_apply(float z, int act) {
   }
 }
 
-DEV_INLINE float act_grad(float y, float z, int act) {
+// act' from the RECOMPUTED pre-activation z = x*scale + shift (+res): the
+// forward's y stream is never read in backward (F3a of
+// docs/DESIGN_bn_conv_fusion.md — one full tensor stream dropped from both
+// backward passes). relu's mask z > 0 is bit-identical to forward's y > 0
+// (same fp32 math on the same bf16 inputs).
+DEV_INLINE float act_grad(float z, int act) {
   switch (act) {
-    case 1: return y > 0.f ? 1.f : 0.f;
+    case 1: return z > 0.f ? 1.f : 0.f;
     case 2: {
       float s = 1.f / (1.f + __expf(-z));
       return s * (1.f + z * (1.f - s));
@@ -314,11 +319,11 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 }
 
 // ---- bwd stage 1: RAW per-channel sums sum(g), sum(g*x) --------------------
-// g = gy * act'; act' needs only y (relu) or z = scale*x+shift(+res)
-// (silu/sigmoid). No mean/rstd/gamma gathers on the hot path.
+// g = gy * act'; act' recomputes z = scale*x+shift(+res) for every act, so
+// the y stream is never read. No mean/rstd/gamma gathers on the hot path.
 template <typename T, bool HAS_RES, int ACT>
 __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
-    const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
+    const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, float* __restrict__ part, int64_t rows,
     int C, int64_t rows_per_block) {
@@ -335,11 +340,10 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
   const int64_t row1 = min(row0 + rows_per_block, rows);
   const P* gp = reinterpret_cast<const P*>(gy);
   const P* xp = reinterpret_cast<const P*>(x);
-  const P* ypk = reinterpret_cast<const P*>(y);
   const P* rp = reinterpret_cast<const P*>(res);
   for (int cp = cp0; cp < cpacks; cp += ncp) {
     float sc[V], sh[V];
-    if (ACT >= 2) {
+    if (ACT != 0) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         sc[j] = scale[cp * V + j];
@@ -355,20 +359,17 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
       const int64_t rstep = (int64_t)nrl * cpacks;
       const P* gq = gp + row * cpacks + cp;
       const P* xq = xp + row * cpacks + cp;
-      const P* yq = ypk + row * cpacks + cp;
       const P* rq = rp + row * cpacks + cp;
       for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
-        P pg4[4], px4[4], py4[4], pr4[4];
+        P pg4[4], px4[4], pr4[4];
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
           pg4[u] = gq[u * rstep];
           px4[u] = xq[u * rstep];
-          if (ACT != 0) py4[u] = yq[u * rstep];
-          if (HAS_RES && ACT >= 2) pr4[u] = rq[u * rstep];
+          if (HAS_RES && ACT != 0) pr4[u] = rq[u * rstep];
         }
         gq += 4 * rstep;
         xq += 4 * rstep;
-        yq += 4 * rstep;
         rq += 4 * rstep;
 #pragma unroll
         for (int u = 0; u < 4; ++u)
@@ -377,36 +378,28 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
             float xv = to_f32(px4[u].v[j]);
             float g = to_f32(pg4[u].v[j]);
             if (ACT != 0) {
-              float z = 0.f;
-              if (ACT >= 2) {
-                z = xv * sc[j] + sh[j];
-                if (HAS_RES) z += to_f32(pr4[u].v[j]);
-              }
-              g *= act_grad(to_f32(py4[u].v[j]), z, ACT);
+              float z = xv * sc[j] + sh[j];
+              if (HAS_RES) z += to_f32(pr4[u].v[j]);
+              g *= act_grad(z, ACT);
             }
             accg[j] += g;
             accgx[j] += g * xv;
           }
       }
       for (; row < row1; row += nrl) {
-        P pg = gq[0], px = xq[0], py, prr;
-        if (ACT != 0) py = yq[0];
-        if (HAS_RES && ACT >= 2) prr = rq[0];
+        P pg = gq[0], px = xq[0], prr;
+        if (HAS_RES && ACT != 0) prr = rq[0];
         gq += rstep;
         xq += rstep;
-        yq += rstep;
         rq += rstep;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float xv = to_f32(px.v[j]);
           float g = to_f32(pg.v[j]);
           if (ACT != 0) {
-            float z = 0.f;
-            if (ACT >= 2) {
-              z = xv * sc[j] + sh[j];
-              if (HAS_RES) z += to_f32(prr.v[j]);
-            }
-            g *= act_grad(to_f32(py.v[j]), z, ACT);
+            float z = xv * sc[j] + sh[j];
+            if (HAS_RES) z += to_f32(prr.v[j]);
+            g *= act_grad(z, ACT);
           }
           accg[j] += g;
           accgx[j] += g * xv;
@@ -500,9 +493,10 @@ __global__ void bn_bwd_finalize_kernel(
 }
 
 // ---- bwd dx: gx = P1*g + P3*x + P2 (+ gres = g) ----------------------------
+// act mask recomputed from z = x*scale+shift(+res); no y stream (F3a).
 template <typename T, int V, bool HAS_RES>
 __global__ void bn_bwd_dx_kernel(
-    const T* __restrict__ gy, const T* __restrict__ x, const T* __restrict__ y,
+    const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, const float* __restrict__ P1c,
     const float* __restrict__ P2c, const float* __restrict__ P3c,
@@ -518,7 +512,6 @@ __global__ void bn_bwd_dx_kernel(
   const F4* h4 = reinterpret_cast<const F4*>(shift);
   const P* gp = reinterpret_cast<const P*>(gy);
   const P* xp = reinterpret_cast<const P*>(x);
-  const P* ypk = reinterpret_cast<const P*>(y);
   const P* rp = reinterpret_cast<const P*>(res);
   P* oxp = reinterpret_cast<P*>(gx);
   P* orp = reinterpret_cast<P*>(gres);
@@ -528,12 +521,11 @@ __global__ void bn_bwd_dx_kernel(
   int c = (int)(i % cpacks);
   gp += i;
   xp += i;
-  ypk += i;
   rp += i;
   oxp += i;
   orp += i;
   for (; i + stride < npacks; i += 2 * stride) {
-    P pg2[2], px2[2], py2[2], pr2[2];
+    P pg2[2], px2[2], pr2[2];
     int cu[2] = {c, c + dstep};
     if (cu[1] >= cpacks) cu[1] -= cpacks;
     c = cu[1] + dstep;
@@ -542,8 +534,7 @@ __global__ void bn_bwd_dx_kernel(
     for (int u = 0; u < 2; ++u) {
       pg2[u] = gp[u * stride];
       px2[u] = xp[u * stride];
-      if (act != 0) py2[u] = ypk[u * stride];
-      if (HAS_RES && act >= 2) pr2[u] = rp[u * stride];
+      if (HAS_RES && act != 0) pr2[u] = rp[u * stride];
     }
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
@@ -553,7 +544,7 @@ __global__ void bn_bwd_dx_kernel(
         v1[q] = p14[cu[u] * NF4 + q];
         v2[q] = p24[cu[u] * NF4 + q];
         v3[q] = p34[cu[u] * NF4 + q];
-        if (act >= 2) {
+        if (act != 0) {
           vs[q] = s4[cu[u] * NF4 + q];
           vh[q] = h4[cu[u] * NF4 + q];
         }
@@ -564,12 +555,9 @@ __global__ void bn_bwd_dx_kernel(
         float xv = to_f32(px2[u].v[j]);
         float g = to_f32(pg2[u].v[j]);
         if (act != 0) {
-          float z = 0.f;
-          if (act >= 2) {
-            z = xv * vs[j / 4].v[j % 4] + vh[j / 4].v[j % 4];
-            if (HAS_RES) z += to_f32(pr2[u].v[j]);
-          }
-          g *= act_grad(to_f32(py2[u].v[j]), z, act);
+          float z = xv * vs[j / 4].v[j % 4] + vh[j / 4].v[j % 4];
+          if (HAS_RES) z += to_f32(pr2[u].v[j]);
+          g *= act_grad(z, act);
         }
         if (HAS_RES) orr.v[j] = from_f32<T>(g);
         ox.v[j] = from_f32<T>(v1[j / 4].v[j % 4] * g + v3[j / 4].v[j % 4] * xv +
@@ -580,16 +568,14 @@ __global__ void bn_bwd_dx_kernel(
     }
     gp += 2 * stride;
     xp += 2 * stride;
-    ypk += 2 * stride;
     rp += 2 * stride;
     oxp += 2 * stride;
     orp += 2 * stride;
   }
   if (i < npacks) {
     const int c0 = c * V;
-    P pg = gp[0], px = xp[0], py, pr;
-    if (act != 0) py = ypk[0];
-    if (HAS_RES && act >= 2) pr = rp[0];
+    P pg = gp[0], px = xp[0], pr;
+    if (HAS_RES && act != 0) pr = rp[0];
     P ox, orr;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -597,12 +583,9 @@ __global__ void bn_bwd_dx_kernel(
       float xv = to_f32(px.v[j]);
       float g = to_f32(pg.v[j]);
       if (act != 0) {
-        float z = 0.f;
-        if (act >= 2) {
-          z = xv * scale[cc] + shift[cc];
-          if (HAS_RES) z += to_f32(pr.v[j]);
-        }
-        g *= act_grad(to_f32(py.v[j]), z, act);
+        float z = xv * scale[cc] + shift[cc];
+        if (HAS_RES) z += to_f32(pr.v[j]);
+        g *= act_grad(z, act);
       }
       if (HAS_RES) orr.v[j] = from_f32<T>(g);
       ox.v[j] = from_f32<T>(P1c[cc] * g + P3c[cc] * xv + P2c[cc]);
@@ -628,13 +611,13 @@ int64_t pick_rows_per_block(int64_t rows, int rows_per_iter) {
 // 90-186 VGPRs (scratch traffic capped the reduce at ~1.5 TB/s).
 template <typename scalar_t>
 void launch_bwd_reduce(const scalar_t* gy, const scalar_t* x,
-                       const scalar_t* y, const scalar_t* resp,
+                       const scalar_t* resp,
                        const float* scale, const float* shift, float* part,
                        int64_t rows, int C, int64_t rpb, int act, int rgrid,
                        hipStream_t stream) {
 #define BR_CASE(HR, A)                                                      \
   hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, HR, A>), dim3(rgrid),  \
-                     dim3(256), 0, stream, gy, x, y, resp, scale, shift,    \
+                     dim3(256), 0, stream, gy, x, resp, scale, shift,       \
                      part, rows, C, rpb)
   if (resp != nullptr) {
     switch (act) {
@@ -656,7 +639,7 @@ void launch_bwd_reduce(const scalar_t* gy, const scalar_t* x,
 
 
 template <typename scalar_t, int V, bool HR>
-void launch_dx(const at::Tensor& gy, const at::Tensor& x, const at::Tensor& y,
+void launch_dx(const at::Tensor& gy, const at::Tensor& x,
                const scalar_t* resp, const at::Tensor& scale,
                const at::Tensor& shift, const at::Tensor& P1,
                const at::Tensor& P2, const at::Tensor& P3, at::Tensor& gx,
@@ -664,8 +647,7 @@ void launch_dx(const at::Tensor& gy, const at::Tensor& x, const at::Tensor& y,
                hipStream_t stream) {
   hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, V, HR>), dim3(grid),
                      dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                     (const scalar_t*)x.data_ptr(),
-                     (const scalar_t*)y.data_ptr(), resp,
+                     (const scalar_t*)x.data_ptr(), resp,
                      scale.data_ptr<float>(), shift.data_ptr<float>(),
                      P1.data_ptr<float>(), P2.data_ptr<float>(),
                      P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,
@@ -697,6 +679,22 @@ std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
 
 }  // namespace
 
+// Column-reduce an [nblocks, width] fp32 partial matrix to [width] — used by
+// SyncBN to turn conv-epilogue partials (F1) into the local [2C] sums it
+// all-reduces across ranks.
+at::Tensor bn_reduce_partials(at::Tensor part) {
+  CHECK_GPU(part);
+  TORCH_CHECK(part.dim() == 2 && part.scalar_type() == at::kFloat,
+              "bn_reduce_partials: [nblocks, width] fp32");
+  const int width = part.size(1);
+  const int nblocks = part.size(0);
+  auto out = at::empty({width}, part.options());
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3((int)ceil_div(width, 64)),
+                     dim3(256), 0, cur_stream(), part.data_ptr<float>(),
+                     out.data_ptr<float>(), width, nblocks);
+  return out;
+}
+
 std::vector<at::Tensor> bn_sums(at::Tensor x) {
   CHECK_GPU(x);
   check_nhwc(x, "x");
@@ -710,11 +708,15 @@ std::vector<at::Tensor> bn_sums(at::Tensor x) {
 }
 
 // One-shot stats: partials -> (mean, rstd, scale, shift) + running update.
+// part_opt: per-block [nblocks, 2C] sum/sumsq partials already produced by
+// the PRODUCING conv's epilogue (F1 of docs/DESIGN_bn_conv_fusion.md) —
+// when given, the bn_sums pass over x is skipped entirely.
 std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
                                  at::Tensor beta,
                                  c10::optional<at::Tensor> rm_opt,
                                  c10::optional<at::Tensor> rv_opt,
-                                 double momentum, double eps, bool training) {
+                                 double momentum, double eps, bool training,
+                                 c10::optional<at::Tensor> part_opt) {
   at::Tensor running_mean = rm_opt.has_value() ? *rm_opt : at::Tensor();
   at::Tensor running_var = rv_opt.has_value() ? *rv_opt : at::Tensor();
   const int C = x.size(1);
@@ -724,7 +726,13 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
   auto scale = at::empty({C}, opts);
   auto shift = at::empty({C}, opts);
   if (training) {
-    auto pg = bn_partials(x);
+    std::pair<at::Tensor, int> pg;
+    if (part_opt.has_value()) {
+      TORCH_CHECK(part_opt->size(1) == 2 * C, "bn_stats: partials width");
+      pg = {*part_opt, (int)part_opt->size(0)};
+    } else {
+      pg = bn_partials(x);
+    }
     const int64_t rows = x.numel() / C;
     const float inv_cnt = 1.f / (float)rows;
     const float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
@@ -784,7 +792,7 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
 
 // Local raw grad-stat sums [2C] = [sum(g), sum(g*x)] (stage 1 + stage 2);
 // used standalone by SyncBN (which all-reduces the result across ranks).
-at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
+at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x,
                         c10::optional<at::Tensor> res, at::Tensor scale,
                         at::Tensor shift, int64_t act) {
   CHECK_GPU(gy);
@@ -807,7 +815,7 @@ at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
     launch_bwd_reduce<scalar_t>(
         (const scalar_t*)gy.data_ptr(), (const scalar_t*)x.data_ptr(),
-        (const scalar_t*)y.data_ptr(), resp, scale.data_ptr<float>(),
+        resp, scale.data_ptr<float>(),
         shift.data_ptr<float>(), part.data_ptr<float>(), rows, C, rpb,
         (int)act, rgrid, stream);
     both = at::empty({2 * C}, fopts);
@@ -819,7 +827,7 @@ at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x, at::Tensor y,
 }
 
 // dx/gw/gb given (possibly cross-rank-reduced) raw sums and the TOTAL count.
-std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
+std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x,
                                      c10::optional<at::Tensor> res,
                                      at::Tensor mean, at::Tensor rstd,
                                      at::Tensor gamma, at::Tensor scale,
@@ -855,11 +863,11 @@ std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
     scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
     if (has_res)
-      launch_dx<scalar_t, V, true>(gy, x, y, resp, scale, shift, P1, P2, P3,
+      launch_dx<scalar_t, V, true>(gy, x, resp, scale, shift, P1, P2, P3,
                                    gx, gresp, npacks, cpacks, (int)act, grid,
                                    stream);
     else
-      launch_dx<scalar_t, V, false>(gy, x, y, resp, scale, shift, P1, P2, P3,
+      launch_dx<scalar_t, V, false>(gy, x, resp, scale, shift, P1, P2, P3,
                                     gx, gresp, npacks, cpacks, (int)act, grid,
                                     stream);
   });
@@ -867,8 +875,8 @@ std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x, at::Tensor y,
 }
 
 // Full backward. scale/shift are the forward's fused coefficients (needed to
-// recompute z for silu/sigmoid activation backward).
-std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
+// recompute z for the activation backward; the y stream is never read).
+std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
                                at::Tensor scale, at::Tensor shift,
@@ -900,7 +908,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
     launch_bwd_reduce<scalar_t>(
         (const scalar_t*)gy.data_ptr(), (const scalar_t*)x.data_ptr(),
-        (const scalar_t*)y.data_ptr(), resp, scale.data_ptr<float>(),
+        resp, scale.data_ptr<float>(),
         shift.data_ptr<float>(), part.data_ptr<float>(), rows, C, rpb,
         (int)act, rgrid, stream);
     const float inv_cnt = 1.f / (float)rows;
@@ -915,11 +923,11 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x, at::Tensor y,
     int grid = grid_1d(npacks, 256);
     scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
     if (has_res)
-      launch_dx<scalar_t, V, true>(gy, x, y, resp, scale, shift, P1, P2, P3,
+      launch_dx<scalar_t, V, true>(gy, x, resp, scale, shift, P1, P2, P3,
                                    gx, gresp, npacks, cpacks, (int)act, grid,
                                    stream);
     else
-      launch_dx<scalar_t, V, false>(gy, x, y, resp, scale, shift, P1, P2, P3,
+      launch_dx<scalar_t, V, false>(gy, x, resp, scale, shift, P1, P2, P3,
                                     gx, gresp, npacks, cpacks, (int)act, grid,
                                     stream);
   });

@@ -92,6 +92,44 @@ DEV_INLINE float block_reduce_max(float v, float* lds) {
 }
 
 // ---------------------------------------------------------------------------
+// BN-conv fusion (F1): per-wave epilogue partial store
+// ---------------------------------------------------------------------------
+// Every conv epilogue restages its 64x64 wave tile through a private fp32
+// slab (16 rows x 68-stride) and stores 16-channel stripes per lane. When the
+// conv's output feeds a BatchNorm, the same epilogue also accumulates
+// per-channel sum / sum-of-squares of the bf16-ROUNDED stored values
+// (bit-identical to what a separate bn_sums pass over y would read) in
+// accs/accq[16] per lane, then calls this helper: transpose-reduce over the
+// 16 er-lanes through the slab and store ONE fp32 partial row slice
+// part[slot_base + col] (+C for sumsq), col = lane, masked to col < kmax.
+// Deterministic: fixed (block, wave) -> slot mapping, no atomics.
+DEV_INLINE void bn_partial_wave_store(float* slab, const float* accs,
+                                      const float* accq,
+                                      float* __restrict__ part,
+                                      int64_t slot_base, int C, int lane,
+                                      int kmax) {
+  const int er = lane >> 2, ec = (lane & 3) << 4;
+  __builtin_amdgcn_wave_barrier();
+#pragma unroll
+  for (int j = 0; j < 16; ++j) slab[er * 68 + ec + j] = accs[j];
+  __builtin_amdgcn_wave_barrier();
+  float s = 0.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) s += slab[r * 68 + lane];
+  __builtin_amdgcn_wave_barrier();
+#pragma unroll
+  for (int j = 0; j < 16; ++j) slab[er * 68 + ec + j] = accq[j];
+  __builtin_amdgcn_wave_barrier();
+  float q = 0.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) q += slab[r * 68 + lane];
+  if (lane < kmax) {
+    part[slot_base + lane] = s;
+    part[slot_base + C + lane] = q;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launch helpers
 // ---------------------------------------------------------------------------
 inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }

@@ -41,8 +41,10 @@ struct ConvParams {
   int oh0, ow0;             // output position base offsets (parity classes)
   int M, nspan, ksteps;     // nspan = ceil(S*Cg/BK), ksteps = R*nspan
   int tiles_m;              // for XCD swizzle
+  float* part;              // EMIT: [tiles_m*2, 2*Kt] BN sum/sumsq partials
 };
 
+template <bool EMIT>
 __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
   const int g = blockIdx.z;
   // XCD-aware swizzle over m-tiles (T1; bijective form)
@@ -171,6 +173,7 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
   const int HoWo = p.Ho * p.Wo;
   const int er = lane >> 2;          // 0..15 row within the 16-row stripe
   const int ec = (lane & 3) << 4;    // 0,16,32,48 col
+  float accs[16] = {}, accq[16] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
     // scatter this mi-stripe's fragments to the slab (conflict-free b32)
@@ -195,6 +198,14 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (EMIT) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const float v = to_f32(u.b[j]);
+          accs[j] += v;
+          accq[j] += v * v;
+        }
+      }
       if (k0 + 16 <= p.Kg) {
         *reinterpret_cast<uint4*>(&p.y[obase + k0]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + k0 + 8]) = u.q[1];
@@ -205,6 +216,14 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
       }
     }
     __builtin_amdgcn_wave_barrier();
+  }
+  if (EMIT) {
+    const int kbase = tile_n * BN + wn * 64;
+    const int kmax = min(64, p.Kg - kbase);
+    bn_partial_wave_store(
+        slab, accs, accq, p.part,
+        (int64_t)(tile_m * 2 + wm) * 2 * p.Kt + g * p.Kg + kbase, p.Kt, lane,
+        kmax);
   }
 }
 
@@ -223,8 +242,10 @@ struct SmallGemmParams {
   __hip_bfloat16* y;
   int C, Kt;
   int64_t M;
+  float* part;  // EMIT: [ceil(M/256)*4, 2*Kt] BN partials
 };
 
+template <bool EMIT>
 __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
     SmallGemmParams p) {
   __shared__ float slab4[4][16 * 68];
@@ -264,6 +285,7 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
     // per-wave LDS-restage epilogue (16-wide bf16 stores; T21)
+    float accs[16] = {}, accq[16] = {};
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -281,12 +303,24 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
 #pragma unroll
         for (int j = 0; j < 16; ++j)
           u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+        if (EMIT) {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            const float v = to_f32(u.b[j]);
+            accs[j] += v;
+            accq[j] += v * v;
+          }
+        }
         __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
         *reinterpret_cast<uint4*>(yp) = u.q[0];
         *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
       }
       __builtin_amdgcn_wave_barrier();
     }
+    if (EMIT)
+      bn_partial_wave_store(slab, accs, accq, p.part,
+                            (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
+                            p.Kt, lane, min(64, p.Kt - k0));
   }
 }
 
@@ -294,7 +328,7 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
 // C <= 64 variant: the whole per-wave A stripe (64 rows x C) is register-
 // cached across k chunks, so x is read exactly once regardless of Kt (the
 // generic kernel re-reads A per 64-wide k chunk, which loses for Kt > 64).
-template <int CS>  // C / 32
+template <int CS, bool EMIT>  // CS = C / 32
 __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
     SmallGemmParams p) {
   __shared__ float slab4[4][16 * 68];
@@ -332,6 +366,7 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr[cs][mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
+    float accs[16] = {}, accq[16] = {};
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -349,12 +384,24 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
 #pragma unroll
         for (int j = 0; j < 16; ++j)
           u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+        if (EMIT) {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            const float v = to_f32(u.b[j]);
+            accs[j] += v;
+            accq[j] += v * v;
+          }
+        }
         __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
         *reinterpret_cast<uint4*>(yp) = u.q[0];
         *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
       }
       __builtin_amdgcn_wave_barrier();
     }
+    if (EMIT)
+      bn_partial_wave_store(slab, accs, accq, p.part,
+                            (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
+                            p.Kt, lane, min(64, p.Kt - k0));
   }
 }
 
@@ -379,12 +426,14 @@ struct SmallConvParams {
   int RSC, RSCp;
   int64_t M;
   unsigned long long magicHoWo, magicWo, magicSC, magicC;
+  float* part;  // EMIT: [ceil(M/256)*4, 2*64] BN partials
 };
 
 DEV_INLINE int magic_div2(int m, unsigned long long magic) {
   return (int)(((unsigned long long)(unsigned)m * magic) >> 47);
 }
 
+template <bool EMIT>
 __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
   __shared__ float slab4[4][16 * 68];
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
@@ -483,6 +532,7 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
   }
   float* slab = slab4[wid];
   const int er = lane >> 2, ec = (lane & 3) << 4;
+  float accs[16] = {}, accq[16] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -500,12 +550,23 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (EMIT) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const float v = to_f32(u.b[j]);
+          accs[j] += v;
+          accq[j] += v * v;
+        }
+      }
       __hip_bfloat16* yp = p.y + m * 64 + ec;
       *reinterpret_cast<uint4*>(yp) = u.q[0];
       *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
     }
     __builtin_amdgcn_wave_barrier();
   }
+  if (EMIT)
+    bn_partial_wave_store(slab, accs, accq, p.part,
+                          (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane, 64);
 }
 
 
@@ -515,7 +576,7 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
 // (OOB -> a 16-byte zero buffer) instead of the loaded value, so no
 // post-load cndmask forces an early waitcnt, and (b) prefetches step i+1's
 // fragments while step i's MFMAs run.
-template <int CSTEPS>  // C / 32
+template <int CSTEPS, bool EMIT>  // CSTEPS = C / 32
 __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
     SmallConvParams p) {
   __shared__ float slab4[4][16 * 68];
@@ -596,6 +657,7 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
   }
   float* slab = slab4[wid];
   const int er = lane >> 2, ec = (lane & 3) << 4;
+  float accs[16] = {}, accq[16] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -613,12 +675,23 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (EMIT) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const float v = to_f32(u.b[j]);
+          accs[j] += v;
+          accq[j] += v * v;
+        }
+      }
       __hip_bfloat16* yp = p.y + m * 64 + ec;
       *reinterpret_cast<uint4*>(yp) = u.q[0];
       *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
     }
     __builtin_amdgcn_wave_barrier();
   }
+  if (EMIT)
+    bn_partial_wave_store(slab, accs, accq, p.part,
+                          (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane, 64);
 }
 
 }  // namespace
@@ -630,7 +703,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
                            int64_t osh = 1, int64_t osw = 1, int64_t oh0 = 0,
-                           int64_t ow0 = 0) {
+                           int64_t ow0 = 0, at::Tensor* part_out = nullptr) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16 only");
   check_nhwc(x, "x");
@@ -655,18 +728,29 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   p.nspan = (S * Cg + BK - 1) / BK;
   p.ksteps = R * p.nspan;
   p.tiles_m = (p.M + BM - 1) / BM;
+  p.part = nullptr;
   dim3 grid(p.tiles_m, (Kg + BN - 1) / BN, groups);
-  hipLaunchKernelGGL(conv_igemm_fwd_kernel, grid, dim3(256), 0, cur_stream(),
-                     p);
+  if (part_out != nullptr) {
+    *part_out = at::empty({(int64_t)p.tiles_m * 2, (int64_t)2 * Kt},
+                          x.options().dtype(at::kFloat));
+    p.part = part_out->data_ptr<float>();
+    hipLaunchKernelGGL(conv_igemm_fwd_kernel<true>, grid, dim3(256), 0,
+                       cur_stream(), p);
+  } else {
+    hipLaunchKernelGGL(conv_igemm_fwd_kernel<false>, grid, dim3(256), 0,
+                       cur_stream(), p);
+  }
   return y;
 }
 
-at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
-                         int64_t ph, int64_t pw, int64_t dh, int64_t dw);
+at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                          int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                          at::Tensor* part_out);
 
-at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
-                      int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                      int64_t groups) {
+static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
+                                  int64_t sw, int64_t ph, int64_t pw,
+                                  int64_t dh, int64_t dw, int64_t groups,
+                                  at::Tensor* part_out) {
   const int N = x.size(0), H = x.size(2), W = x.size(3);
   const int Kt = w.size(0), R = w.size(2), S = w.size(3);
   // v2 (3-slot glds ring) wins when the reduction is deep enough to fill
@@ -678,7 +762,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   const int C_ = x.size(1);
   if (!v2_off && groups == 1 && Kt >= 192 && R * S * C_ >= 512 &&
       x.scalar_type() == at::kBFloat16)
-    return conv2d_fwd_v2(x, w, sh, sw, ph, pw, dh, dw);
+    return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, part_out);
   static const bool small_off = []() {
     const char* e = getenv("DISTRIBUUUU_CONV_SMALL");
     return e && e[0] == '0';
@@ -694,15 +778,29 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     sp.w = (const __hip_bfloat16*)w.data_ptr();
     sp.y = (__hip_bfloat16*)y.data_ptr();
     sp.C = C_; sp.Kt = Kt; sp.M = M;
+    sp.part = nullptr;
     const dim3 sg((int)((M + 255) / 256));
-    if (C_ == 64)
-      hipLaunchKernelGGL(conv_gemm_smallc_kernel<2>, sg, dim3(256), 0,
-                         cur_stream(), sp);
+    if (part_out != nullptr) {
+      *part_out = at::empty({(int64_t)sg.x * 4, (int64_t)2 * Kt},
+                            x.options().dtype(at::kFloat));
+      sp.part = part_out->data_ptr<float>();
+      if (C_ == 64)
+        hipLaunchKernelGGL((conv_gemm_smallc_kernel<2, true>), sg, dim3(256),
+                           0, cur_stream(), sp);
+      else if (C_ == 32)
+        hipLaunchKernelGGL((conv_gemm_smallc_kernel<1, true>), sg, dim3(256),
+                           0, cur_stream(), sp);
+      else
+        hipLaunchKernelGGL(conv_gemm_small_kernel<true>, sg, dim3(256), 0,
+                           cur_stream(), sp);
+    } else if (C_ == 64)
+      hipLaunchKernelGGL((conv_gemm_smallc_kernel<2, false>), sg, dim3(256),
+                         0, cur_stream(), sp);
     else if (C_ == 32)
-      hipLaunchKernelGGL(conv_gemm_smallc_kernel<1>, sg, dim3(256), 0,
-                         cur_stream(), sp);
+      hipLaunchKernelGGL((conv_gemm_smallc_kernel<1, false>), sg, dim3(256),
+                         0, cur_stream(), sp);
     else
-      hipLaunchKernelGGL(conv_gemm_small_kernel, sg, dim3(256), 0,
+      hipLaunchKernelGGL(conv_gemm_small_kernel<false>, sg, dim3(256), 0,
                          cur_stream(), sp);
     return y;
   }
@@ -739,21 +837,55 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     if (!zbuf.defined() || zbuf.device() != x.device())
       zbuf = at::zeros({8}, x.options());
     sp.zbuf = (const __hip_bfloat16*)zbuf.data_ptr();
+    sp.part = nullptr;
     const dim3 skg((int)((sp.M + 255) / 256));
-    if (C_ == 64)
-      hipLaunchKernelGGL(conv_smallk_pipe_kernel<2>, skg, dim3(256), 0,
-                         cur_stream(), sp);
+    if (part_out != nullptr) {
+      *part_out = at::empty({(int64_t)skg.x * 4, 128},
+                            x.options().dtype(at::kFloat));
+      sp.part = part_out->data_ptr<float>();
+      if (C_ == 64)
+        hipLaunchKernelGGL((conv_smallk_pipe_kernel<2, true>), skg, dim3(256),
+                           0, cur_stream(), sp);
+      else if (C_ == 32)
+        hipLaunchKernelGGL((conv_smallk_pipe_kernel<1, true>), skg, dim3(256),
+                           0, cur_stream(), sp);
+      else
+        hipLaunchKernelGGL(conv_smallk_kernel<true>, skg, dim3(256), 0,
+                           cur_stream(), sp);
+    } else if (C_ == 64)
+      hipLaunchKernelGGL((conv_smallk_pipe_kernel<2, false>), skg, dim3(256),
+                         0, cur_stream(), sp);
     else if (C_ == 32)
-      hipLaunchKernelGGL(conv_smallk_pipe_kernel<1>, skg, dim3(256), 0,
-                         cur_stream(), sp);
+      hipLaunchKernelGGL((conv_smallk_pipe_kernel<1, false>), skg, dim3(256),
+                         0, cur_stream(), sp);
     else
-      hipLaunchKernelGGL(conv_smallk_kernel, skg, dim3(256), 0, cur_stream(),
-                         sp);
+      hipLaunchKernelGGL(conv_smallk_kernel<false>, skg, dim3(256), 0,
+                         cur_stream(), sp);
     return y;
   }
   auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups);
+  return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups, 1,
+                         1, 0, 0, part_out);
+}
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                      int64_t groups) {
+  return conv2d_fwd_impl(x, w, sh, sw, ph, pw, dh, dw, groups, nullptr);
+}
+
+// conv forward + BN sum/sumsq partials emitted from the epilogue (F1):
+// returns {y, part} with part a [nblocks, 2*Kt] fp32 tensor consumable by
+// bn_stats(part_opt=...). Stats are over the bf16-rounded stored y, exactly
+// matching what bn_sums(y) would compute.
+std::vector<at::Tensor> conv2d_fwd_bn(at::Tensor x, at::Tensor w, int64_t sh,
+                                      int64_t sw, int64_t ph, int64_t pw,
+                                      int64_t dh, int64_t dw, int64_t groups) {
+  at::Tensor part;
+  auto y = conv2d_fwd_impl(x, w, sh, sw, ph, pw, dh, dw, groups, &part);
+  TORCH_CHECK(part.defined(), "conv2d_fwd_bn: dispatch path without partials");
+  return {y, part};
 }
 
 // NT GEMM through the conv kernel: y[M, N] = a[M, K] @ b[N, K]^T

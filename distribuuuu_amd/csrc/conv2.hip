@@ -31,6 +31,7 @@ struct Conv2Params {
   int Ho, Wo;
   int M, nspan, ksteps;
   int tiles_m;
+  float* part;  // EMIT: [tiles_m*4, 2*K] BN sum/sumsq partials
 };
 
 // st_16x32 swizzle on a byte offset within a tile (1024-B subtiles)
@@ -38,6 +39,7 @@ DEV_INLINE int swz(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
 
 #define WAITVM(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
 
+template <bool EMIT>
 __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   int tile_m = blockIdx.x, tile_n = blockIdx.y;
   {  // XCD-aware bijective remap over m-tiles (T1)
@@ -179,6 +181,7 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   float* slab = reinterpret_cast<float*>(smem) + wid * (16 * 68);
   const int er = lane >> 2;
   const int ec = (lane & 3) << 4;
+  float accs[16] = {}, accq[16] = {};
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -201,6 +204,14 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
+      if (EMIT) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const float v = to_f32(u.b[j]);
+          accs[j] += v;
+          accq[j] += v * v;
+        }
+      }
       if (k0 + 16 <= p.K) {
         *reinterpret_cast<uint4*>(&p.y[obase + ec]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + ec + 8]) = u.q[1];
@@ -211,6 +222,13 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
       }
     }
     __builtin_amdgcn_wave_barrier();
+  }
+  if (EMIT) {
+    const int kbase = tile_n * BN2 + wn * 64;
+    const int kmax = min(64, p.K - kbase);
+    bn_partial_wave_store(slab, accs, accq, p.part,
+                          (int64_t)(tile_m * 4 + wm) * 2 * p.K + kbase, p.K,
+                          lane, kmax);
   }
 }
 
@@ -257,8 +275,9 @@ __global__ void pad_weight_span_kernel(const T* __restrict__ w,
 
 }  // namespace
 
-at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
-                         int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
+at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                          int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                          at::Tensor* part_out) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
   check_nhwc(x, "x");
@@ -324,8 +343,22 @@ at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   p.nspan = SPAN64 / BK2;
   p.ksteps = R * p.nspan;
   p.tiles_m = (p.M + BM2 - 1) / BM2;
+  p.part = nullptr;
   dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2);
-  hipLaunchKernelGGL(conv_igemm_v2_kernel, grid, dim3(512), 0, cur_stream(),
-                     p);
+  if (part_out != nullptr) {
+    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * K},
+                          x.options().dtype(at::kFloat));
+    p.part = part_out->data_ptr<float>();
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<true>, grid, dim3(512), 0,
+                       cur_stream(), p);
+  } else {
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<false>, grid, dim3(512), 0,
+                       cur_stream(), p);
+  }
   return y;
+}
+
+at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
+  return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, nullptr);
 }

@@ -42,10 +42,16 @@ class _HIPConv2d(torch.autograd.Function):
     squeeze widths like 58) run on zero-padded channels — padding C cannot
     change the output, and padded-K rows are sliced off. Without this, such
     shapes fell back to ATen/MIOpen naive kernels (~230 us per tiny SE conv
-    on RegNetY)."""
+    on RegNetY).
+
+    ``emit_part``: the conv's epilogue also emits BN sum/sumsq partials
+    (F1 of docs/DESIGN_bn_conv_fusion.md), returned as a non-differentiable
+    second output — the consuming BatchNorm skips its whole stats pass over
+    y. Enabled lazily: batch_norm_act flags the producing conv's weight the
+    first time it sees its output (see conv2d/batch_norm_act below)."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, dilation, groups):
+    def forward(ctx, x, w, bias, stride, padding, dilation, groups, emit_part):
         x = _cl(x)
         w = _cl(w)
         e = ext()
@@ -59,8 +65,14 @@ class _HIPConv2d(torch.autograd.Function):
             wp = e.pad_channels(w, _pad8(cin))
         if kpad:
             wp = _pad_k(wp, _pad8(kout))
-        y = e.conv2d_fwd(xp, wp, stride[0], stride[1], padding[0], padding[1],
-                         dilation[0], dilation[1], groups)
+        if emit_part and not kpad and bias is None:
+            y, part = e.conv2d_fwd_bn(xp, wp, stride[0], stride[1],
+                                      padding[0], padding[1], dilation[0],
+                                      dilation[1], groups)
+        else:
+            y = e.conv2d_fwd(xp, wp, stride[0], stride[1], padding[0],
+                             padding[1], dilation[0], dilation[1], groups)
+            part = y.new_empty(0)
         if kpad:
             y = _cl(y[:, :kout])
         if bias is not None:
@@ -68,10 +80,11 @@ class _HIPConv2d(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         ctx.conf = (stride, padding, dilation, groups, bias is not None, cin,
                     kout, cpad, kpad)
-        return y
+        ctx.mark_non_differentiable(part)
+        return y, part
 
     @staticmethod
-    def backward(ctx, gy):
+    def backward(ctx, gy, _gpart):
         x, w = ctx.saved_tensors
         (stride, padding, dilation, groups, has_bias, cin, kout, cpad,
          kpad) = ctx.conf
@@ -102,7 +115,7 @@ class _HIPConv2d(torch.autograd.Function):
                 gw = _cl(gw)
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
-        return gx, gw, gb, None, None, None, None
+        return gx, gw, gb, None, None, None, None, None
 
 
 class _HIPDepthwiseConv2d(torch.autograd.Function):
@@ -163,7 +176,17 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
             and use_hip(x, "dwconv_fwd")):
         return _HIPDepthwiseConv2d.apply(x, weight, bias, stride, padding)
     if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
-        return _HIPConv2d.apply(x, weight, bias, stride, padding, dilation, groups)
+        emit = (bias is None and torch.is_grad_enabled()
+                and getattr(weight, "_emit_bn_partials", False))
+        y, part = _HIPConv2d.apply(x, weight, bias, stride, padding, dilation,
+                                   groups, emit)
+        if part.numel():
+            y._bn_partials = part
+        elif bias is None:
+            # discovery hook: if a BatchNorm consumes this output it flags the
+            # weight, and from the next step on the epilogue emits partials
+            y._bn_src_weight = weight
+        return y
     return F.conv2d(x, weight, bias, stride, padding, dilation, groups)
 
 
@@ -171,13 +194,16 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
 # BatchNorm (+residual add + activation) — K6/K8/K20 fused
 # ---------------------------------------------------------------------------
 class _HIPBatchNormAct(torch.autograd.Function):
-    """Training-mode fused BN: one per-channel fp32 sums reduction kernel +
-    one normalize(+add)(+act) elementwise kernel; backward is one grad-stats
-    reduction + one dx kernel. Running stats updated in fp32."""
+    """Training-mode fused BN: stats from the producing conv's epilogue
+    partials when available (F1 — no pass over x at all), else one per-channel
+    fp32 sums reduction kernel; then one normalize(+add)(+act) elementwise
+    kernel. Backward is one grad-stats reduction + one dx kernel, both
+    recomputing the act mask from x (F3a — the y stream is never re-read).
+    Running stats updated in fp32."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, act, residual):
+                momentum, eps, act, residual, part):
         x = _cl(x)
         e = ext()
         act_id = _ACTS[act]
@@ -190,14 +216,14 @@ class _HIPBatchNormAct(torch.autograd.Function):
             copy_back = True
         # one fused kernel: sums -> mean/rstd/scale/shift + running update
         mean, rstd, scale, shift = e.bn_stats(
-            x, gamma, beta, rm, rv, momentum, eps, training)
+            x, gamma, beta, rm, rv, momentum, eps, training, part)
         if copy_back:
             with torch.no_grad():
                 running_mean.copy_(rm)
                 running_var.copy_(rv)
         res = _cl(residual) if residual is not None else None
         y = e.bn_apply_act(x, scale, shift, act_id, res)
-        ctx.save_for_backward(x, gamma, scale, shift, mean, rstd, y,
+        ctx.save_for_backward(x, gamma, scale, shift, mean, rstd,
                               res if res is not None else x.new_empty(0))
         ctx.act_id = act_id
         ctx.training = training
@@ -207,21 +233,28 @@ class _HIPBatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy):
-        x, gamma, scale, shift, mean, rstd, y, res = ctx.saved_tensors
+        x, gamma, scale, shift, mean, rstd, res = ctx.saved_tensors
         e = ext()
         gy = _cl(gy)
         gx, gw, gb, gres = e.bn_bwd(
-            gy, x, y, res if ctx.has_res else None, mean, rstd, gamma,
+            gy, x, res if ctx.has_res else None, mean, rstd, gamma,
             scale, shift, ctx.act_id, ctx.training, ctx.has_res)
         return (gx, gw.to(ctx.w_dtype), gb.to(ctx.w_dtype), None, None, None,
-                None, None, None, gres if ctx.has_res else None)
+                None, None, None, gres if ctx.has_res else None, None)
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
                    momentum=0.1, eps=1e-5, act="none", residual=None):
     if use_hip(x, "bn_sums"):
-        return _HIPBatchNormAct.apply(x, weight, bias, running_mean, running_var,
-                                      training, momentum, eps, act, residual)
+        part = getattr(x, "_bn_partials", None) if training else None
+        if training and part is None:
+            srcw = getattr(x, "_bn_src_weight", None)
+            if srcw is not None:
+                # flag the producing conv: it emits partials from now on
+                srcw._emit_bn_partials = True
+        return _HIPBatchNormAct.apply(x, weight, bias, running_mean,
+                                      running_var, training, momentum, eps,
+                                      act, residual, part)
     y = F.batch_norm(x, running_mean, running_var, weight, bias, training,
                      momentum, eps)
     if residual is not None:

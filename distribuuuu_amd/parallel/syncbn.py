@@ -29,8 +29,17 @@ class _SyncBNFunction(torch.autograd.Function):
         beta = bias.float().contiguous()
         hip = use_hip(x, "bn_sums")
         if hip:
+            part = getattr(x, "_bn_partials", None)
+            srcw = getattr(x, "_bn_src_weight", None)
             x = DF._cl(x)
-            s, ss = ext().bn_sums(x)  # fp32 [C] views of one [2C] tensor
+            if part is not None:
+                # conv-epilogue partials (F1): no pass over x at all
+                both = ext().bn_reduce_partials(part)
+                s, ss = both[:c], both[c:]
+            else:
+                if srcw is not None:
+                    srcw._emit_bn_partials = True
+                s, ss = ext().bn_sums(x)  # fp32 [C] views of one [2C] tensor
         else:
             xf = x.float()
             s = xf.sum(dim=(0, 2, 3))
@@ -63,8 +72,10 @@ class _SyncBNFunction(torch.autograd.Function):
             if res is not None:
                 y = y + res.float()
             y = DF._apply_act(y, act).to(x.dtype)
+        # hip backward recomputes the act mask from x (F3a): y not saved
         ctx.save_for_backward(x, gamma, scale.contiguous(), shift.contiguous(),
-                              mean.contiguous(), rstd.contiguous(), y,
+                              mean.contiguous(), rstd.contiguous(),
+                              y if not hip else x.new_empty(0),
                               res if res is not None else x.new_empty(0))
         ctx.meta = (act, total, residual is not None, process_group, hip,
                     weight.dtype)
@@ -78,11 +89,11 @@ class _SyncBNFunction(torch.autograd.Function):
         if hip:
             gy = DF._cl(gy)
             e = ext()
-            sums = e.bn_bwd_stats(gy, x, y, res if has_res else None,
+            sums = e.bn_bwd_stats(gy, x, res if has_res else None,
                                   scale, shift, act_id)      # raw [2C]
             dist.all_reduce(sums, group=group)
             gx, gw, gb, gres = e.bn_bwd_apply(
-                gy, x, y, res if has_res else None, mean, rstd, gamma,
+                gy, x, res if has_res else None, mean, rstd, gamma,
                 scale, shift, sums, total, act_id, True, has_res)
             return (gx, gw.to(w_dtype), gb.to(w_dtype), None, None, None,
                     None, None, gres if has_res else None, None)

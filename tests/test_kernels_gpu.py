@@ -292,3 +292,126 @@ def test_mhsa_fwd_bwd_vs_torch():
             (got.shape, err.mean().item(), scale_m)
         assert err.max().item() < 0.05 * want.grad.abs().max().item() + 0.3, \
             (got.shape, err.max().item())
+
+
+# ---------------------------------------------------------------------------
+# BN-conv fusion (F1/F3a)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("shape,conv", [
+    # (N, C, H, W, K, R, stride) chosen to hit each conv kernel family
+    ((4, 64, 28, 28, 256, 1, 1), "small"),     # smallc 1x1
+    ((4, 64, 28, 28, 256, 3, 1), "v2"),        # v2 ring 3x3
+    ((4, 96, 28, 28, 96, 3, 2), "v1"),         # v1 tile (K<192 RSC>=512... )
+    ((4, 32, 28, 28, 64, 3, 1), "smallk"),     # smallk pipe K=64
+])
+def test_conv_fwd_bn_partials(shape, conv):
+    """conv2d_fwd_bn's epilogue partials must reduce to exactly bn_sums(y)."""
+    e = _ext()
+    torch.manual_seed(0)
+    n, c, h, w, k, r, s = shape
+    x = _cl(torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16))
+    wt = _cl(torch.randn(k, c, r, r, device="cuda", dtype=torch.bfloat16) * 0.1)
+    pad = r // 2
+    y, part = e.conv2d_fwd_bn(x, wt, s, s, pad, pad, 1, 1, 1)
+    y2 = e.conv2d_fwd(x, wt, s, s, pad, pad, 1, 1, 1)
+    assert torch.equal(y.float(), y2.float())
+    sums = e.bn_reduce_partials(part)
+    s_ref, ss_ref = e.bn_sums(y)
+    assert torch.allclose(sums[:k], s_ref, atol=1e-1, rtol=1e-4)
+    assert torch.allclose(sums[k:], ss_ref, atol=1e-1, rtol=1e-4)
+
+
+def test_conv_fwd_bn_partials_grouped():
+    e = _ext()
+    torch.manual_seed(1)
+    g, cg, kg = 4, 16, 32
+    x = _cl(torch.randn(2, g * cg, 14, 14, device="cuda",
+                        dtype=torch.bfloat16))
+    wt = _cl(torch.randn(g * kg, cg, 3, 3, device="cuda",
+                         dtype=torch.bfloat16) * 0.1)
+    y, part = e.conv2d_fwd_bn(x, wt, 1, 1, 1, 1, 1, 1, g)
+    sums = e.bn_reduce_partials(part)
+    s_ref, ss_ref = e.bn_sums(y)
+    k = g * kg
+    assert torch.allclose(sums[:k], s_ref, atol=1e-1, rtol=1e-4)
+    assert torch.allclose(sums[k:], ss_ref, atol=1e-1, rtol=1e-4)
+
+
+@pytest.mark.parametrize("act,res", [(0, False), (1, False), (1, True),
+                                     (2, False), (3, True)])
+def test_bn_bwd_no_y_stream(act, res):
+    """bn_bwd recomputes the act mask from x (F3a) — compare against the
+    autograd fp32 reference of act(x*scale+shift (+res))."""
+    e = _ext()
+    torch.manual_seed(0)
+    n, c, h, w = 4, 32, 9, 9
+    x = _cl(torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16))
+    rt = _cl(torch.randn_like(x)) if res else None
+    gamma = torch.randn(c, device="cuda") * 0.5 + 1
+    beta = torch.randn(c, device="cuda") * 0.1
+    mean, rstd, scale, shift = e.bn_stats(x, gamma, beta, None, None,
+                                          0.1, 1e-5, True, None)
+    gy = _cl(torch.randn_like(x))
+    gx, gw, gb, gres = e.bn_bwd(gy, x, rt, mean, rstd, gamma, scale, shift,
+                                act, True, res)
+    # fp32 autograd reference
+    xf = x.float().detach().requires_grad_(True)
+    gamf = gamma.detach().requires_grad_(True)
+    betf = beta.detach().requires_grad_(True)
+    mu = xf.mean(dim=(0, 2, 3), keepdim=True)
+    var = xf.var(dim=(0, 2, 3), unbiased=False, keepdim=True)
+    z = (xf - mu) / (var + 1e-5).sqrt() * gamf.reshape(1, -1, 1, 1) \
+        + betf.reshape(1, -1, 1, 1)
+    if res:
+        rf = rt.float().detach().requires_grad_(True)
+        z = z + rf
+    if act == 1:
+        out = F.relu(z)
+    elif act == 2:
+        out = F.silu(z)
+    elif act == 3:
+        out = torch.sigmoid(z)
+    else:
+        out = z
+    out.backward(gy.float())
+    assert torch.allclose(gx.float(), xf.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(gw, gamf.grad, atol=2.0, rtol=2e-2)
+    assert torch.allclose(gb, betf.grad, atol=2.0, rtol=2e-2)
+    if res:
+        assert torch.allclose(gres.float(), rf.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_conv_bn_discovery_hook():
+    """Step 1: BN flags the producing conv's weight; step 2: conv emits
+    partials and BN stats from partials match the unfused stats."""
+    from distribuuuu_amd.ops import functional as DF
+
+    _ext()
+    torch.manual_seed(0)
+    x = _cl(torch.randn(2, 64, 14, 14, device="cuda", dtype=torch.bfloat16))
+    wt = torch.nn.Parameter(
+        _cl(torch.randn(64, 64, 3, 3, device="cuda",
+                        dtype=torch.bfloat16) * 0.1))
+    gamma = torch.nn.Parameter(torch.ones(64, device="cuda"))
+    beta = torch.nn.Parameter(torch.zeros(64, device="cuda"))
+    rm = torch.zeros(64, device="cuda")
+    rv = torch.ones(64, device="cuda")
+
+    def step():
+        y = DF.conv2d(x, wt, None, (1, 1), (1, 1), (1, 1), 1)
+        has_part = getattr(y, "_bn_partials", None) is not None
+        out = DF.batch_norm_act(y, gamma, beta, rm, rv, True, 0.1, 1e-5,
+                                "relu", None)
+        out.float().sum().backward()
+        g = wt.grad.detach().clone()
+        wt.grad = None
+        gamma.grad = None
+        beta.grad = None
+        return has_part, g
+
+    p1, g1 = step()
+    assert not p1  # discovery step: no partials yet
+    assert getattr(wt, "_emit_bn_partials", False)
+    p2, g2 = step()
+    assert p2      # fused step: conv emitted partials
+    assert torch.allclose(g1.float(), g2.float(), atol=1e-3, rtol=1e-3)

@@ -75,7 +75,7 @@ def bench_bn():
         b = torch.zeros(c, device=DEV)
         sc = torch.ones(c, device=DEV)
         sh = torch.zeros(c, device=DEV)
-        t = timeit(lambda: e.bn_bwd(gy, x, y, res, mean, rstd, g, sc, sh, 1,
+        t = timeit(lambda: e.bn_bwd(gy, x, res, mean, rstd, g, sc, sh, 1,
                                     True, True))
         print(f"bn_bwd(all)  {str(shp):22s} {t*1e6:8.1f} us  {8*nbytes/t/1e9:7.0f} GB/s")
 

@@ -57,7 +57,7 @@ def main():
             scale = torch.rand(c, device="cuda")
             shift = torch.rand(c, device="cuda")
             r = torch.randn_like(x) if res else None
-            dt = bench(lambda: e.bn_bwd_stats(gy, x, y, r, scale, shift, 1))
+            dt = bench(lambda: e.bn_bwd_stats(gy, x, r, scale, shift, 1))
             gb = 3 * x.numel() * 2 / 1e9  # act=relu reads gy, x, y
             print(f"  C={c:4d} {h:3d}x{w:<3d} res={res}: {dt*1e6:7.1f}us "
                   f"{gb/dt/1e9*1e9:6.0f} GB/s")
