@@ -654,6 +654,43 @@ void launch_dx(const at::Tensor& gy, const at::Tensor& x,
                      npacks, cpacks, act);
 }
 
+// Stage-A collapse for conv-epilogue partial matrices (F1) with many rows:
+// big 56^2 layers produce ~12k per-wave partial rows, and feeding those to
+// the 16-row-lane finalize kernel directly leaves one ~800-iteration load
+// chain per thread on a near-idle grid. Collapse to 64 rows first with a
+// (width/256 x 64) grid, then finalize.
+__global__ void collapse_partials_kernel(const float* __restrict__ part,
+                                         float* __restrict__ out, int width,
+                                         int nblocks, int rows_per) {
+  const int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= width) return;
+  const int r0 = blockIdx.y * rows_per;
+  const int r1 = min(r0 + rows_per, nblocks);
+  float acc = 0.f;
+  int r = r0;
+  for (; r + 3 < r1; r += 4)
+    acc += (part[(int64_t)r * width + c] +
+            part[(int64_t)(r + 1) * width + c]) +
+           (part[(int64_t)(r + 2) * width + c] +
+            part[(int64_t)(r + 3) * width + c]);
+  for (; r < r1; ++r) acc += part[(int64_t)r * width + c];
+  out[(int64_t)blockIdx.y * width + c] = acc;
+}
+
+at::Tensor maybe_collapse_partials(at::Tensor part) {
+  const int nblocks = part.size(0);
+  if (nblocks <= 512) return part;
+  const int width = part.size(1);
+  constexpr int CROWS = 64;
+  auto out = at::empty({CROWS, width}, part.options());
+  const int rows_per = (int)ceil_div(nblocks, CROWS);
+  dim3 grid((int)ceil_div(width, 256), CROWS);
+  hipLaunchKernelGGL(collapse_partials_kernel, grid, dim3(256), 0,
+                     cur_stream(), part.data_ptr<float>(),
+                     out.data_ptr<float>(), width, nblocks, rows_per);
+  return out;
+}
+
 // stage-1 only helper: per-block partials [grid, 2C]
 std::pair<at::Tensor, int> bn_partials(const at::Tensor& x) {
   const int C = x.size(1);
@@ -686,6 +723,7 @@ at::Tensor bn_reduce_partials(at::Tensor part) {
   CHECK_GPU(part);
   TORCH_CHECK(part.dim() == 2 && part.scalar_type() == at::kFloat,
               "bn_reduce_partials: [nblocks, width] fp32");
+  part = maybe_collapse_partials(part);
   const int width = part.size(1);
   const int nblocks = part.size(0);
   auto out = at::empty({width}, part.options());
@@ -729,7 +767,8 @@ std::vector<at::Tensor> bn_stats(at::Tensor x, at::Tensor gamma,
     std::pair<at::Tensor, int> pg;
     if (part_opt.has_value()) {
       TORCH_CHECK(part_opt->size(1) == 2 * C, "bn_stats: partials width");
-      pg = {*part_opt, (int)part_opt->size(0)};
+      auto part = maybe_collapse_partials(*part_opt);
+      pg = {part, (int)part.size(0)};
     } else {
       pg = bn_partials(x);
     }

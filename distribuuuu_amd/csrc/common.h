@@ -92,37 +92,37 @@ DEV_INLINE float block_reduce_max(float v, float* lds) {
 }
 
 // ---------------------------------------------------------------------------
-// BN-conv fusion (F1): per-wave epilogue partial store
+// BN-conv fusion (F1): per-wave epilogue partial accumulation
 // ---------------------------------------------------------------------------
 // Every conv epilogue restages its 64x64 wave tile through a private fp32
-// slab (16 rows x 68-stride) and stores 16-channel stripes per lane. When the
-// conv's output feeds a BatchNorm, the same epilogue also accumulates
-// per-channel sum / sum-of-squares of the bf16-ROUNDED stored values
-// (bit-identical to what a separate bn_sums pass over y would read) in
-// accs/accq[16] per lane, then calls this helper: transpose-reduce over the
-// 16 er-lanes through the slab and store ONE fp32 partial row slice
-// part[slot_base + col] (+C for sumsq), col = lane, masked to col < kmax.
-// Deterministic: fixed (block, wave) -> slot mapping, no atomics.
-DEV_INLINE void bn_partial_wave_store(float* slab, const float* accs,
-                                      const float* accq,
-                                      float* __restrict__ part,
-                                      int64_t slot_base, int C, int lane,
-                                      int kmax) {
-  const int er = lane >> 2, ec = (lane & 3) << 4;
-  __builtin_amdgcn_wave_barrier();
+// slab (16 rows x 68-stride), one 16-row stripe at a time. When the conv's
+// output feeds a BatchNorm, each lane ALSO reads its own column (col = lane)
+// of the live stripe and accumulates sum / sum-of-squares of the
+// bf16-rounded value — bit-identical to what a separate bn_sums pass over
+// the stored y would compute. Only 2 accumulator VGPRs per lane (a 16-wide
+// per-lane register accumulator cost conv_gemm_smallc a full wave of
+// occupancy: 168 -> 208 VGPRs).
+//
+// Call between the two wave barriers of a stripe (slab contents valid);
+// rows_valid = clamp(M - stripe_row_base, 0, 16) masks M-tail garbage rows.
+DEV_INLINE void bn_partial_col_accum(const float* slab, float& s, float& q,
+                                     int rows_valid, int lane) {
 #pragma unroll
-  for (int j = 0; j < 16; ++j) slab[er * 68 + ec + j] = accs[j];
-  __builtin_amdgcn_wave_barrier();
-  float s = 0.f;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) s += slab[r * 68 + lane];
-  __builtin_amdgcn_wave_barrier();
-#pragma unroll
-  for (int j = 0; j < 16; ++j) slab[er * 68 + ec + j] = accq[j];
-  __builtin_amdgcn_wave_barrier();
-  float q = 0.f;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) q += slab[r * 68 + lane];
+  for (int r = 0; r < 16; ++r) {
+    if (r < rows_valid) {
+      const float v =
+          to_f32(from_f32<__hip_bfloat16>(slab[r * 68 + lane]));
+      s += v;
+      q += v * v;
+    }
+  }
+}
+
+// Final per-wave store: one fp32 partial row slice part[slot_base + col]
+// (+C for sumsq), col = lane, masked to col < kmax. Deterministic: fixed
+// (block, wave) -> slot mapping, no atomics.
+DEV_INLINE void bn_partial_store(float* __restrict__ part, int64_t slot_base,
+                                 int C, int lane, int kmax, float s, float q) {
   if (lane < kmax) {
     part[slot_base + lane] = s;
     part[slot_base + C + lane] = q;

@@ -173,7 +173,7 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
   const int HoWo = p.Ho * p.Wo;
   const int er = lane >> 2;          // 0..15 row within the 16-row stripe
   const int ec = (lane & 3) << 4;    // 0,16,32,48 col
-  float accs[16] = {}, accq[16] = {};
+  float ps = 0.f, pq = 0.f;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
     // scatter this mi-stripe's fragments to the slab (conflict-free b32)
@@ -198,14 +198,6 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (EMIT) {
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const float v = to_f32(u.b[j]);
-          accs[j] += v;
-          accq[j] += v * v;
-        }
-      }
       if (k0 + 16 <= p.Kg) {
         *reinterpret_cast<uint4*>(&p.y[obase + k0]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + k0 + 8]) = u.q[1];
@@ -215,15 +207,18 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
           if (k0 + j < p.Kg) p.y[obase + k0 + j] = u.b[j];
       }
     }
+    if (EMIT) {
+      const int base = tile_m * BM + wm * 64 + mi * 16;
+      bn_partial_col_accum(slab, ps, pq,
+                           (int)min((int64_t)16, (int64_t)p.M - base), lane);
+    }
     __builtin_amdgcn_wave_barrier();
   }
   if (EMIT) {
     const int kbase = tile_n * BN + wn * 64;
-    const int kmax = min(64, p.Kg - kbase);
-    bn_partial_wave_store(
-        slab, accs, accq, p.part,
-        (int64_t)(tile_m * 2 + wm) * 2 * p.Kt + g * p.Kg + kbase, p.Kt, lane,
-        kmax);
+    bn_partial_store(p.part,
+                     (int64_t)(tile_m * 2 + wm) * 2 * p.Kt + g * p.Kg + kbase,
+                     p.Kt, lane, min(64, p.Kg - kbase), ps, pq);
   }
 }
 
@@ -285,7 +280,7 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
     // per-wave LDS-restage epilogue (16-wide bf16 stores; T21)
-    float accs[16] = {}, accq[16] = {};
+    float ps = 0.f, pq = 0.f;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -303,24 +298,19 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
 #pragma unroll
         for (int j = 0; j < 16; ++j)
           u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-        if (EMIT) {
-#pragma unroll
-          for (int j = 0; j < 16; ++j) {
-            const float v = to_f32(u.b[j]);
-            accs[j] += v;
-            accq[j] += v * v;
-          }
-        }
         __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
         *reinterpret_cast<uint4*>(yp) = u.q[0];
         *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
       }
+      if (EMIT)
+        bn_partial_col_accum(
+            slab, ps, pq,
+            (int)min((int64_t)16, p.M - (mbase + mi * 16)), lane);
       __builtin_amdgcn_wave_barrier();
     }
     if (EMIT)
-      bn_partial_wave_store(slab, accs, accq, p.part,
-                            (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
-                            p.Kt, lane, min(64, p.Kt - k0));
+      bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
+                       p.Kt, lane, min(64, p.Kt - k0), ps, pq);
   }
 }
 
@@ -328,8 +318,10 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
 // C <= 64 variant: the whole per-wave A stripe (64 rows x C) is register-
 // cached across k chunks, so x is read exactly once regardless of Kt (the
 // generic kernel re-reads A per 64-wide k chunk, which loses for Kt > 64).
+// min 3 blocks/CU (3 waves/SIMD -> <=170 VGPRs): the EMIT variant's natural
+// allocation is 180, which silently dropped smallc to 2 waves/SIMD
 template <int CS, bool EMIT>  // CS = C / 32
-__global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
+__global__ __launch_bounds__(256, 3) void conv_gemm_smallc_kernel(
     SmallGemmParams p) {
   __shared__ float slab4[4][16 * 68];
   const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
@@ -366,7 +358,7 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr[cs][mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
-    float accs[16] = {}, accq[16] = {};
+    float ps = 0.f, pq = 0.f;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -384,24 +376,19 @@ __global__ __launch_bounds__(256) void conv_gemm_smallc_kernel(
 #pragma unroll
         for (int j = 0; j < 16; ++j)
           u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-        if (EMIT) {
-#pragma unroll
-          for (int j = 0; j < 16; ++j) {
-            const float v = to_f32(u.b[j]);
-            accs[j] += v;
-            accq[j] += v * v;
-          }
-        }
         __hip_bfloat16* yp = p.y + m * p.Kt + k0 + ec;
         *reinterpret_cast<uint4*>(yp) = u.q[0];
         *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
       }
+      if (EMIT)
+        bn_partial_col_accum(
+            slab, ps, pq,
+            (int)min((int64_t)16, p.M - (mbase + mi * 16)), lane);
       __builtin_amdgcn_wave_barrier();
     }
     if (EMIT)
-      bn_partial_wave_store(slab, accs, accq, p.part,
-                            (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
-                            p.Kt, lane, min(64, p.Kt - k0));
+      bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
+                       p.Kt, lane, min(64, p.Kt - k0), ps, pq);
   }
 }
 
@@ -532,7 +519,7 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
   }
   float* slab = slab4[wid];
   const int er = lane >> 2, ec = (lane & 3) << 4;
-  float accs[16] = {}, accq[16] = {};
+  float ps = 0.f, pq = 0.f;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -550,23 +537,19 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (EMIT) {
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const float v = to_f32(u.b[j]);
-          accs[j] += v;
-          accq[j] += v * v;
-        }
-      }
       __hip_bfloat16* yp = p.y + m * 64 + ec;
       *reinterpret_cast<uint4*>(yp) = u.q[0];
       *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
     }
+    if (EMIT)
+      bn_partial_col_accum(slab, ps, pq,
+                           (int)min((int64_t)16, p.M - (mbase + mi * 16)),
+                           lane);
     __builtin_amdgcn_wave_barrier();
   }
   if (EMIT)
-    bn_partial_wave_store(slab, accs, accq, p.part,
-                          (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane, 64);
+    bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane,
+                     64, ps, pq);
 }
 
 
@@ -657,7 +640,7 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
   }
   float* slab = slab4[wid];
   const int er = lane >> 2, ec = (lane & 3) << 4;
-  float accs[16] = {}, accq[16] = {};
+  float ps = 0.f, pq = 0.f;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -675,23 +658,19 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (EMIT) {
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const float v = to_f32(u.b[j]);
-          accs[j] += v;
-          accq[j] += v * v;
-        }
-      }
       __hip_bfloat16* yp = p.y + m * 64 + ec;
       *reinterpret_cast<uint4*>(yp) = u.q[0];
       *reinterpret_cast<uint4*>(yp + 8) = u.q[1];
     }
+    if (EMIT)
+      bn_partial_col_accum(slab, ps, pq,
+                           (int)min((int64_t)16, p.M - (mbase + mi * 16)),
+                           lane);
     __builtin_amdgcn_wave_barrier();
   }
   if (EMIT)
-    bn_partial_wave_store(slab, accs, accq, p.part,
-                          (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane, 64);
+    bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane,
+                     64, ps, pq);
 }
 
 }  // namespace

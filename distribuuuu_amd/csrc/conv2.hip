@@ -181,7 +181,7 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   float* slab = reinterpret_cast<float*>(smem) + wid * (16 * 68);
   const int er = lane >> 2;
   const int ec = (lane & 3) << 4;
-  float accs[16] = {}, accq[16] = {};
+  float ps = 0.f, pq = 0.f;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -204,14 +204,6 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (EMIT) {
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const float v = to_f32(u.b[j]);
-          accs[j] += v;
-          accq[j] += v * v;
-        }
-      }
       if (k0 + 16 <= p.K) {
         *reinterpret_cast<uint4*>(&p.y[obase + ec]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + ec + 8]) = u.q[1];
@@ -221,14 +213,17 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
           if (k0 + j < p.K) p.y[obase + ec + j] = u.b[j];
       }
     }
+    if (EMIT) {
+      const int base = tile_m * BM2 + wm * 64 + mi * 16;
+      bn_partial_col_accum(slab, ps, pq,
+                           (int)min((int64_t)16, (int64_t)p.M - base), lane);
+    }
     __builtin_amdgcn_wave_barrier();
   }
   if (EMIT) {
     const int kbase = tile_n * BN2 + wn * 64;
-    const int kmax = min(64, p.K - kbase);
-    bn_partial_wave_store(slab, accs, accq, p.part,
-                          (int64_t)(tile_m * 4 + wm) * 2 * p.K + kbase, p.K,
-                          lane, kmax);
+    bn_partial_store(p.part, (int64_t)(tile_m * 4 + wm) * 2 * p.K + kbase,
+                     p.K, lane, min(64, p.K - kbase), ps, pq);
   }
 }
 
