@@ -237,84 +237,77 @@ __global__ void bn_eval_prep_kernel(const float* __restrict__ running_mean,
 }
 
 // ---- fused apply: y = act(x*scale[c] + shift[c] (+ res)) -------------------
-// Channel-aware grid-stride loops are written with per-thread pack POINTERS
-// and an INCREMENTAL channel-pack index: a naive `p[i + u*stride]` +
-// `i % cpacks` body compiles to ~786 instructions per 4-pack iteration
-// (64-bit address muls + a float-division-trick modulo) and caps at
-// ~1.1 TB/s; the grid stride is uniform, so c advances by a precomputed
-// (stride % cpacks) with a conditional wrap instead, and scale/shift load as
-// float4s.
-template <typename T, int V, bool HAS_RES>
-__global__ void bn_apply_kernel(const T* __restrict__ x,
-                                const float* __restrict__ scale,
-                                const float* __restrict__ shift,
-                                const T* __restrict__ res, T* __restrict__ y,
-                                int64_t npacks, int cpacks, int act) {
+// Row-walk structure (same as bn_bwd_reduce/dx): each thread owns a fixed
+// channel pack so scale/shift load ONCE per column, and rows advance with
+// pointer increments (no per-iteration modulo or table gathers — the
+// previous channel-incrementing grid-stride form re-gathered the float4
+// tables every pack).
+template <typename T, bool HAS_RES>
+__global__ __launch_bounds__(256, 2) void bn_apply_kernel(
+    const T* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, const T* __restrict__ res,
+    T* __restrict__ y, int64_t rows, int C, int64_t rows_per_block, int act) {
+  constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
-  using F4 = Pack<float, 4>;
-  constexpr int NF4 = V / 4;  // float4s of params per pack (bf16: 2, fp32: 1)
-  const F4* s4 = reinterpret_cast<const F4*>(scale);
-  const F4* h4 = reinterpret_cast<const F4*>(shift);
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, rows);
   const P* xp = reinterpret_cast<const P*>(x);
   const P* rp = reinterpret_cast<const P*>(res);
   P* yp = reinterpret_cast<P*>(y);
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  const int dstep = (int)(stride % cpacks);
-  int c = (int)(i % cpacks);  // the ONE division this thread does
-  xp += i;
-  yp += i;
-  if (HAS_RES) rp += i;
-  for (; i + 3 * stride < npacks; i += 4 * stride) {
-    P px4[4], pr4[4];
-    int cu[4];
-    int cc = c;
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      px4[u] = xp[u * stride];
-      if (HAS_RES) pr4[u] = rp[u * stride];
-      cu[u] = cc;
-      cc += dstep;
-      if (cc >= cpacks) cc -= cpacks;
-    }
-    c = cc;
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      F4 sv[NF4], hv[NF4];
-#pragma unroll
-      for (int q = 0; q < NF4; ++q) {
-        sv[q] = s4[cu[u] * NF4 + q];
-        hv[q] = h4[cu[u] * NF4 + q];
-      }
-#pragma unroll
-      for (int j = 0; j < V; ++j) {
-        float z = to_f32(px4[u].v[j]) * sv[j / 4].v[j % 4] + hv[j / 4].v[j % 4];
-        if (HAS_RES) z += to_f32(pr4[u].v[j]);
-        px4[u].v[j] = from_f32<T>(act_apply(z, act));
-      }
-      yp[u * stride] = px4[u];
-    }
-    xp += 4 * stride;
-    yp += 4 * stride;
-    if (HAS_RES) rp += 4 * stride;
-  }
-  for (; i < npacks; i += stride) {
-    const int c0 = c * V;
-    P px = xp[0];
-    P pr;
-    if (HAS_RES) pr = rp[0];
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sc[V], sh[V];
 #pragma unroll
     for (int j = 0; j < V; ++j) {
-      float z = to_f32(px.v[j]) * scale[c0 + j] + shift[c0 + j];
-      if (HAS_RES) z += to_f32(pr.v[j]);
-      px.v[j] = from_f32<T>(act_apply(z, act));
+      sc[j] = scale[cp * V + j];
+      sh[j] = shift[cp * V + j];
     }
-    yp[0] = px;
-    xp += stride;
-    yp += stride;
-    if (HAS_RES) rp += stride;
-    c += dstep;
-    if (c >= cpacks) c -= cpacks;
+    if (!active) continue;
+    int64_t row = row0 + rl;
+    const int64_t rstep = (int64_t)nrl * cpacks;
+    const P* xq = xp + row * cpacks + cp;
+    const P* rq = rp + row * cpacks + cp;
+    P* yq = yp + row * cpacks + cp;
+    for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+      P px4[4], pr4[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        px4[u] = xq[u * rstep];
+        if (HAS_RES) pr4[u] = rq[u * rstep];
+      }
+      xq += 4 * rstep;
+      rq += 4 * rstep;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float z = to_f32(px4[u].v[j]) * sc[j] + sh[j];
+          if (HAS_RES) z += to_f32(pr4[u].v[j]);
+          px4[u].v[j] = from_f32<T>(act_apply(z, act));
+        }
+        yq[u * rstep] = px4[u];
+      }
+      yq += 4 * rstep;
+    }
+    for (; row < row1; row += nrl) {
+      P px = xq[0], pr;
+      if (HAS_RES) pr = rq[0];
+      xq += rstep;
+      rq += rstep;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float z = to_f32(px.v[j]) * sc[j] + sh[j];
+        if (HAS_RES) z += to_f32(pr.v[j]);
+        px.v[j] = from_f32<T>(act_apply(z, act));
+      }
+      yq[0] = px;
+      yq += rstep;
+    }
   }
 }
 
@@ -494,104 +487,110 @@ __global__ void bn_bwd_finalize_kernel(
 
 // ---- bwd dx: gx = P1*g + P3*x + P2 (+ gres = g) ----------------------------
 // act mask recomputed from z = x*scale+shift(+res); no y stream (F3a).
-template <typename T, int V, bool HAS_RES>
-__global__ void bn_bwd_dx_kernel(
+// Row-walk structure (same as bn_bwd_reduce): each thread owns a fixed
+// channel pack, so the 3-5 per-channel coefficient tables are loaded ONCE
+// per column instead of per grid-stride iteration — the flat grid-stride
+// form re-gathered them every pack and cost ~0.5 ms/step on ResNet-50.
+template <typename T, bool HAS_RES, int ACT>
+__global__ __launch_bounds__(256, 2) void bn_bwd_dx_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, const float* __restrict__ P1c,
     const float* __restrict__ P2c, const float* __restrict__ P3c,
-    T* __restrict__ gx, T* __restrict__ gres, int64_t npacks, int cpacks,
-    int act) {
+    T* __restrict__ gx, T* __restrict__ gres, int64_t rows, int C,
+    int64_t rows_per_block) {
+  constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
-  using F4 = Pack<float, 4>;
-  constexpr int NF4 = V / 4;
-  const F4* p14 = reinterpret_cast<const F4*>(P1c);
-  const F4* p24 = reinterpret_cast<const F4*>(P2c);
-  const F4* p34 = reinterpret_cast<const F4*>(P3c);
-  const F4* s4 = reinterpret_cast<const F4*>(scale);
-  const F4* h4 = reinterpret_cast<const F4*>(shift);
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, rows);
   const P* gp = reinterpret_cast<const P*>(gy);
   const P* xp = reinterpret_cast<const P*>(x);
   const P* rp = reinterpret_cast<const P*>(res);
   P* oxp = reinterpret_cast<P*>(gx);
   P* orp = reinterpret_cast<P*>(gres);
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  int64_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  const int dstep = (int)(stride % cpacks);
-  int c = (int)(i % cpacks);
-  gp += i;
-  xp += i;
-  rp += i;
-  oxp += i;
-  orp += i;
-  for (; i + stride < npacks; i += 2 * stride) {
-    P pg2[2], px2[2], pr2[2];
-    int cu[2] = {c, c + dstep};
-    if (cu[1] >= cpacks) cu[1] -= cpacks;
-    c = cu[1] + dstep;
-    if (c >= cpacks) c -= cpacks;
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sc[V], sh[V], a1[V], a2[V], a3[V];
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      pg2[u] = gp[u * stride];
-      px2[u] = xp[u * stride];
-      if (HAS_RES && act != 0) pr2[u] = rp[u * stride];
-    }
-#pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      F4 v1[NF4], v2[NF4], v3[NF4], vs[NF4], vh[NF4];
-#pragma unroll
-      for (int q = 0; q < NF4; ++q) {
-        v1[q] = p14[cu[u] * NF4 + q];
-        v2[q] = p24[cu[u] * NF4 + q];
-        v3[q] = p34[cu[u] * NF4 + q];
-        if (act != 0) {
-          vs[q] = s4[cu[u] * NF4 + q];
-          vh[q] = h4[cu[u] * NF4 + q];
-        }
+    for (int j = 0; j < V; ++j) {
+      const int cc = cp * V + j;
+      a1[j] = P1c[cc];
+      a2[j] = P2c[cc];
+      a3[j] = P3c[cc];
+      if (ACT != 0) {
+        sc[j] = scale[cc];
+        sh[j] = shift[cc];
       }
+    }
+    if (!active) continue;
+    int64_t row = row0 + rl;
+    const int64_t rstep = (int64_t)nrl * cpacks;
+    const P* gq = gp + row * cpacks + cp;
+    const P* xq = xp + row * cpacks + cp;
+    const P* rq = rp + row * cpacks + cp;
+    P* oq = oxp + row * cpacks + cp;
+    P* orq = orp + row * cpacks + cp;
+    for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+      P pg4[4], px4[4], pr4[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        pg4[u] = gq[u * rstep];
+        px4[u] = xq[u * rstep];
+        if (HAS_RES && ACT != 0) pr4[u] = rq[u * rstep];
+      }
+      gq += 4 * rstep;
+      xq += 4 * rstep;
+      rq += 4 * rstep;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        P ox, orr;
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float xv = to_f32(px4[u].v[j]);
+          float g = to_f32(pg4[u].v[j]);
+          if (ACT != 0) {
+            float z = xv * sc[j] + sh[j];
+            if (HAS_RES) z += to_f32(pr4[u].v[j]);
+            g *= act_grad(z, ACT);
+          }
+          if (HAS_RES) orr.v[j] = from_f32<T>(g);
+          ox.v[j] = from_f32<T>(a1[j] * g + a3[j] * xv + a2[j]);
+        }
+        oq[u * rstep] = ox;
+        if (HAS_RES) orq[u * rstep] = orr;
+      }
+      oq += 4 * rstep;
+      orq += 4 * rstep;
+    }
+    for (; row < row1; row += nrl) {
+      P pg = gq[0], px = xq[0], pr;
+      if (HAS_RES && ACT != 0) pr = rq[0];
+      gq += rstep;
+      xq += rstep;
+      rq += rstep;
       P ox, orr;
 #pragma unroll
       for (int j = 0; j < V; ++j) {
-        float xv = to_f32(px2[u].v[j]);
-        float g = to_f32(pg2[u].v[j]);
-        if (act != 0) {
-          float z = xv * vs[j / 4].v[j % 4] + vh[j / 4].v[j % 4];
-          if (HAS_RES) z += to_f32(pr2[u].v[j]);
-          g *= act_grad(z, act);
+        float xv = to_f32(px.v[j]);
+        float g = to_f32(pg.v[j]);
+        if (ACT != 0) {
+          float z = xv * sc[j] + sh[j];
+          if (HAS_RES) z += to_f32(pr.v[j]);
+          g *= act_grad(z, ACT);
         }
         if (HAS_RES) orr.v[j] = from_f32<T>(g);
-        ox.v[j] = from_f32<T>(v1[j / 4].v[j % 4] * g + v3[j / 4].v[j % 4] * xv +
-                              v2[j / 4].v[j % 4]);
+        ox.v[j] = from_f32<T>(a1[j] * g + a3[j] * xv + a2[j]);
       }
-      oxp[u * stride] = ox;
-      if (HAS_RES) orp[u * stride] = orr;
+      oq[0] = ox;
+      if (HAS_RES) orq[0] = orr;
+      oq += rstep;
+      orq += rstep;
     }
-    gp += 2 * stride;
-    xp += 2 * stride;
-    rp += 2 * stride;
-    oxp += 2 * stride;
-    orp += 2 * stride;
-  }
-  if (i < npacks) {
-    const int c0 = c * V;
-    P pg = gp[0], px = xp[0], pr;
-    if (HAS_RES && act != 0) pr = rp[0];
-    P ox, orr;
-#pragma unroll
-    for (int j = 0; j < V; ++j) {
-      const int cc = c0 + j;
-      float xv = to_f32(px.v[j]);
-      float g = to_f32(pg.v[j]);
-      if (act != 0) {
-        float z = xv * scale[cc] + shift[cc];
-        if (HAS_RES) z += to_f32(pr.v[j]);
-        g *= act_grad(z, act);
-      }
-      if (HAS_RES) orr.v[j] = from_f32<T>(g);
-      ox.v[j] = from_f32<T>(P1c[cc] * g + P3c[cc] * xv + P2c[cc]);
-    }
-    oxp[0] = ox;
-    if (HAS_RES) orp[0] = orr;
   }
 }
 
@@ -638,20 +637,37 @@ void launch_bwd_reduce(const scalar_t* gy, const scalar_t* x,
 }
 
 
-template <typename scalar_t, int V, bool HR>
+template <typename scalar_t>
 void launch_dx(const at::Tensor& gy, const at::Tensor& x,
                const scalar_t* resp, const at::Tensor& scale,
                const at::Tensor& shift, const at::Tensor& P1,
                const at::Tensor& P2, const at::Tensor& P3, at::Tensor& gx,
-               scalar_t* gresp, int64_t npacks, int cpacks, int act, int grid,
-               hipStream_t stream) {
-  hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, V, HR>), dim3(grid),
-                     dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),
-                     (const scalar_t*)x.data_ptr(), resp,
-                     scale.data_ptr<float>(), shift.data_ptr<float>(),
-                     P1.data_ptr<float>(), P2.data_ptr<float>(),
-                     P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,
-                     npacks, cpacks, act);
+               scalar_t* gresp, int64_t rows, int C, int64_t rpb, int act,
+               int rgrid, hipStream_t stream) {
+#define DX_CASE(HR, A)                                                       \
+  hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, HR, A>), dim3(rgrid),       \
+                     dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),   \
+                     (const scalar_t*)x.data_ptr(), resp,                    \
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),       \
+                     P1.data_ptr<float>(), P2.data_ptr<float>(),             \
+                     P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,  \
+                     rows, C, rpb)
+  if (resp != nullptr) {
+    switch (act) {
+      case 0: DX_CASE(true, 0); break;
+      case 1: DX_CASE(true, 1); break;
+      case 2: DX_CASE(true, 2); break;
+      default: DX_CASE(true, 3); break;
+    }
+  } else {
+    switch (act) {
+      case 0: DX_CASE(false, 0); break;
+      case 1: DX_CASE(false, 1); break;
+      case 2: DX_CASE(false, 2); break;
+      default: DX_CASE(false, 3); break;
+    }
+  }
+#undef DX_CASE
 }
 
 // Stage-A collapse for conv-epilogue partial matrices (F1) with many rows:
@@ -809,22 +825,24 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_apply_act", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
-    int64_t npacks = x.numel() / V;
-    int cpacks = C / V;
-    int grid = grid_1d(npacks, 256);
+    const int64_t rows = x.numel() / C;
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
     auto stream = cur_stream();
     if (has_res)
-      hipLaunchKernelGGL((bn_apply_kernel<scalar_t, V, true>), dim3(grid),
+      hipLaunchKernelGGL((bn_apply_kernel<scalar_t, true>), dim3(rgrid),
                          dim3(256), 0, stream, (const scalar_t*)x.data_ptr(),
                          scale.data_ptr<float>(), shift.data_ptr<float>(),
                          (const scalar_t*)res->data_ptr(),
-                         (scalar_t*)y.data_ptr(), npacks, cpacks, (int)act);
+                         (scalar_t*)y.data_ptr(), rows, C, rpb, (int)act);
     else
-      hipLaunchKernelGGL((bn_apply_kernel<scalar_t, V, false>), dim3(grid),
+      hipLaunchKernelGGL((bn_apply_kernel<scalar_t, false>), dim3(rgrid),
                          dim3(256), 0, stream, (const scalar_t*)x.data_ptr(),
                          scale.data_ptr<float>(), shift.data_ptr<float>(),
                          (const scalar_t*)nullptr, (scalar_t*)y.data_ptr(),
-                         npacks, cpacks, (int)act);
+                         rows, C, rpb, (int)act);
   });
   return y;
 }
@@ -874,19 +892,21 @@ std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x,
                                      double total_count, int64_t act,
                                      bool training, bool need_gres) {
   const int C = x.size(1);
+  const int64_t rows = x.numel() / C;
   auto fopts = x.options().dtype(at::kFloat);
   const bool has_res = res.has_value();
   auto gx = at::empty_like(x);
-  auto gres = need_gres ? at::empty_like(x) : at::Tensor();
+  auto gres = has_res ? at::empty_like(x) : at::Tensor();
   auto gw = at::empty({C}, fopts);
   auto gb = at::empty({C}, fopts);
   auto P1 = at::empty({C}, fopts);
   auto P2 = at::empty({C}, fopts);
   auto P3 = at::empty({C}, fopts);
+  (void)need_gres;
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd_apply", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
     auto stream = cur_stream();
-    const int cpacks = C / V;
     const float inv_cnt = 1.f / (float)total_count;
     // sums is a flat [2C] row => treat as a 1-block partial
     hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
@@ -896,19 +916,15 @@ std::vector<at::Tensor> bn_bwd_apply(at::Tensor gy, at::Tensor x,
                        gb.data_ptr<float>(), P1.data_ptr<float>(),
                        P2.data_ptr<float>(), P3.data_ptr<float>(), C, inv_cnt,
                        training ? 1 : 0);
-    int64_t npacks = x.numel() / V;
-    int grid = grid_1d(npacks, 256);
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
-    scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
-    if (has_res)
-      launch_dx<scalar_t, V, true>(gy, x, resp, scale, shift, P1, P2, P3,
-                                   gx, gresp, npacks, cpacks, (int)act, grid,
-                                   stream);
-    else
-      launch_dx<scalar_t, V, false>(gy, x, resp, scale, shift, P1, P2, P3,
-                                    gx, gresp, npacks, cpacks, (int)act, grid,
-                                    stream);
+    scalar_t* gresp = has_res ? (scalar_t*)gres.data_ptr() : nullptr;
+    launch_dx<scalar_t>(gy, x, resp, scale, shift, P1, P2, P3, gx, gresp,
+                        rows, C, rpb, (int)act, rgrid, stream);
   });
   return {gx, gw, gb, gres};
 }
@@ -927,12 +943,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
   auto fopts = x.options().dtype(at::kFloat);
   const bool has_res = res.has_value();
   auto gx = at::empty_like(x);
-  auto gres = need_gres ? at::empty_like(x) : at::Tensor();
+  auto gres = has_res ? at::empty_like(x) : at::Tensor();
   auto gw = at::empty({C}, fopts);
   auto gb = at::empty({C}, fopts);
   auto P1 = at::empty({C}, fopts);
   auto P2 = at::empty({C}, fopts);
   auto P3 = at::empty({C}, fopts);
+  (void)need_gres;
 
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
@@ -958,17 +975,9 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                        gb.data_ptr<float>(), P1.data_ptr<float>(),
                        P2.data_ptr<float>(), P3.data_ptr<float>(), C, inv_cnt,
                        training ? 1 : 0);
-    int64_t npacks = x.numel() / V;
-    int grid = grid_1d(npacks, 256);
-    scalar_t* gresp = need_gres ? (scalar_t*)gres.data_ptr() : nullptr;
-    if (has_res)
-      launch_dx<scalar_t, V, true>(gy, x, resp, scale, shift, P1, P2, P3,
-                                   gx, gresp, npacks, cpacks, (int)act, grid,
-                                   stream);
-    else
-      launch_dx<scalar_t, V, false>(gy, x, resp, scale, shift, P1, P2, P3,
-                                    gx, gresp, npacks, cpacks, (int)act, grid,
-                                    stream);
+    scalar_t* gresp = has_res ? (scalar_t*)gres.data_ptr() : nullptr;
+    launch_dx<scalar_t>(gy, x, resp, scale, shift, P1, P2, P3, gx, gresp,
+                        rows, C, rpb, (int)act, rgrid, stream);
   });
   return {gx, gw, gb, gres};
 }
