@@ -7,6 +7,8 @@ dropout and gradient-checkpointed (memory_efficient) bottleneck recompute
 (reference densenet.py:81-110) via torch.utils.checkpoint.
 """
 
+import re
+
 import torch
 import torch.nn as nn
 import torch.utils.checkpoint as cp
@@ -97,22 +99,54 @@ class DenseNet(nn.Module):
         return self.classifier(out)
 
 
-def _densenet(growth_rate, block_config, num_init_features, **kwargs):
-    kwargs.pop("pretrained", None)
-    return DenseNet(growth_rate, block_config, num_init_features, **kwargs)
+model_urls = {
+    "densenet121": "https://download.pytorch.org/models/densenet121-a639ec97.pth",
+    "densenet169": "https://download.pytorch.org/models/densenet169-b2777c0a.pth",
+    "densenet201": "https://download.pytorch.org/models/densenet201-c1103571.pth",
+    "densenet161": "https://download.pytorch.org/models/densenet161-8d451a50.pth",
+}
+
+# legacy torchvision DenseNet checkpoints use dotted sub-layer names
+# ('denselayer1.norm.1.weight'); current module names drop the dot
+_LEGACY_KEY = re.compile(
+    r"^(.*denselayer\d+\.(?:norm|relu|conv))\."
+    r"((?:[12])\.(?:weight|bias|running_mean|running_var))$")
+
+
+def remap_legacy_densenet_keys(state_dict):
+    """norm.1 -> norm1 etc., matching the reference's pretrained-key remap
+    (reference densenet.py:266-282)."""
+    for key in list(state_dict.keys()):
+        res = _LEGACY_KEY.match(key)
+        if res:
+            state_dict[res.group(1) + res.group(2)] = state_dict.pop(key)
+    return state_dict
+
+
+def _densenet(arch, growth_rate, block_config, num_init_features,
+              pretrained=False, progress=True, **kwargs):
+    model = DenseNet(growth_rate, block_config, num_init_features, **kwargs)
+    if pretrained:
+        from torch.hub import load_state_dict_from_url
+
+        state_dict = load_state_dict_from_url(model_urls[arch],
+                                              progress=progress,
+                                              map_location="cpu")
+        model.load_state_dict(remap_legacy_densenet_keys(state_dict))
+    return model
 
 
 def densenet121(**kw):
-    return _densenet(32, (6, 12, 24, 16), 64, **kw)
+    return _densenet("densenet121", 32, (6, 12, 24, 16), 64, **kw)
 
 
 def densenet161(**kw):
-    return _densenet(48, (6, 12, 36, 24), 96, **kw)
+    return _densenet("densenet161", 48, (6, 12, 36, 24), 96, **kw)
 
 
 def densenet169(**kw):
-    return _densenet(32, (6, 12, 32, 32), 64, **kw)
+    return _densenet("densenet169", 32, (6, 12, 32, 32), 64, **kw)
 
 
 def densenet201(**kw):
-    return _densenet(32, (6, 12, 48, 32), 64, **kw)
+    return _densenet("densenet201", 32, (6, 12, 48, 32), 64, **kw)

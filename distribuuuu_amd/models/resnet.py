@@ -123,48 +123,75 @@ class ResNet(nn.Module):
         return self.fc(x.flatten(1))
 
 
-def _resnet(block, layers, **kwargs):
-    kwargs.pop("pretrained", None)  # no network in this environment
-    return ResNet(block, layers, **kwargs)
+# Published torchvision weight URLs (the reference loads the same files,
+# resnet.py:23-33 there); the state_dict layout of these models matches
+# torchvision key-for-key, so the checkpoints load strict.
+model_urls = {
+    "resnet18": "https://download.pytorch.org/models/resnet18-5c106cde.pth",
+    "resnet34": "https://download.pytorch.org/models/resnet34-333f7ec4.pth",
+    "resnet50": "https://download.pytorch.org/models/resnet50-19c8e357.pth",
+    "resnet101": "https://download.pytorch.org/models/resnet101-5d3b4d8f.pth",
+    "resnet152": "https://download.pytorch.org/models/resnet152-b121ed2d.pth",
+    "resnext50_32x4d":
+        "https://download.pytorch.org/models/resnext50_32x4d-7cdf4587.pth",
+    "resnext101_32x8d":
+        "https://download.pytorch.org/models/resnext101_32x8d-8ba56ff5.pth",
+    "wide_resnet50_2":
+        "https://download.pytorch.org/models/wide_resnet50_2-95faca4d.pth",
+    "wide_resnet101_2":
+        "https://download.pytorch.org/models/wide_resnet101_2-32ee1156.pth",
+}
+
+
+def _resnet(arch, block, layers, pretrained=False, progress=True, **kwargs):
+    model = ResNet(block, layers, **kwargs)
+    if pretrained:
+        from torch.hub import load_state_dict_from_url
+
+        state_dict = load_state_dict_from_url(model_urls[arch],
+                                              progress=progress,
+                                              map_location="cpu")
+        model.load_state_dict(state_dict)
+    return model
 
 
 def resnet18(**kw):
-    return _resnet(BasicBlock, [2, 2, 2, 2], **kw)
+    return _resnet("resnet18", BasicBlock, [2, 2, 2, 2], **kw)
 
 
 def resnet34(**kw):
-    return _resnet(BasicBlock, [3, 4, 6, 3], **kw)
+    return _resnet("resnet34", BasicBlock, [3, 4, 6, 3], **kw)
 
 
 def resnet50(**kw):
-    return _resnet(Bottleneck, [3, 4, 6, 3], **kw)
+    return _resnet("resnet50", Bottleneck, [3, 4, 6, 3], **kw)
 
 
 def resnet101(**kw):
-    return _resnet(Bottleneck, [3, 4, 23, 3], **kw)
+    return _resnet("resnet101", Bottleneck, [3, 4, 23, 3], **kw)
 
 
 def resnet152(**kw):
-    return _resnet(Bottleneck, [3, 8, 36, 3], **kw)
+    return _resnet("resnet152", Bottleneck, [3, 8, 36, 3], **kw)
 
 
 def resnext50_32x4d(**kw):
     kw.setdefault("groups", 32)
     kw.setdefault("width_per_group", 4)
-    return _resnet(Bottleneck, [3, 4, 6, 3], **kw)
+    return _resnet("resnext50_32x4d", Bottleneck, [3, 4, 6, 3], **kw)
 
 
 def resnext101_32x8d(**kw):
     kw.setdefault("groups", 32)
     kw.setdefault("width_per_group", 8)
-    return _resnet(Bottleneck, [3, 4, 23, 3], **kw)
+    return _resnet("resnext101_32x8d", Bottleneck, [3, 4, 23, 3], **kw)
 
 
 def wide_resnet50_2(**kw):
     kw.setdefault("width_per_group", 128)
-    return _resnet(Bottleneck, [3, 4, 6, 3], **kw)
+    return _resnet("wide_resnet50_2", Bottleneck, [3, 4, 6, 3], **kw)
 
 
 def wide_resnet101_2(**kw):
     kw.setdefault("width_per_group", 128)
-    return _resnet(Bottleneck, [3, 4, 23, 3], **kw)
+    return _resnet("wide_resnet101_2", Bottleneck, [3, 4, 23, 3], **kw)

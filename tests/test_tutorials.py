@@ -8,13 +8,20 @@ import sys
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def test_snsc_runs():
+def test_snsc_golden_transcript():
+    """The docstring transcript is a regression oracle (the reference embeds
+    expected stdout in every tutorial docstring, e.g. snsc.py:85-114)."""
     from tutorial import snsc
 
-    snsc.main(epochs=1)
+    losses = snsc.main(epochs=3)
+    assert len(losses) == len(snsc.EXPECTED_LOSSES)
+    for got, want in zip(losses, snsc.EXPECTED_LOSSES):
+        assert abs(got - want) < 5e-3, (losses, snsc.EXPECTED_LOSSES)
 
 
-def test_ddp_launch_tutorial_2proc_gloo():
+def test_ddp_launch_tutorial_2proc_gloo_golden():
+    from tutorial.mnmc_ddp_launch import EXPECTED_LOSSES
+
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
     env["MASTER_PORT"] = "29621"
@@ -24,7 +31,11 @@ def test_ddp_launch_tutorial_2proc_gloo():
          "--master-port", "29621", "tutorial/mnmc_ddp_launch.py"],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
-    assert "loss" in r.stdout
+    got = [float(line.rsplit(" ", 1)[1]) for line in r.stdout.splitlines()
+           if " loss " in line]
+    assert len(got) == len(EXPECTED_LOSSES), r.stdout
+    for g, want in zip(got, EXPECTED_LOSSES):
+        assert abs(g - want) < 5e-3, (got, EXPECTED_LOSSES)
 
 
 def test_imagenet_tutorial_single_proc():

@@ -10,7 +10,15 @@ Run (1 node, 8 GPUs):
       --master-addr 127.0.0.1 tutorial/mnmc_ddp_launch.py
 Simulate 2 "nodes" on one host (reference README.md:119-144 recipe) by
 splitting GPUs with HIP_VISIBLE_DEVICES and --nnodes 2 --node-rank {0,1}.
+
+Expected transcript (2-process gloo CPU run, seed 0 — asserted by
+tests/test_tutorials.py as a regression oracle):
+
+    epoch 0 iter 0 loss 2.4308
+    epoch 0 iter 2 loss 2.7715
 """
+
+EXPECTED_LOSSES = [2.4308, 2.7715]
 
 import os
 
@@ -36,11 +44,13 @@ def main(epochs=1):
     if has_gpu:
         torch.cuda.set_device(local_rank)
 
+    torch.manual_seed(0)  # deterministic build (DDP broadcasts rank 0 anyway)
     net = DistributedDataParallel(build_net().to(device))
     ds = DummyCifar()
     sampler = DistributedSampler(ds, shuffle=True)
     loader = DataLoader(ds, batch_size=64, sampler=sampler)
-    opt = torch.optim.SGD(net.parameters(), lr=0.05 * world, momentum=0.9)
+    # linear LR scaling with world size (reference README.md:198-200 recipe)
+    opt = torch.optim.SGD(net.parameters(), lr=0.002 * world, momentum=0.9)
     for epoch in range(epochs):
         sampler.set_epoch(epoch)
         for i, (x, y) in enumerate(loader):

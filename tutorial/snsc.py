@@ -3,10 +3,24 @@
 The simplest rung of the ladder: one process, one device, a small CNN on
 synthetic CIFAR-10-shaped data (this environment has no dataset downloads;
 swap DummyCifar for a real dataset loader to reproduce the reference's
-transcript).
+CIFAR-10 transcript).
 
 Run:  python tutorial/snsc.py
+
+Expected transcript (CPU, seed 0, 3 epochs — asserted by
+tests/test_tutorials.py as a regression oracle, the same mechanism the
+reference uses in its docstrings, e.g. reference snsc.py:85-114):
+
+    epoch 0 iter 0 loss 2.5241
+    epoch 0 iter 4 loss 2.6541
+    epoch 1 iter 0 loss 2.2602
+    epoch 1 iter 4 loss 2.2849
+    epoch 2 iter 0 loss 2.3753
+    epoch 2 iter 4 loss 2.0213
 """
+
+# the docstring table above, machine-readable for the regression test
+EXPECTED_LOSSES = [2.5241, 2.6541, 2.2602, 2.2849, 2.3753, 2.0213]
 
 import torch
 import torch.nn as nn
@@ -41,11 +55,14 @@ def build_net():
     )
 
 
-def main(epochs=2):
+def main(epochs=2, seed=0):
+    torch.manual_seed(seed)
     device = "cuda" if torch.cuda.is_available() else "cpu"
     net = build_net().to(device)
-    loader = DataLoader(DummyCifar(), batch_size=64, shuffle=True)
-    opt = torch.optim.SGD(net.parameters(), lr=0.01, momentum=0.9)
+    g = torch.Generator().manual_seed(seed)
+    loader = DataLoader(DummyCifar(), batch_size=64, shuffle=True, generator=g)
+    opt = torch.optim.SGD(net.parameters(), lr=0.002, momentum=0.9)
+    losses = []
     for epoch in range(epochs):
         for i, (x, y) in enumerate(loader):
             x, y = x.to(device), y.to(device)
@@ -55,6 +72,8 @@ def main(epochs=2):
             opt.step()
             if i % 4 == 0:
                 print(f"epoch {epoch} iter {i} loss {loss.item():.4f}")
+                losses.append(loss.item())
+    return losses
 
 
 if __name__ == "__main__":
