@@ -678,13 +678,24 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
+// fp32 kernel path (conv_fp32.hip)
+at::Tensor conv2d_fwd_into_fp32(at::Tensor x, at::Tensor w, at::Tensor y,
+                                int64_t Ho, int64_t Wo, int64_t sh, int64_t sw,
+                                int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                                int64_t groups, int64_t osh, int64_t osw,
+                                int64_t oh0, int64_t ow0,
+                                at::Tensor* part_out);
+
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
                            int64_t osh = 1, int64_t osw = 1, int64_t oh0 = 0,
                            int64_t ow0 = 0, at::Tensor* part_out = nullptr) {
   CHECK_GPU(x);
-  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16 only");
+  if (x.scalar_type() == at::kFloat)
+    return conv2d_fwd_into_fp32(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw,
+                                groups, osh, osw, oh0, ow0, part_out);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16/fp32 only");
   check_nhwc(x, "x");
   check_nhwc(w, "w");
   const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);

@@ -659,11 +659,17 @@ __global__ void cast_acc_kernel(const float* __restrict__ acc,
 
 }  // namespace
 
+at::Tensor conv2d_wgrad_fp32(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
+                             int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                             int64_t dh, int64_t dw, int64_t groups);
+
 at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                         int64_t sh, int64_t sw, int64_t ph, int64_t pw,
                         int64_t dh, int64_t dw, int64_t groups) {
   CHECK_GPU(gy);
-  TORCH_CHECK(gy.scalar_type() == at::kBFloat16, "wgrad: bf16 only");
+  if (gy.scalar_type() == at::kFloat)
+    return conv2d_wgrad_fp32(gy, x, R, S, sh, sw, ph, pw, dh, dw, groups);
+  TORCH_CHECK(gy.scalar_type() == at::kBFloat16, "wgrad: bf16/fp32 only");
   check_nhwc(gy, "gy");
   check_nhwc(x, "x");
   const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);

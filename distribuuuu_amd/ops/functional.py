@@ -155,8 +155,8 @@ def _is_depthwise(x, weight, groups, dilation):
 
 
 def _hip_conv_ok(x, weight, groups):
-    if x.dtype != torch.bfloat16:
-        fallback_warn("conv2d", f"dtype {x.dtype} (only bf16 has MFMA kernels)")
+    if x.dtype not in (torch.bfloat16, torch.float32):
+        fallback_warn("conv2d", f"dtype {x.dtype} (bf16/fp32 MFMA kernels)")
         return False
     if groups == 1:
         return True  # any C/K: the pad-channels path covers %8 misfits
@@ -410,7 +410,8 @@ class _HIPLinear(torch.autograd.Function):
 
 
 def linear(x, weight, bias=None):
-    if (use_hip(x, "gemm_nt") and x.dtype == torch.bfloat16 and x.dim() == 2
+    if (use_hip(x, "gemm_nt")
+            and x.dtype in (torch.bfloat16, torch.float32) and x.dim() == 2
             and x.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
         return _HIPLinear.apply(x, weight, bias)
     if x.is_cuda:
