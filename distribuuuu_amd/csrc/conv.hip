@@ -735,7 +735,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
 
 at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                           int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                          at::Tensor* part_out);
+                          int64_t groups, at::Tensor* part_out);
 
 static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
                                   int64_t sw, int64_t ph, int64_t pw,
@@ -750,9 +750,14 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     return e && e[0] == '0';
   }();
   const int C_ = x.size(1);
-  if (!v2_off && groups == 1 && Kt >= 192 && R * S * C_ >= 512 &&
-      x.scalar_type() == at::kBFloat16)
-    return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, part_out);
+  // per-group geometry decides the ring gate: deep-enough reduction fills
+  // the 3-slot pipeline (RegNetY's 232-wide groups qualify: Kg=232,
+  // R*S*Cg=2088)
+  const int v2Kg = Kt / (int)groups;
+  const int v2Cg = C_ / (int)groups;
+  if (!v2_off && v2Kg >= 192 && (int64_t)R * S * v2Cg >= 512 &&
+      v2Cg % 8 == 0 && x.scalar_type() == at::kBFloat16)
+    return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, groups, part_out);
   static const bool small_off = []() {
     const char* e = getenv("DISTRIBUUUU_CONV_SMALL");
     return e && e[0] == '0';
@@ -1065,11 +1070,12 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
     return gx;
   }
   const int Kt_ = gy.size(1);
-  if (sh == 1 && sw == 1 && groups == 1 && dh * (R - 1) == 2 * ph &&
-      dw * (S - 1) == 2 * pw) {
-    // same-size conv: the plain fwd path (incl. the v2 ring kernel) applies
+  if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
+      dw * (S - 1) == 2 * pw && (groups == 1 || Kt_ % (8 * groups) == 0)) {
+    // same-size conv: the plain fwd path (incl. the v2 ring kernel) applies;
+    // grouped too — wt is [Ct, Kg, R, S], a valid grouped conv weight
     return conv2d_fwd(gy, wt, 1, 1, dh * (R - 1) - ph, dw * (S - 1) - pw,
-                      dh, dw, 1);
+                      dh, dw, groups);
   }
   (void)Kt_;
   auto gyd = (sh == 1 && sw == 1) ? gy : dilate_nhwc(gy, sh, sw);

@@ -22,16 +22,17 @@ constexpr int BUF_B = ATILE_B + BTILE_B;        // 48 KB per ring slot
 constexpr int NBUF = 3;                         // 144 KB ring
 
 struct Conv2Params {
-  const __hip_bfloat16* x;  // padded [N, Hp, Wp, C] (+slack)
-  const __hip_bfloat16* w;  // span-padded [K, R, SPAN64]
-  __hip_bfloat16* y;        // [N, Ho, Wo, K]
-  int N, Hp, Wp, C, K;
+  const __hip_bfloat16* x;  // padded [N, Hp, Wp, Ct] (+slack)
+  const __hip_bfloat16* w;  // span-padded [Kt, R, SPAN64]
+  __hip_bfloat16* y;        // [N, Ho, Wo, Kt]
+  int N, Hp, Wp, C, K;      // C/K are PER-GROUP (Cg/Kg)
+  int Ct, Kt;               // totals (group stride in x/w/y)
   int R, SPAN64, Cg, S;
   int sh, sw, dh, dw;
   int Ho, Wo;
   int M, nspan, ksteps;
   int tiles_m;
-  float* part;  // EMIT: [tiles_m*4, 2*K] BN sum/sumsq partials
+  float* part;  // EMIT: [tiles_m*4, 2*Kt] BN sum/sumsq partials
 };
 
 // st_16x32 swizzle on a byte offset within a tile (1024-B subtiles)
@@ -41,6 +42,7 @@ DEV_INLINE int swz(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
 
 template <bool EMIT>
 __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
+  const int g = blockIdx.z;
   int tile_m = blockIdx.x, tile_n = blockIdx.y;
   {  // XCD-aware bijective remap over m-tiles (T1)
     const int nwg = p.tiles_m;
@@ -81,8 +83,8 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
     const int sl = s ^ (((s >> 5) & 1) << 1);
     b_sl[it] = sl;
     int k = tile_n * BN2 + (sl >> 3);
-    if (k >= p.K) k = 0;  // clamp
-    b_base[it] = k * p.R * p.SPAN64 + (sl & 7) * 8;
+    if (k >= p.K) k = 0;  // clamp (per-group Kg tail)
+    b_base[it] = (g * p.K + k) * p.R * p.SPAN64 + (sl & 7) * 8;
   }
 
   auto stage = [&](int buf, int ks) {
@@ -98,7 +100,8 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
       const __hip_bfloat16* src =
           p.x +
           ((int64_t)a_n[it] * p.Hp * p.Wp +
-           (a_hwbase[it] + r * p.dh * p.Wp + s_ * p.dw)) * p.C + c;
+           (a_hwbase[it] + r * p.dh * p.Wp + s_ * p.dw)) * p.Ct +
+          g * p.Cg + c;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)src,
           (__attribute__((address_space(3))) uint32_t*)(base + it * 8192),
@@ -194,8 +197,8 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
     if (m < p.M) {
       const int n = m / HoWo;
       const int rem = m - n * HoWo;
-      const int64_t obase = ((int64_t)n * HoWo + rem) * p.K + tile_n * BN2 +
-                            wn * 64;
+      const int64_t obase = ((int64_t)n * HoWo + rem) * p.Kt + g * p.K +
+                            tile_n * BN2 + wn * 64;
       const int k0 = tile_n * BN2 + wn * 64 + ec;
       union {
         __hip_bfloat16 b[16];
@@ -222,8 +225,9 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   }
   if (EMIT) {
     const int kbase = tile_n * BN2 + wn * 64;
-    bn_partial_store(p.part, (int64_t)(tile_m * 4 + wm) * 2 * p.K + kbase,
-                     p.K, lane, min(64, p.K - kbase), ps, pq);
+    bn_partial_store(p.part,
+                     (int64_t)(tile_m * 4 + wm) * 2 * p.Kt + g * p.K + kbase,
+                     p.Kt, lane, min(64, p.K - kbase), ps, pq);
   }
 }
 
@@ -272,17 +276,18 @@ __global__ void pad_weight_span_kernel(const T* __restrict__ w,
 
 at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                           int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                          at::Tensor* part_out) {
+                          int64_t groups, at::Tensor* part_out) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
   check_nhwc(x, "x");
   check_nhwc(w, "w");
-  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  const int K = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
-  TORCH_CHECK(Cg == C, "v2 is groups==1 only");
+  const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(Cg * groups == Ct, "v2: channel/group mismatch");
+  const int C = Cg, K = Kt / groups;
   const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
-  const int SC = S * C;
+  const int SC = S * Cg;
   const int SPAN64 = (SC + BK2 - 1) / BK2 * BK2;
 
   // padded input (+ slack for span-tail over-read)
@@ -291,46 +296,47 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   if (ph > 0 || pw > 0) {
     Hp = H + 2 * ph;
     Wp = W + 2 * pw;
-    auto xp = at::empty({(int64_t)N * Hp * Wp * C + SPAN64 + 64},
+    auto xp = at::empty({(int64_t)N * Hp * Wp * Ct + SPAN64 + 64},
                         x.options());
-    int64_t total = (int64_t)N * Hp * Wp * C;
+    int64_t total = (int64_t)N * Hp * Wp * Ct;
     // span-tail reads of the LAST pixel land in the slack: it must be finite
     // (tail A values multiply zero-padded B weights, but Inf*0 = NaN)
     xp.narrow(0, total, SPAN64 + 64).zero_();
     hipLaunchKernelGGL((pad_image_kernel<__hip_bfloat16>),
-                       dim3((Wp * C / 8 + 255) / 256, N * Hp), dim3(256), 0,
+                       dim3((Wp * Ct / 8 + 255) / 256, N * Hp), dim3(256), 0,
                        cur_stream(),
                        (const __hip_bfloat16*)x.data_ptr(),
-                       (__hip_bfloat16*)xp.data_ptr(), N, H, W, C, Hp, Wp, ph,
+                       (__hip_bfloat16*)xp.data_ptr(), N, H, W, Ct, Hp, Wp, ph,
                        pw);
     xin = xp;
   } else if (SPAN64 != SC) {
     // slack for span-tail reads past the logical end
-    auto xp = at::empty({(int64_t)N * H * W * C + SPAN64 + 64}, x.options());
+    auto xp = at::empty({(int64_t)N * H * W * Ct + SPAN64 + 64}, x.options());
     // copy in PHYSICAL (NHWC) order; zero the slack (read by span tails)
-    xp.narrow(0, 0, (int64_t)N * H * W * C)
+    xp.narrow(0, 0, (int64_t)N * H * W * Ct)
         .copy_(x.permute({0, 2, 3, 1}).reshape({-1}));
-    xp.narrow(0, (int64_t)N * H * W * C, SPAN64 + 64).zero_();
+    xp.narrow(0, (int64_t)N * H * W * Ct, SPAN64 + 64).zero_();
     xin = xp;
   }  // else: no padding, no span tail -> use x directly
   // span-padded weight
   at::Tensor wp = w;
   if (SPAN64 != SC) {
-    wp = at::empty({(int64_t)K * R * SPAN64}, w.options());
-    int64_t total = (int64_t)K * R * SPAN64;
+    wp = at::empty({(int64_t)Kt * R * SPAN64}, w.options());
+    int64_t total = (int64_t)Kt * R * SPAN64;
     hipLaunchKernelGGL((pad_weight_span_kernel<__hip_bfloat16>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
                        (const __hip_bfloat16*)w.data_ptr(),
-                       (__hip_bfloat16*)wp.data_ptr(), K, R, SC, SPAN64);
+                       (__hip_bfloat16*)wp.data_ptr(), Kt, R, SC, SPAN64);
   }
 
-  auto y = at::empty({N, K, Ho, Wo},
+  auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   Conv2Params p;
   p.x = (const __hip_bfloat16*)xin.data_ptr();
   p.w = (const __hip_bfloat16*)wp.data_ptr();
   p.y = (__hip_bfloat16*)y.data_ptr();
   p.N = N; p.Hp = Hp; p.Wp = Wp; p.C = C; p.K = K;
+  p.Ct = Ct; p.Kt = Kt;
   p.R = R; p.SPAN64 = SPAN64; p.Cg = C; p.S = S;
   p.sh = sh; p.sw = sw; p.dh = dh; p.dw = dw;
   p.Ho = Ho; p.Wo = Wo;
@@ -339,9 +345,9 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   p.ksteps = R * p.nspan;
   p.tiles_m = (p.M + BM2 - 1) / BM2;
   p.part = nullptr;
-  dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2);
+  dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2, groups);
   if (part_out != nullptr) {
-    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * K},
+    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
     hipLaunchKernelGGL(conv_igemm_v2_kernel<true>, grid, dim3(512), 0,
@@ -355,5 +361,5 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
 
 at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                          int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
-  return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, nullptr);
+  return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, 1, nullptr);
 }

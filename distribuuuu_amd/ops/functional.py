@@ -169,12 +169,38 @@ def _hip_conv_ok(x, weight, groups):
     return True
 
 
+def _pack_narrow_groups(weight, groups, f):
+    """Block-diagonal packing for sub-8-wide groups (ResNeXt-50's 4-channel
+    stage-1 groups): f neighbouring groups of width cg merge into one compute
+    group of width f*cg with zeros off the diagonal, putting the conv on the
+    MFMA path at f x the MACs (still far ahead of a naive fallback).
+    Differentiable: autograd routes the packed grads back to the original
+    weight and discards the zero blocks' grads (docs/DESIGN_grouped_conv.md)."""
+    kt, cg, r, s = weight.shape
+    kg = kt // groups
+    w6 = weight.reshape(groups // f, f, kg, cg, r, s)
+    z = weight.new_zeros(groups // f, 1, kg, cg, r, s)
+    rows = []
+    for i in range(f):
+        blocks = [w6[:, i:i + 1] if j == i else z for j in range(f)]
+        rows.append(torch.cat(blocks, dim=3))
+    return torch.cat(rows, dim=1).reshape(kt, f * cg, r, s)
+
+
 def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
            groups=1):
     stride, padding, dilation = tuple(stride), tuple(padding), tuple(dilation)
     if (x.dtype == torch.bfloat16 and _is_depthwise(x, weight, groups, dilation)
             and use_hip(x, "dwconv_fwd")):
         return _HIPDepthwiseConv2d.apply(x, weight, bias, stride, padding)
+    if (x.is_cuda and groups > 1 and weight.shape[1] < 8
+            and 8 % weight.shape[1] == 0
+            and x.dtype in (torch.bfloat16, torch.float32)):
+        f = 8 // weight.shape[1]
+        kg = weight.shape[0] // groups
+        if groups % f == 0 and (kg * f) % 8 == 0 and use_hip(x, "conv2d_fwd"):
+            wp = _pack_narrow_groups(_cl(weight), groups, f)
+            return conv2d(x, wp, bias, stride, padding, dilation, groups // f)
     if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
         emit = (bias is None and torch.is_grad_enabled()
                 and getattr(weight, "_emit_bn_partials", False))

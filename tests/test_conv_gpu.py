@@ -59,6 +59,8 @@ CONV_CASES = [
     (1, 64, 23, 19, 72, 3, 2, 1, 1),   # odd sizes
     (2, 64, 28, 28, 64, 3, 1, 1, 4),   # grouped
     (2, 232, 28, 28, 232, 3, 2, 1, 1),  # regnety width
+    (2, 464, 14, 14, 464, 3, 1, 1, 2),  # regnety grouped -> v2 ring
+    (2, 464, 14, 14, 464, 3, 2, 1, 2),  # grouped stride-2 (v1 + parity dgrad)
 ]
 
 
@@ -343,3 +345,61 @@ def test_resnet18_fp32_native_step():
     assert torch.isfinite(loss)
     for p_ in net.parameters():
         assert p_.grad is not None and torch.isfinite(p_.grad).all()
+
+
+# ---------------------------------------------------------------------------
+# Narrow-group packing (ResNeXt 32x4d stage 1) — docs/DESIGN_grouped_conv.md
+# ---------------------------------------------------------------------------
+def test_narrow_group_packed_conv_fwd_bwd():
+    """groups=32 with 4-wide channels pack 2:1 into 8-wide block-diagonal
+    groups; results must match ATen's grouped conv, fwd and both grads."""
+    from distribuuuu_amd.ops import functional as DF
+
+    _ext()
+    torch.manual_seed(3)
+    n, c, h, w_, k, g = 2, 128, 14, 14, 128, 32
+    x = _cl(torch.randn(n, c, h, w_, device="cuda",
+                        dtype=torch.bfloat16)).requires_grad_(True)
+    wt = (torch.randn(k, c // g, 3, 3, device="cuda", dtype=torch.bfloat16)
+          * 0.2).requires_grad_(True)
+    y = DF.conv2d(x, wt, None, (1, 1), (1, 1), (1, 1), g)
+    gy = _cl(torch.randn_like(y))
+    y.backward(gy)
+    xr = x.detach().float().requires_grad_(True)
+    wr = wt.detach().float().requires_grad_(True)
+    yr = F.conv2d(xr, wr, None, 1, 1, 1, g)
+    yr.backward(gy.float())
+    for got, ref, tol in ((y.float(), yr, 2e-2),
+                          (x.grad.float(), xr.grad, 2e-2),
+                          (wt.grad.float(), wr.grad, 5e-2)):
+        err = (got - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err < tol * max(scale, 1.0), (err, scale)
+
+
+def test_resnext50_step_no_aten_conv_fallback():
+    """The resnext50_32x4d train step must touch zero ATen conv kernels:
+    every fallback emits a one-time (op, reason) warning — assert none fired
+    for conv2d (VERDICT round-1 item #3 done-criterion)."""
+    from distribuuuu_amd import models
+    from distribuuuu_amd.ops import functional as DF
+    from distribuuuu_amd.ops import dispatch
+
+    _ext()
+    dispatch._WARNED.clear()
+    torch.manual_seed(0)
+    net = models.build_model("resnext50_32x4d", num_classes=10)
+    net = net.to("cuda").to(torch.bfloat16)
+    for m in net.modules():
+        if hasattr(m, "running_mean") and m.running_mean is not None:
+            m.float()
+    net = net.to(memory_format=torch.channels_last)
+    net.train()
+    x = _cl(torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16))
+    yt = torch.randint(0, 10, (4,), device="cuda")
+    loss = DF.cross_entropy(net(x).float(), yt)
+    loss.backward()
+    assert torch.isfinite(loss)
+    conv_warns = [w for w in dispatch._WARNED
+                  if isinstance(w, tuple) and w[0] == "conv2d"]
+    assert not conv_warns, conv_warns
