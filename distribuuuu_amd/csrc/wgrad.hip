@@ -486,12 +486,15 @@ __global__ __launch_bounds__(256) void conv_wgrad_ring128_kernel(
   // staging: subtile s = it*4 + wid (it<4) per operand; within a subtile
   // lane l fills bytes l*16 = element (m = l>>1, col8 = (l&1)*8) — the glds
   // lane-linear image IS the packed [32 m][16 col] subtile layout.
-  int s_mloc[4], s_col[4];
+  int s_mloc[4], s_col[4], s_kcol[4];
 #pragma unroll
   for (int it = 0; it < 4; ++it) {
     const int sub = it * 4 + wid;
     s_mloc[it] = (sub >> 3) * 32 + (lane >> 1);
     s_col[it] = (sub & 7) * 16 + (lane & 1) * 8;
+    int kc = ktile * R128_BM + s_col[it];
+    if (kc >= p.Kg) kc = 0;  // Kg tail: its output row is discarded
+    s_kcol[it] = kc;
   }
   const int SCg = p.S * p.Cg;
   int b_r[4], b_s[4], b_c[4];
@@ -514,7 +517,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_ring128_kernel(
     for (int it = 0; it < 4; ++it) {
       const int m = m0 + ks * WBK + s_mloc[it];  // < M (M%64==0)
       const __hip_bfloat16* asrc =
-          p.gy + (int64_t)m * p.Kt + g * p.Kg + ktile * R128_BM + s_col[it];
+          p.gy + (int64_t)m * p.Kt + g * p.Kg + s_kcol[it];
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)asrc,
           (__attribute__((address_space(3))) uint32_t*)(base + it * 4096), 16,
@@ -614,6 +617,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_ring128_kernel(
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int k = ktile * R128_BM + wm * 64 + mi * 16 + kq * 4 + rr;
+      if (k >= p.Kg) continue;  // Kg tail (grouped shapes)
       const int64_t rowbase = (int64_t)(g * p.Kg + k) * p.RSC;
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
@@ -730,14 +734,19 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
 
   // 128x128 ring kernel where it measured faster (tools/probes/convmap.py):
   // 1x1 convolutions (no pad pass, B columns are plain channels) whose
-  // launch has enough tiles+chunks to fill the 256 CUs. 3x3 shapes keep the
-  // 64x64 kernel: the pad pass plus tap-spanning B columns cost more than
-  // the bigger tile saves.
-  const int64_t blocks64 = (Kg / 128) *
+  // launch has enough tiles+chunks to fill the 256 CUs, and GROUPED deep
+  // 3x3s (RegNetY 232-wide groups: the 64x64 kernel runs out of math to
+  // cover its staging there too — Kg tails are clamped/dropped). Dense 3x3
+  // keeps the 64x64 kernel: pad pass + tap-spanning B columns cost more
+  // than the bigger tile saves.
+  const int64_t blocks64 = (((int64_t)Kg + 127) / 128) *
                            (((int64_t)R * S * Cg + 127) / 128) *
                            ((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK));
-  if (R == 1 && S == 1 && Kg % 128 == 0 && Cg >= 128 && M64 % 64 == 0 &&
-      Cg % 8 == 0 && blocks64 * groups >= 320) {
+  const bool dense1x1 = R == 1 && S == 1 && Kg % 128 == 0 && Cg >= 128;
+  const bool grouped_deep = groups > 1 && Kg >= 96 &&
+                            (int64_t)R * S * Cg >= 512;
+  if ((dense1x1 || grouped_deep) && M64 % 64 == 0 && Cg % 8 == 0 &&
+      blocks64 * groups >= 320) {
     at::Tensor xin = x;
     int Hp = H, Wp = W;
     if (ph > 0 || pw > 0) {
@@ -761,7 +770,7 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
     q.Ho = Ho; q.Wo = Wo;
     q.M = (int)M64;
     q.RSC = R * S * Cg;
-    q.ktiles = Kg / R128_BM;
+    q.ktiles = (Kg + R128_BM - 1) / R128_BM;
     q.ntiles = (q.RSC + R128_BN - 1) / R128_BN;
     q.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
     q.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
