@@ -743,10 +743,18 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                            (((int64_t)R * S * Cg + 127) / 128) *
                            ((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK));
   const bool dense1x1 = R == 1 && S == 1 && Kg % 128 == 0 && Cg >= 128;
+  // M >= 8192: measured crossover (tools/probes/wgrad_group_ab.py) — the
+  // 7x7 grouped shapes stay on the 64x64 kernel (85 vs 118 us)
   const bool grouped_deep = groups > 1 && Kg >= 96 &&
-                            (int64_t)R * S * Cg >= 512;
-  if ((dense1x1 || grouped_deep) && M64 % 64 == 0 && Cg % 8 == 0 &&
-      blocks64 * groups >= 320) {
+                            (int64_t)R * S * Cg >= 512 && M64 >= 8192;
+  // DISTRIBUUUU_WGRAD_128: 1 forces the ring128 route (A/B probes, tests of
+  // the Kg-tail path on small shapes), 0 disables it; re-read per call
+  const char* f128 = getenv("DISTRIBUUUU_WGRAD_128");
+  const int force128 = f128 ? atoi(f128) : -1;
+  const bool gate_ok = (dense1x1 || grouped_deep) &&
+                       blocks64 * groups >= 320;
+  if (force128 != 0 && M64 % 64 == 0 && Cg % 8 == 0 &&
+      (gate_ok || force128 == 1)) {
     at::Tensor xin = x;
     int Hp = H, Wp = W;
     if (ph > 0 || pw > 0) {

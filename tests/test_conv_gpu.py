@@ -403,3 +403,21 @@ def test_resnext50_step_no_aten_conv_fallback():
     conv_warns = [w for w in dispatch._WARNED
                   if isinstance(w, tuple) and w[0] == "conv2d"]
     assert not conv_warns, conv_warns
+
+
+def test_wgrad_ring128_grouped_kg_tail(monkeypatch):
+    """Force the ring128 route on a grouped shape with a Kg tail (232 % 128
+    != 0) and M % 64 == 0 — the production gate needs huge launches, so the
+    tail path is exercised here via the force env."""
+    e = _ext()
+    monkeypatch.setenv("DISTRIBUUUU_WGRAD_128", "1")
+    torch.manual_seed(8)
+    n, c, h, g = 4, 464, 16, 2  # M = 4*16*16 = 1024 (%64 == 0), Kg = 232
+    x = _cl(torch.randn(n, c, h, h, device="cuda", dtype=torch.bfloat16))
+    gy = _cl(torch.randn(n, c, h, h, device="cuda", dtype=torch.bfloat16))
+    gw = e.conv2d_wgrad(gy, x, 3, 3, 1, 1, 1, 1, 1, 1, g)
+    wf = torch.zeros(c, c // g, 3, 3, device="cuda", requires_grad=True)
+    F.conv2d(x.float(), wf, None, 1, 1, 1, g).backward(gy.float())
+    err = (gw.float() - wf.grad).abs().max().item()
+    scale = wf.grad.abs().max().item()
+    assert err < 3e-2 * max(scale, 1.0), (err, scale)
