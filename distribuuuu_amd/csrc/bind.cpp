@@ -77,6 +77,10 @@ at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
                     at::Tensor rh, int64_t H, int64_t W,
                     c10::optional<at::Tensor> pout);
+std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
+                                 at::Tensor kt, at::Tensor v,
+                                 at::Tensor rel_w, at::Tensor rel_h,
+                                 int64_t H, int64_t W);
 // augment.hip
 at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
                               std::vector<double> mean,
@@ -132,5 +136,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_dgrad", &dwconv_dgrad);
   m.def("dwconv_wgrad", &dwconv_wgrad);
   m.def("mhsa_fwd", &mhsa_fwd);
+  m.def("mhsa_bwd", &mhsa_bwd);
   m.def("aug_crop_flip_norm", &aug_crop_flip_norm);
 }

@@ -78,25 +78,30 @@ def _rel_bwd_idx(h, w, device):
 
 class _HIPMHSARelPos(torch.autograd.Function):
     """Fused CDNA4 forward (one kernel: QK^T + decomposed rel-pos + softmax
-    + PV, score tile LDS-resident). The kernel also writes the softmax
-    probs P (39 MB/layer at L=196), so backward is the short analytic
-    chain dV = P^T dO, dP = dO V^T, dS = P(dP - rowsum(dP P)), dQ/dK bmms
-    and padded-gather rel-pos reductions — the previous full torch
-    recompute cost ~5.4 ms/step on BoTNet-50."""
+    + PV, score tile LDS-resident; the per-row rel-logit tables come from
+    the MFMA NT GEMM). The kernel also writes the softmax probs P
+    (39 MB/layer at L=196), and backward is three HIP kernels (mhsa_bwd):
+    per-q-tile dP -> dS -> dQ with the rel-pos fold, a dK/dV accumulation
+    pass, and the rel-table grad reduction — the previous ATen bmm/
+    elementwise chain cost ~3 ms/step on BoTNet-50."""
 
     @staticmethod
     def forward(ctx, q, k, v, rel_h, rel_w, h, w):
         n, heads, l, d = q.shape
+        e = ext()
         qf = q.reshape(n * heads, l, d).contiguous()
         kf = k.reshape(n * heads, l, d).contiguous()
         vt = v.reshape(n * heads, l, d).transpose(1, 2).contiguous()
-        # per-row relative-logit tables (fp32): RW = q@rel_w^T, RH = q@rel_h^T
-        rw = torch.matmul(qf.float(), rel_w.float().t()).contiguous()
-        rh = torch.matmul(qf.float(), rel_h.float().t()).contiguous()
+        # per-row relative-logit tables RW = q@rel_w^T, RH = q@rel_h^T on
+        # the MFMA NT GEMM (fp32 accumulate), expanded fp32 for the kernel
+        rw = e.gemm_nt(qf.reshape(-1, d), rel_w.contiguous()).float()
+        rh = e.gemm_nt(qf.reshape(-1, d), rel_h.contiguous()).float()
+        rw = rw.reshape(n * heads, l, -1).contiguous()
+        rh = rh.reshape(n * heads, l, -1).contiguous()
         need_grad = any(t.requires_grad for t in (q, k, v, rel_h, rel_w))
         pout = (torch.empty(n * heads, l, l, dtype=q.dtype, device=q.device)
                 if need_grad else None)
-        out = ext().mhsa_fwd(qf, kf, vt, rw, rh, h, w, pout)
+        out = e.mhsa_fwd(qf, kf, vt, rw, rh, h, w, pout)
         ctx.save_for_backward(q, k, v, rel_h, rel_w,
                               pout if pout is not None else q.new_empty(0))
         ctx.hw = (h, w)
@@ -108,30 +113,15 @@ class _HIPMHSARelPos(torch.autograd.Function):
         h, w = ctx.hw
         n, heads, l, d = q.shape
         bh = n * heads
-        qf = q.reshape(bh, l, d)
-        kf = k.reshape(bh, l, d)
-        vf = v.reshape(bh, l, d)
+        qf = q.reshape(bh, l, d).contiguous()
+        kt = k.reshape(bh, l, d).transpose(1, 2).contiguous()
+        vf = v.reshape(bh, l, d).contiguous()
         dO = gout.reshape(bh, l, d).contiguous()
-        dV = torch.bmm(P.transpose(1, 2), dO)
-        dP = torch.bmm(dO, vf.transpose(1, 2)).float()
-        Pf = P.float()
-        dS = Pf * (dP - (dP * Pf).sum(-1, keepdim=True))
-        dSb = dS.to(q.dtype)
-        dq = torch.bmm(dSb, kf)
-        dK = torch.bmm(dSb.transpose(1, 2), qf)
-        iw, ih = _rel_bwd_idx(h, w, q.device)
-        dSw = F.pad(dS.reshape(bh, l, h, w).sum(2), (w - 1, w - 1))
-        dRW = dSw.gather(2, iw.expand(bh, l, 2 * w - 1))
-        dSh = F.pad(dS.reshape(bh, l, h, w).sum(3), (h - 1, h - 1))
-        dRH = dSh.gather(2, ih.expand(bh, l, 2 * h - 1))
-        dq = dq + (torch.matmul(dRW, rel_w.float())
-                   + torch.matmul(dRH, rel_h.float())).to(q.dtype)
-        grw = torch.matmul(dRW.reshape(-1, 2 * w - 1).t(),
-                           qf.reshape(-1, d).float()).to(rel_w.dtype)
-        grh = torch.matmul(dRH.reshape(-1, 2 * h - 1).t(),
-                           qf.reshape(-1, d).float()).to(rel_h.dtype)
-        return (dq.reshape(n, heads, l, d), dK.reshape(n, heads, l, d),
-                dV.reshape(n, heads, l, d), grh, grw, None, None)
+        dq, dk, dv, grw, grh = ext().mhsa_bwd(
+            dO, P, qf, kt, vf, rel_w.contiguous(), rel_h.contiguous(), h, w)
+        return (dq.reshape(n, heads, l, d), dk.reshape(n, heads, l, d),
+                dv.reshape(n, heads, l, d), grh.to(rel_h.dtype),
+                grw.to(rel_w.dtype), None, None)
 
 
 def mhsa_relpos(q, k, v, rel_h, rel_w, h, w):
