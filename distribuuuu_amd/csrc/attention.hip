@@ -41,14 +41,16 @@ struct MhsaParams {
   int lpad;      // LDS row width for S (multiple of 16 + pad)
 };
 
+// Logits live in REGISTERS (LT tiles x 4 rows per lane); LDS holds only the
+// bf16 P tile for the O pass. The round-1 version kept a [QT][lpad] fp32
+// score slab in LDS (~86 KB -> 1 block/CU -> 1 wave/SIMD, nothing to hide
+// latency with); register logits cut LDS to 28 KB -> 5 blocks/CU.
+template <int LT>
 __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
   const int b = blockIdx.x;
   const int q0 = blockIdx.y * QT;
   extern __shared__ __align__(16) char smem[];
-  // S: [QT][lpad] fp32 ; P: [QT][lpad] bf16 (after S)
-  float* S = reinterpret_cast<float*>(smem);
-  __hip_bfloat16* P =
-      reinterpret_cast<__hip_bfloat16*>(smem + QT * p.lpad * 4);
+  __hip_bfloat16* P = reinterpret_cast<__hip_bfloat16*>(smem);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -57,7 +59,6 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
   const int qrow_base = q0 + wid * 16;
 
   // ---- load this wave's q fragments (A operand, reused across k tiles) ----
-  // A[i = lane%16][kk = (lane/16)*8 + j], 4 kk-chunks of 32 over D=128
   const int nd = p.D / 32;
   bf16x8a qfrag[4];
   {
@@ -73,67 +74,100 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
     }
   }
 
-  // ---- S = q@k^T + rel, one 16-wide k tile at a time ----------------------
-  for (int nt = 0; nt < p.ltiles16; ++nt) {
+  // ---- S = q@k^T + rel in registers ---------------------------------------
+  // D layout per tile: col kpos = nt*16 + il, row = kq*4 + rr
+  f32x4a sacc[LT];
+#pragma unroll
+  for (int nt = 0; nt < LT; ++nt) {
     f32x4a acc = {0.f, 0.f, 0.f, 0.f};
-    const int kr = nt * 16 + il;
-    const bool kok = kr < p.L;
-    const __hip_bfloat16* kp = p.k + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
+    if (nt < p.ltiles16) {
+      const int kr = nt * 16 + il;
+      const bool kok = kr < p.L;
+      const __hip_bfloat16* kp =
+          p.k + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      if (c >= nd) break;
-      bf16x8a kfrag = kok ? *reinterpret_cast<const bf16x8a*>(kp + c * 32 + kq * 8)
-                          : zero8();
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], kfrag, acc, 0,
-                                                    0, 0);
+      for (int c = 0; c < 4; ++c) {
+        if (c >= nd) break;
+        bf16x8a kfrag =
+            kok ? *reinterpret_cast<const bf16x8a*>(kp + c * 32 + kq * 8)
+                : zero8();
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], kfrag, acc, 0,
+                                                      0, 0);
+      }
     }
-    // D layout: col kpos = nt*16 + lane%16, row qrow = (lane/16)*4 + rr
+    sacc[nt] = acc;
+  }
+  // rel-logit add + L-tail mask (per lane: 4 rows x LT cols)
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int qr_local = kq * 4 + rr;
-      const int qr = qrow_base + qr_local;
+  for (int rr = 0; rr < 4; ++rr) {
+    const int qr = qrow_base + kq * 4 + rr;
+    const bool qok = qr < p.L;
+    const int wi = qok ? qr % p.W : 0;
+    const int hi = qok ? qr / p.W : 0;
+    const float* rwp = p.rw + ((int64_t)b * p.L + (qok ? qr : 0)) * (2 * p.W - 1);
+    const float* rhp = p.rh + ((int64_t)b * p.L + (qok ? qr : 0)) * (2 * p.H - 1);
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) {
       const int kpos = nt * 16 + il;
-      float v = acc[rr];
-      if (qr < p.L && kpos < p.L) {
-        const int dy = (kpos % p.W) - (qr % p.W) + p.W - 1;
-        const int dx = (kpos / p.W) - (qr / p.W) + p.H - 1;
-        v += p.rw[((int64_t)b * p.L + qr) * (2 * p.W - 1) + dy] +
-             p.rh[((int64_t)b * p.L + qr) * (2 * p.H - 1) + dx];
+      float v = sacc[nt][rr];
+      if (qok && kpos < p.L) {
+        const int dy = (kpos % p.W) - wi + p.W - 1;
+        const int dx = (kpos / p.W) - hi + p.H - 1;
+        v += rwp[dy] + rhp[dx];
       } else {
         v = -INFINITY;
       }
-      S[(wid * 16 + qr_local) * p.lpad + kpos] = v;
+      sacc[nt][rr] = v;
     }
   }
-  __builtin_amdgcn_wave_barrier();
 
-  // ---- row softmax (wave handles its own 16 rows; S rows are private) -----
-  const int lrounds = (p.lpad + 63) / 64;
-  for (int r = 0; r < 16; ++r) {
-    float* row = &S[(wid * 16 + r) * p.lpad];
+  // ---- register softmax: reduce over the 16 il-lanes of each kq group -----
+  float rmax[4], rsum[4];
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
     float m = -INFINITY;
-    for (int j = lane; j < p.L; j += 64) m = fmaxf(m, row[j]);
-    m = wave_reduce_max(m);
-    m = __shfl(m, 0);
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) m = fmaxf(m, sacc[nt][rr]);
+#pragma unroll
+    for (int s = 1; s < 16; s <<= 1) m = fmaxf(m, __shfl_xor(m, s));
+    rmax[rr] = m;
+  }
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const int qr = qrow_base + kq * 4 + rr;
     float s = 0.f;
-    for (int j = lane; j < p.L; j += 64) s += __expf(row[j] - m);
-    s = wave_reduce_sum(s);
-    s = __shfl(s, 0);
-    const float inv = 1.f / s;
-    __hip_bfloat16* prow = &P[(wid * 16 + r) * p.lpad];
-    const int qr = qrow_base + r;
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) {
+      const float e = (qr < p.L && nt * 16 + il < p.L)
+                          ? __expf(sacc[nt][rr] - rmax[rr]) : 0.f;
+      sacc[nt][rr] = e;  // exp stored back (0 for pad cols/garbage rows)
+      s += e;
+    }
+#pragma unroll
+    for (int st = 1; st < 16; st <<= 1) s += __shfl_xor(s, st);
+    rsum[rr] = (qr < p.L) ? 1.f / s : 0.f;
+  }
+  // write P to LDS (+ optional global), zero-fill the lpad tail
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const int row = wid * 16 + kq * 4 + rr;
+    const int qr = q0 + row;
+    __hip_bfloat16* prow = &P[row * p.lpad];
     __hip_bfloat16* gp = (p.pout != nullptr && qr < p.L)
                              ? p.pout + ((int64_t)b * p.L + qr) * p.L
                              : nullptr;
-    for (int jr = 0; jr < lrounds; ++jr) {
-      const int j = jr * 64 + lane;
-      if (j < p.lpad) {
-        const __hip_bfloat16 pv = from_f32<__hip_bfloat16>(
-            j < p.L ? __expf(row[j] - m) * inv : 0.f);
-        prow[j] = pv;
-        if (gp && j < p.L) gp[j] = pv;
-      }
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) {
+      const int kpos = nt * 16 + il;
+      const __hip_bfloat16 pv =
+          from_f32<__hip_bfloat16>(sacc[nt][rr] * rsum[rr]);
+      prow[kpos] = pv;
+      if (gp && kpos < p.L) gp[kpos] = pv;
     }
+    // lpad tail beyond LT*16 (LT*16 >= ltiles16*16 covers it; pad cols in
+    // written tiles already got exp=0)
+    for (int j = LT * 16 + il; j < p.lpad; j += 16)
+      prow[j] = from_f32<__hip_bfloat16>(0.f);
   }
   __builtin_amdgcn_wave_barrier();
 
@@ -202,14 +236,13 @@ struct MhsaBwdQParams {
   int ltiles16, lpad;
 };
 
+template <int LT>
 __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
   const int b = blockIdx.x;
   const int q0 = blockIdx.y * QT;
   extern __shared__ __align__(16) char smem[];
-  float* S = reinterpret_cast<float*>(smem);                  // dP then free
-  __hip_bfloat16* DS =
-      reinterpret_cast<__hip_bfloat16*>(smem + QT * p.lpad * 4);
-  float* RWs = reinterpret_cast<float*>(smem + QT * p.lpad * 6);  // [QT][32]
+  __hip_bfloat16* DS = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* RWs = reinterpret_cast<float*>(smem + QT * p.lpad * 2);  // [QT][32]
   float* RHs = RWs + QT * 32;                                     // [QT][32]
 
   const int tid = threadIdx.x;
@@ -233,50 +266,70 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
                       : zero8();
   }
 
-  // ---- dP = dO @ v^T into the fp32 slab -----------------------------------
-  for (int nt = 0; nt < p.ltiles16; ++nt) {
+  // ---- dP = dO @ v^T in registers (same layout as the forward logits) ----
+  f32x4a dacc[LT];
+#pragma unroll
+  for (int nt = 0; nt < LT; ++nt) {
     f32x4a acc = {0.f, 0.f, 0.f, 0.f};
-    const int kr = nt * 16 + il;
-    const bool kok = kr < p.L;
-    const __hip_bfloat16* vp = p.v + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
+    if (nt < p.ltiles16) {
+      const int kr = nt * 16 + il;
+      const bool kok = kr < p.L;
+      const __hip_bfloat16* vp =
+          p.v + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      if (c >= nd) break;
-      bf16x8a vfrag =
-          kok ? *reinterpret_cast<const bf16x8a*>(vp + c * 32 + kq * 8)
-              : zero8();
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[c], vfrag, acc, 0,
-                                                    0, 0);
+      for (int c = 0; c < 4; ++c) {
+        if (c >= nd) break;
+        bf16x8a vfrag =
+            kok ? *reinterpret_cast<const bf16x8a*>(vp + c * 32 + kq * 8)
+                : zero8();
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[c], vfrag, acc,
+                                                      0, 0, 0);
+      }
     }
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int qr_local = kq * 4 + rr;
-      const int kpos = nt * 16 + il;
-      S[(wid * 16 + qr_local) * p.lpad + kpos] = acc[rr];
-    }
+    dacc[nt] = acc;
   }
-  __builtin_amdgcn_wave_barrier();
 
-  // ---- dS = P (dP - rowdot); write LDS bf16 + global ----------------------
-  for (int r = 0; r < 16; ++r) {
-    const int qr = qrow_base + r;
-    float* row = &S[(wid * 16 + r) * p.lpad];
+  // ---- rowdot = sum(dP * P) via register pass + lane shuffles -------------
+  float dot[4];
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const int qr = qrow_base + kq * 4 + rr;
     const __hip_bfloat16* prow =
         p.P + ((int64_t)b * p.L + (qr < p.L ? qr : 0)) * p.L;
-    float dot = 0.f;
-    for (int j = lane; j < p.L; j += 64) dot += row[j] * to_f32(prow[j]);
-    dot = wave_reduce_sum(dot);
-    dot = __shfl(dot, 0);
-    __hip_bfloat16* dsrow = &DS[(wid * 16 + r) * p.lpad];
+    float s = 0.f;
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) {
+      const int kpos = nt * 16 + il;
+      if (qr < p.L && kpos < p.L)
+        s += dacc[nt][rr] * to_f32(prow[kpos]);
+    }
+#pragma unroll
+    for (int st = 1; st < 16; st <<= 1) s += __shfl_xor(s, st);
+    dot[rr] = s;
+  }
+
+  // ---- dS = P (dP - dot): write LDS bf16 + global (P re-read from L1) -----
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+    const int row = wid * 16 + kq * 4 + rr;
+    const int qr = q0 + row;
+    const __hip_bfloat16* prow =
+        p.P + ((int64_t)b * p.L + (qr < p.L ? qr : 0)) * p.L;
+    __hip_bfloat16* dsrow = &DS[row * p.lpad];
     __hip_bfloat16* gds =
         (qr < p.L) ? p.ds + ((int64_t)b * p.L + qr) * p.L : nullptr;
-    for (int j = lane; j < p.lpad; j += 64) {
+#pragma unroll
+    for (int nt = 0; nt < LT; ++nt) {
+      const int kpos = nt * 16 + il;
+      const bool ok = qr < p.L && kpos < p.L;
       const float dsv =
-          (j < p.L) ? to_f32(prow[j]) * (row[j] - dot) : 0.f;
+          ok ? to_f32(prow[kpos]) * (dacc[nt][rr] - dot[rr]) : 0.f;
       const __hip_bfloat16 dsb = from_f32<__hip_bfloat16>(dsv);
-      dsrow[j] = dsb;
-      if (gds && j < p.L) gds[j] = dsb;
+      dsrow[kpos] = dsb;
+      if (gds && kpos < p.L) gds[kpos] = dsb;
     }
+    for (int j = LT * 16 + il; j < p.lpad; j += 16)
+      dsrow[j] = from_f32<__hip_bfloat16>(0.f);
   }
   __builtin_amdgcn_wave_barrier();
 
@@ -449,36 +502,90 @@ struct MhsaRelGradParams {
 
 __global__ __launch_bounds__(256) void mhsa_rel_grad_kernel(
     MhsaRelGradParams p) {
-  __shared__ float drow[32];
-  __shared__ __hip_bfloat16 qrow[128];
+  // sync-free: thread (half, d) accumulates its half of the m-range for one
+  // d column; q[r, d] loads coalesce across the 128 d-threads, the shared
+  // dr[r, m] scalars broadcast within each wave (same address per lane
+  // group). The first LDS-staged version synced twice per row and was
+  // latency-bound at ~2 ms/step for a 350 MFLOP reduction.
   const int tid = threadIdx.x;
   const int d = tid & 127;
-  const int half = tid >> 7;               // m in [half*14, ...)
+  const int half = tid >> 7;
   const int m0 = half * ((p.M + 1) / 2);
   const int mn = min(p.M - m0, (p.M + 1) / 2);
+  if (d >= p.D) return;
   float acc[16] = {};
   const int64_t r0 = (int64_t)blockIdx.x * p.rows_per_block;
   const int64_t r1 = min(r0 + p.rows_per_block, p.rows);
   for (int64_t r = r0; r < r1; ++r) {
-    if (tid < p.M) drow[tid] = p.dr[r * p.M + tid];
-    if (tid >= 128 && tid - 128 < p.D / 8) {
-      const int c8 = (tid - 128) * 8;
-      *reinterpret_cast<bf16x8a*>(&qrow[c8]) =
-          *reinterpret_cast<const bf16x8a*>(p.q + r * p.D + c8);
+    const float qv = to_f32(p.q[r * p.D + d]);
+    const float* dr = p.dr + r * p.M + m0;
+#pragma unroll 7
+    for (int m = 0; m < mn; ++m) acc[m] += dr[m] * qv;
+  }
+  for (int m = 0; m < mn; ++m)
+    atomicAdd(&p.out[(int64_t)(m0 + m) * p.D + d], acc[m]);
+}
+
+// ---- per-row rel-logit tables: out[row, m] = sum_d q[row, d] rel[m, d] ----
+// fp32 accumulate AND fp32 output (a bf16 GEMM output rounds the table and
+// measurably shifts the softmax when logits are large).
+struct RelTabParams {
+  const __hip_bfloat16* q;    // [rows, D]
+  const __hip_bfloat16* rel;  // [M, D]
+  float* out;                 // [rows, M]
+  int64_t rows;
+  int M, D;
+};
+
+__global__ __launch_bounds__(256) void mhsa_rel_tables_kernel(
+    RelTabParams p) {
+  __shared__ __hip_bfloat16 rel_s[32 * 128];
+  __shared__ __hip_bfloat16 q_s[8][128];
+  for (int i = threadIdx.x; i < p.M * p.D; i += 256) rel_s[i] = p.rel[i];
+  __syncthreads();
+  const int m = threadIdx.x & 31;
+  const int rl = threadIdx.x >> 5;  // 8 rows per iteration
+  for (int64_t row0 = (int64_t)blockIdx.x * 8; row0 < p.rows;
+       row0 += (int64_t)gridDim.x * 8) {
+    if (threadIdx.x < 16 * (p.D / 8)) {
+      const int rr = threadIdx.x / (p.D / 8);
+      const int c8 = (threadIdx.x % (p.D / 8)) * 8;
+      if (rr < 8 && row0 + rr < p.rows)
+        *reinterpret_cast<bf16x8a*>(&q_s[rr][c8]) =
+            *reinterpret_cast<const bf16x8a*>(p.q + (row0 + rr) * p.D + c8);
     }
     __syncthreads();
-    if (d < p.D) {
-      const float qv = to_f32(qrow[d]);
-      for (int m = 0; m < mn; ++m) acc[m] += drow[m0 + m] * qv;
+    if (m < p.M && row0 + rl < p.rows) {
+      float acc = 0.f;
+      for (int dd = 0; dd < p.D; ++dd)
+        acc += to_f32(q_s[rl][dd]) * to_f32(rel_s[m * p.D + dd]);
+      p.out[(row0 + rl) * p.M + m] = acc;
     }
     __syncthreads();
   }
-  if (d < p.D)
-    for (int m = 0; m < mn; ++m)
-      atomicAdd(&p.out[(int64_t)(m0 + m) * p.D + d], acc[m]);
 }
 
 }  // namespace
+
+// rw = q @ rel^T, fp32 out: [rows, M]
+at::Tensor mhsa_rel_tables(at::Tensor q2d, at::Tensor rel) {
+  CHECK_GPU(q2d);
+  const int64_t rows = q2d.size(0);
+  const int D = q2d.size(1), M = rel.size(0);
+  TORCH_CHECK(M <= 32 && D <= 128 && D % 8 == 0, "rel tables: M<=32 D<=128");
+  auto out = at::empty({rows, (int64_t)M}, q2d.options().dtype(at::kFloat));
+  RelTabParams p;
+  p.q = (const __hip_bfloat16*)q2d.data_ptr();
+  p.rel = (const __hip_bfloat16*)rel.data_ptr();
+  p.out = out.data_ptr<float>();
+  p.rows = rows;
+  p.M = M;
+  p.D = D;
+  hipLaunchKernelGGL(mhsa_rel_tables_kernel,
+                     dim3((int)std::min<int64_t>(ceil_div(rows, 8), 2048)),
+                     dim3(256), 0, cur_stream(), p);
+  return out;
+}
 
 // Backward: dO + saved (P, q, k, v) -> dq (incl. rel-pos fold), dk, dv,
 // grw, grh. kt/vt are the transposed operands ([B, D, L]).
@@ -512,16 +619,21 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
   p.B = B; p.L = L; p.D = D; p.H = H; p.W = W;
   p.ltiles16 = (L + 15) / 16;
   p.lpad = lpad;
-  const int smem_bytes = QT * lpad * 6 + QT * 32 * 8;
-  static int bwd_smem_set = 0;
-  if (smem_bytes > 65536 && !bwd_smem_set) {
-    hipFuncSetAttribute((const void*)mhsa_bwd_q_kernel,
-                        hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
-    bwd_smem_set = 1;
-  }
+  TORCH_CHECK(p.ltiles16 <= 16, "mhsa_bwd: L must be <= 256");
+  const int smem_bytes = QT * lpad * 2 + QT * 32 * 8;
   dim3 grid(B, (L + QT - 1) / QT);
-  hipLaunchKernelGGL(mhsa_bwd_q_kernel, grid, dim3(256), smem_bytes,
-                     cur_stream(), p);
+  if (p.ltiles16 <= 4)
+    hipLaunchKernelGGL(mhsa_bwd_q_kernel<4>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else if (p.ltiles16 <= 8)
+    hipLaunchKernelGGL(mhsa_bwd_q_kernel<8>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else if (p.ltiles16 <= 13)
+    hipLaunchKernelGGL(mhsa_bwd_q_kernel<13>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else
+    hipLaunchKernelGGL(mhsa_bwd_q_kernel<16>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
 
   MhsaBwdKVParams kv;
   kv.ds = (const __hip_bfloat16*)ds.data_ptr();
@@ -575,15 +687,20 @@ at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
   p.B = B; p.L = L; p.D = D; p.H = H; p.W = W;
   p.ltiles16 = (L + 15) / 16;
   p.lpad = ((L + 31) / 32) * 32;  // multiple of 32 for the P fragment chunks
-  const int smem_bytes = QT * p.lpad * 4 + QT * p.lpad * 2;
-  static int max_smem_set = 0;
-  if (smem_bytes > 65536 && !max_smem_set) {
-    hipFuncSetAttribute((const void*)mhsa_fwd_kernel,
-                        hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
-    max_smem_set = 1;
-  }
+  TORCH_CHECK(p.ltiles16 <= 16, "mhsa: L must be <= 256");
+  const int smem_bytes = QT * p.lpad * 2;  // P tile only (logits in regs)
   dim3 grid(B, (L + QT - 1) / QT);
-  hipLaunchKernelGGL(mhsa_fwd_kernel, grid, dim3(256), smem_bytes,
-                     cur_stream(), p);
+  if (p.ltiles16 <= 4)
+    hipLaunchKernelGGL(mhsa_fwd_kernel<4>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else if (p.ltiles16 <= 8)
+    hipLaunchKernelGGL(mhsa_fwd_kernel<8>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else if (p.ltiles16 <= 13)
+    hipLaunchKernelGGL(mhsa_fwd_kernel<13>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
+  else
+    hipLaunchKernelGGL(mhsa_fwd_kernel<16>, grid, dim3(256), smem_bytes,
+                       cur_stream(), p);
   return o;
 }

@@ -81,6 +81,7 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
                                  at::Tensor kt, at::Tensor v,
                                  at::Tensor rel_w, at::Tensor rel_h,
                                  int64_t H, int64_t W);
+at::Tensor mhsa_rel_tables(at::Tensor q2d, at::Tensor rel);
 // augment.hip
 at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
                               std::vector<double> mean,
@@ -137,5 +138,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_wgrad", &dwconv_wgrad);
   m.def("mhsa_fwd", &mhsa_fwd);
   m.def("mhsa_bwd", &mhsa_bwd);
+  m.def("mhsa_rel_tables", &mhsa_rel_tables);
   m.def("aug_crop_flip_norm", &aug_crop_flip_norm);
 }

@@ -92,12 +92,13 @@ class _HIPMHSARelPos(torch.autograd.Function):
         qf = q.reshape(n * heads, l, d).contiguous()
         kf = k.reshape(n * heads, l, d).contiguous()
         vt = v.reshape(n * heads, l, d).transpose(1, 2).contiguous()
-        # per-row relative-logit tables RW = q@rel_w^T, RH = q@rel_h^T on
-        # the MFMA NT GEMM (fp32 accumulate), expanded fp32 for the kernel
-        rw = e.gemm_nt(qf.reshape(-1, d), rel_w.contiguous()).float()
-        rh = e.gemm_nt(qf.reshape(-1, d), rel_h.contiguous()).float()
-        rw = rw.reshape(n * heads, l, -1).contiguous()
-        rh = rh.reshape(n * heads, l, -1).contiguous()
+        # per-row relative-logit tables RW = q@rel_w^T, RH = q@rel_h^T —
+        # dedicated kernel with fp32 accumulate AND fp32 output (a bf16
+        # GEMM output rounds the tables and shifts large-logit softmaxes)
+        rw = e.mhsa_rel_tables(qf.reshape(-1, d), rel_w.contiguous())
+        rh = e.mhsa_rel_tables(qf.reshape(-1, d), rel_h.contiguous())
+        rw = rw.reshape(n * heads, l, -1)
+        rh = rh.reshape(n * heads, l, -1)
         need_grad = any(t.requires_grad for t in (q, k, v, rel_h, rel_w))
         pout = (torch.empty(n * heads, l, l, dtype=q.dtype, device=q.device)
                 if need_grad else None)
