@@ -42,9 +42,19 @@ struct ConvParams {
   int M, nspan, ksteps;     // nspan = ceil(S*Cg/BK), ksteps = R*nspan
   int tiles_m;              // for XCD swizzle
   float* part;              // EMIT: [tiles_m*2, 2*Kt] BN sum/sumsq partials
+  // EMODE 2 (dgrad -> BN-backward stats): this conv's output IS the
+  // consuming BatchNorm's upstream grad gy; the epilogue also reads the
+  // BN's input x at the SAME positions and emits [sum(g), sum(g*x)]
+  // partials with g = act'(x*scale+shift) * gy — the standalone
+  // bn_bwd_reduce pass over the whole tensor disappears.
+  const __hip_bfloat16* bnx;
+  const float* bnscale;
+  const float* bnshift;
+  int bnact;                // 0 none, 1 relu
 };
 
-template <bool EMIT>
+// EMODE: 0 plain, 1 forward BN sum/sumsq partials, 2 backward masked stats
+template <int EMODE>
 __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
   const int g = blockIdx.z;
   // XCD-aware swizzle over m-tiles (T1; bijective form)
@@ -207,14 +217,68 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
           if (k0 + j < p.Kg) p.y[obase + k0 + j] = u.b[j];
       }
     }
-    if (EMIT) {
+    if (EMODE == 1) {
       const int base = tile_m * BM + wm * 64 + mi * 16;
       bn_partial_col_accum(slab, ps, pq,
                            (int)min((int64_t)16, (int64_t)p.M - base), lane);
     }
+    if (EMODE == 2) {
+      // load x at the SAME positions, build g = act'(x*scale+shift)*gy,
+      // run the two-pass slab column accumulation (g, then g*x)
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } xv;
+      const int k0 = tile_n * BN + wn * 64 + ec;
+      bool valid = m < p.M;
+      if (valid) {
+        const int n = m / HoWo;
+        const int rem = m - n * HoWo;
+        const int64_t obase =
+            (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh + p.oh0) * p.WoA +
+             (rem % p.Wo) * p.osw + p.ow0) * p.Kt + g * p.Kg;
+        if (k0 + 16 <= p.Kg) {
+          xv.q[0] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0]);
+          xv.q[1] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0 + 8]);
+        } else {
+          xv.q[0] = uint4{0, 0, 0, 0};
+          xv.q[1] = uint4{0, 0, 0, 0};
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            if (k0 + j < p.Kg) xv.b[j] = p.bnx[obase + k0 + j];
+        }
+      }
+      __builtin_amdgcn_wave_barrier();  // everyone done with the y stripe
+      // pass 1: g   (invalid rows contribute zeros)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        float gj = 0.f;
+        if (valid) {
+          const int cc = g * p.Kg + k0 + j;
+          const float yj = to_f32(from_f32<__hip_bfloat16>(
+              slab[er * 68 + ec + j]));  // the ROUNDED stored gy
+          gj = yj;
+          if (p.bnact == 1 &&
+              to_f32(xv.b[j]) * p.bnscale[cc] + p.bnshift[cc] <= 0.f)
+            gj = 0.f;
+        }
+        slab[er * 68 + ec + j] = gj;
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int r = 0; r < 16; ++r) ps += slab[r * 68 + lane];
+      __builtin_amdgcn_wave_barrier();
+      // pass 2: g*x (read own g back, multiply by x)
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        slab[er * 68 + ec + j] *= valid ? to_f32(xv.b[j]) : 0.f;
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pq += slab[r * 68 + lane];
+    }
     __builtin_amdgcn_wave_barrier();
   }
-  if (EMIT) {
+  if (EMODE == 1 || EMODE == 2) {
     const int kbase = tile_n * BN + wn * 64;
     bn_partial_store(p.part,
                      (int64_t)(tile_m * 2 + wm) * 2 * p.Kt + g * p.Kg + kbase,
@@ -686,15 +750,27 @@ at::Tensor conv2d_fwd_into_fp32(at::Tensor x, at::Tensor w, at::Tensor y,
                                 int64_t oh0, int64_t ow0,
                                 at::Tensor* part_out);
 
+// BN-backward stat emission (EMODE 2) descriptor: the conv being launched is
+// a DGRAD whose output is the gy of the BatchNorm that produced this conv's
+// logical input; x/scale/shift/act describe that BN.
+struct BnBwdEmit {
+  const __hip_bfloat16* x;
+  const float* scale;
+  const float* shift;
+  int act;
+};
+
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
                            int64_t osh = 1, int64_t osw = 1, int64_t oh0 = 0,
-                           int64_t ow0 = 0, at::Tensor* part_out = nullptr) {
+                           int64_t ow0 = 0, at::Tensor* part_out = nullptr,
+                           const BnBwdEmit* bemit = nullptr) {
   CHECK_GPU(x);
   if (x.scalar_type() == at::kFloat)
     return conv2d_fwd_into_fp32(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw,
-                                groups, osh, osw, oh0, ow0, part_out);
+                                groups, osh, osw, oh0, ow0,
+                                bemit ? nullptr : part_out);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16/fp32 only");
   check_nhwc(x, "x");
   check_nhwc(w, "w");
@@ -719,15 +795,29 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   p.ksteps = R * p.nspan;
   p.tiles_m = (p.M + BM - 1) / BM;
   p.part = nullptr;
+  p.bnx = nullptr;
+  p.bnscale = nullptr;
+  p.bnshift = nullptr;
+  p.bnact = 0;
   dim3 grid(p.tiles_m, (Kg + BN - 1) / BN, groups);
-  if (part_out != nullptr) {
+  if (bemit != nullptr) {
     *part_out = at::empty({(int64_t)p.tiles_m * 2, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
-    hipLaunchKernelGGL(conv_igemm_fwd_kernel<true>, grid, dim3(256), 0,
+    p.bnx = bemit->x;
+    p.bnscale = bemit->scale;
+    p.bnshift = bemit->shift;
+    p.bnact = bemit->act;
+    hipLaunchKernelGGL(conv_igemm_fwd_kernel<2>, grid, dim3(256), 0,
+                       cur_stream(), p);
+  } else if (part_out != nullptr) {
+    *part_out = at::empty({(int64_t)p.tiles_m * 2, (int64_t)2 * Kt},
+                          x.options().dtype(at::kFloat));
+    p.part = part_out->data_ptr<float>();
+    hipLaunchKernelGGL(conv_igemm_fwd_kernel<1>, grid, dim3(256), 0,
                        cur_stream(), p);
   } else {
-    hipLaunchKernelGGL(conv_igemm_fwd_kernel<false>, grid, dim3(256), 0,
+    hipLaunchKernelGGL(conv_igemm_fwd_kernel<0>, grid, dim3(256), 0,
                        cur_stream(), p);
   }
   return y;
@@ -735,12 +825,14 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
 
 at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                           int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                          int64_t groups, at::Tensor* part_out);
+                          int64_t groups, at::Tensor* part_out,
+                          const BnBwdEmit* bemit = nullptr);
 
 static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
                                   int64_t sw, int64_t ph, int64_t pw,
                                   int64_t dh, int64_t dw, int64_t groups,
-                                  at::Tensor* part_out) {
+                                  at::Tensor* part_out,
+                                  const BnBwdEmit* bemit = nullptr) {
   const int N = x.size(0), H = x.size(2), W = x.size(3);
   const int Kt = w.size(0), R = w.size(2), S = w.size(3);
   // v2 (3-slot glds ring) wins when the reduction is deep enough to fill
@@ -757,7 +849,8 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
   const int v2Cg = C_ / (int)groups;
   if (!v2_off && v2Kg >= 192 && (int64_t)R * S * v2Cg >= 512 &&
       v2Cg % 8 == 0 && x.scalar_type() == at::kBFloat16)
-    return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, groups, part_out);
+    return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, groups, part_out,
+                          bemit);
   static const bool small_off = []() {
     const char* e = getenv("DISTRIBUUUU_CONV_SMALL");
     return e && e[0] == '0';
@@ -775,6 +868,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     sp.C = C_; sp.Kt = Kt; sp.M = M;
     sp.part = nullptr;
     const dim3 sg((int)((M + 255) / 256));
+    if (bemit != nullptr) part_out = nullptr;  // small 1x1: no bwd emission
     if (part_out != nullptr) {
       *part_out = at::empty({(int64_t)sg.x * 4, (int64_t)2 * Kt},
                             x.options().dtype(at::kFloat));
@@ -834,6 +928,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     sp.zbuf = (const __hip_bfloat16*)zbuf.data_ptr();
     sp.part = nullptr;
     const dim3 skg((int)((sp.M + 255) / 256));
+    if (bemit != nullptr) part_out = nullptr;  // smallk: no bwd emission
     if (part_out != nullptr) {
       *part_out = at::empty({(int64_t)skg.x * 4, 128},
                             x.options().dtype(at::kFloat));
@@ -861,7 +956,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
   auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups, 1,
-                         1, 0, 0, part_out);
+                         1, 0, 0, part_out, bemit);
 }
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
@@ -1028,9 +1123,12 @@ at::Tensor pad_channels(at::Tensor x, int64_t Cn) {
 }
 
 // dgrad: gx = conv(dilate(gy), flipT(w)) written into a zeroed [H, W] canvas
-at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
-                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
-                        int64_t dh, int64_t dw, int64_t groups) {
+static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
+                                    int64_t W, int64_t sh, int64_t sw,
+                                    int64_t ph, int64_t pw, int64_t dh,
+                                    int64_t dw, int64_t groups,
+                                    at::Tensor* part_out,
+                                    const BnBwdEmit* bemit) {
   CHECK_GPU(gy);
   check_nhwc(gy, "gy");
   const int N = gy.size(0);
@@ -1038,12 +1136,14 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
   const int Ct = Cg * groups;
   auto wt = weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
   if (R == 1 && S == 1 && (sh > 1 || sw > 1) && ph == 0 && pw == 0) {
-    // strided-output GEMM: gx[ho*sh, wo*sw] = gy[ho, wo] @ w^T, rest zero
+    // strided-output GEMM: gx[ho*sh, wo*sw] = gy[ho, wo] @ w^T, rest zero.
+    // The zeroed gap positions contribute g = 0 to the BN-backward sums, so
+    // partials over written positions alone are the full sums.
     auto gx1 = at::empty({N, Ct, H, W}, gy.options().memory_format(
                                             at::MemoryFormat::ChannelsLast));
     gx1.zero_();
     conv2d_fwd_into(gy, wt, gx1, gy.size(2), gy.size(3), 1, 1, 0, 0, 1, 1,
-                    groups, sh, sw);
+                    groups, sh, sw, 0, 0, part_out, bemit);
     return gx1;
   }
   if (R == 3 && S == 3 && sh == 2 && sw == 2 && dh == 1 && dw == 1 &&
@@ -1056,6 +1156,7 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
     // weights are flipT(w) rows (1 - h0)::2.
     auto gx = at::empty({N, Ct, H, W}, gy.options().memory_format(
                                            at::MemoryFormat::ChannelsLast));
+    std::vector<at::Tensor> parts;
     for (int h0 = 0; h0 < 2; ++h0) {
       const int64_t hu = (H - h0 + 1) >> 1;
       auto wr = (h0 == 0) ? wt.slice(2, 1, 2) : wt.slice(2, 0, 3, 2);
@@ -1063,10 +1164,14 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
         const int64_t wu = (W - w0 + 1) >> 1;
         auto wsub = ((w0 == 0) ? wr.slice(3, 1, 2) : wr.slice(3, 0, 3, 2))
                         .contiguous(at::MemoryFormat::ChannelsLast);
+        at::Tensor sub_part;
         conv2d_fwd_into(gy, wsub, gx, hu, wu, 1, 1, 0, 0, 1, 1, groups, 2, 2,
-                        h0, w0);
+                        h0, w0, bemit ? &sub_part : nullptr, bemit);
+        if (bemit && sub_part.defined()) parts.push_back(sub_part);
       }
     }
+    if (bemit && part_out && parts.size() == 4)
+      *part_out = at::cat(parts, 0);
     return gx;
   }
   const int Kt_ = gy.size(1);
@@ -1074,8 +1179,9 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
       dw * (S - 1) == 2 * pw && (groups == 1 || Kt_ % (8 * groups) == 0)) {
     // same-size conv: the plain fwd path (incl. the v2 ring kernel) applies;
     // grouped too — wt is [Ct, Kg, R, S], a valid grouped conv weight
-    return conv2d_fwd(gy, wt, 1, 1, dh * (R - 1) - ph, dw * (S - 1) - pw,
-                      dh, dw, groups);
+    return conv2d_fwd_impl(gy, wt, 1, 1, dh * (R - 1) - ph,
+                           dw * (S - 1) - pw, dh, dw, groups, part_out,
+                           bemit);
   }
   (void)Kt_;
   auto gyd = (sh == 1 && sw == 1) ? gy : dilate_nhwc(gy, sh, sw);
@@ -1089,6 +1195,38 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
   (void)Hd; (void)Wd;
   auto gx = at::empty({N, Ct, H, W}, gy.options().memory_format(
                                            at::MemoryFormat::ChannelsLast));
-  conv2d_fwd_into(gyd, wt, gx, H, W, 1, 1, pph, ppw, dh, dw, groups);
+  conv2d_fwd_into(gyd, wt, gx, H, W, 1, 1, pph, ppw, dh, dw, groups, 1, 1, 0,
+                  0, part_out, bemit);
   return gx;
+}
+
+at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
+                        int64_t sh, int64_t sw, int64_t ph, int64_t pw,
+                        int64_t dh, int64_t dw, int64_t groups) {
+  return conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
+                           nullptr, nullptr);
+}
+
+// dgrad + BN-backward stat partials (docs/DESIGN_bn_conv_fusion.md, the
+// dgrad-side analogue of conv2d_fwd_bn): bnx/scale/shift/act describe the
+// BatchNorm whose gy this dgrad produces. Returns {gx, part}; part is an
+// EMPTY tensor when the dispatched kernel family doesn't emit (caller falls
+// back to the standalone reduce).
+std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor gy, at::Tensor w,
+                                        int64_t H, int64_t W, int64_t sh,
+                                        int64_t sw, int64_t ph, int64_t pw,
+                                        int64_t dh, int64_t dw,
+                                        int64_t groups, at::Tensor bnx,
+                                        at::Tensor bnscale, at::Tensor bnshift,
+                                        int64_t bnact) {
+  BnBwdEmit em;
+  em.x = (const __hip_bfloat16*)bnx.data_ptr();
+  em.scale = bnscale.data_ptr<float>();
+  em.shift = bnshift.data_ptr<float>();
+  em.act = (int)bnact;
+  at::Tensor part;
+  auto gx = conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
+                              &part, &em);
+  if (!part.defined()) part = at::empty({0, 0}, bnscale.options());
+  return {gx, part};
 }

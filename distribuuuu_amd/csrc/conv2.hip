@@ -33,6 +33,11 @@ struct Conv2Params {
   int M, nspan, ksteps;
   int tiles_m;
   float* part;  // EMIT: [tiles_m*4, 2*Kt] BN sum/sumsq partials
+  // EMODE 2: BN-backward masked stats (see conv.hip ConvParams)
+  const __hip_bfloat16* bnx;
+  const float* bnscale;
+  const float* bnshift;
+  int bnact;
 };
 
 // st_16x32 swizzle on a byte offset within a tile (1024-B subtiles)
@@ -40,7 +45,8 @@ DEV_INLINE int swz(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
 
 #define WAITVM(N) asm volatile("s_waitcnt vmcnt(" #N ")" ::: "memory")
 
-template <bool EMIT>
+// EMODE: 0 plain, 1 forward BN sum/sumsq partials, 2 backward masked stats
+template <int EMODE>
 __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   const int g = blockIdx.z;
   int tile_m = blockIdx.x, tile_n = blockIdx.y;
@@ -216,14 +222,61 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
           if (k0 + j < p.K) p.y[obase + ec + j] = u.b[j];
       }
     }
-    if (EMIT) {
+    if (EMODE == 1) {
       const int base = tile_m * BM2 + wm * 64 + mi * 16;
       bn_partial_col_accum(slab, ps, pq,
                            (int)min((int64_t)16, (int64_t)p.M - base), lane);
     }
+    if (EMODE == 2) {
+      // g = act'(x*scale+shift) * gy; two slab column passes (g, g*x)
+      union {
+        __hip_bfloat16 b[16];
+        uint4 q[2];
+      } xv;
+      const int k0 = tile_n * BN2 + wn * 64 + ec;
+      const bool valid = m < p.M;
+      if (valid) {
+        const int n = m / HoWo;
+        const int rem = m - n * HoWo;
+        const int64_t obase = ((int64_t)n * HoWo + rem) * p.Kt + g * p.K;
+        if (k0 + 16 <= p.K) {
+          xv.q[0] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0]);
+          xv.q[1] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0 + 8]);
+        } else {
+          xv.q[0] = uint4{0, 0, 0, 0};
+          xv.q[1] = uint4{0, 0, 0, 0};
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            if (k0 + j < p.K) xv.b[j] = p.bnx[obase + k0 + j];
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        float gj = 0.f;
+        if (valid) {
+          const int cc = g * p.K + k0 + j;
+          gj = to_f32(from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]));
+          if (p.bnact == 1 &&
+              to_f32(xv.b[j]) * p.bnscale[cc] + p.bnshift[cc] <= 0.f)
+            gj = 0.f;
+        }
+        slab[er * 68 + ec + j] = gj;
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int r = 0; r < 16; ++r) ps += slab[r * 68 + lane];
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        slab[er * 68 + ec + j] *= valid ? to_f32(xv.b[j]) : 0.f;
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pq += slab[r * 68 + lane];
+    }
     __builtin_amdgcn_wave_barrier();
   }
-  if (EMIT) {
+  if (EMODE != 0) {
     const int kbase = tile_n * BN2 + wn * 64;
     bn_partial_store(p.part,
                      (int64_t)(tile_m * 4 + wm) * 2 * p.Kt + g * p.K + kbase,
@@ -274,9 +327,17 @@ __global__ void pad_weight_span_kernel(const T* __restrict__ w,
 
 }  // namespace
 
+struct BnBwdEmit {
+  const __hip_bfloat16* x;
+  const float* scale;
+  const float* shift;
+  int act;
+};
+
 at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                           int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                          int64_t groups, at::Tensor* part_out) {
+                          int64_t groups, at::Tensor* part_out,
+                          const BnBwdEmit* bemit) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
   check_nhwc(x, "x");
@@ -345,15 +406,29 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   p.ksteps = R * p.nspan;
   p.tiles_m = (p.M + BM2 - 1) / BM2;
   p.part = nullptr;
+  p.bnx = nullptr;
+  p.bnscale = nullptr;
+  p.bnshift = nullptr;
+  p.bnact = 0;
   dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2, groups);
-  if (part_out != nullptr) {
+  if (bemit != nullptr) {
     *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
-    hipLaunchKernelGGL(conv_igemm_v2_kernel<true>, grid, dim3(512), 0,
+    p.bnx = bemit->x;
+    p.bnscale = bemit->scale;
+    p.bnshift = bemit->shift;
+    p.bnact = bemit->act;
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<2>, grid, dim3(512), 0,
+                       cur_stream(), p);
+  } else if (part_out != nullptr) {
+    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * Kt},
+                          x.options().dtype(at::kFloat));
+    p.part = part_out->data_ptr<float>();
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<1>, grid, dim3(512), 0,
                        cur_stream(), p);
   } else {
-    hipLaunchKernelGGL(conv_igemm_v2_kernel<false>, grid, dim3(512), 0,
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<0>, grid, dim3(512), 0,
                        cur_stream(), p);
   }
   return y;
@@ -361,5 +436,5 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
 
 at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                          int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
-  return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, 1, nullptr);
+  return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, 1, nullptr, nullptr);
 }

@@ -51,7 +51,8 @@ class _HIPConv2d(torch.autograd.Function):
     first time it sees its output (see conv2d/batch_norm_act below)."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, dilation, groups, emit_part):
+    def forward(ctx, x, w, bias, stride, padding, dilation, groups, emit_part,
+                bnx, bnscale, bnshift, bnact, bnslot):
         x = _cl(x)
         w = _cl(w)
         e = ext()
@@ -77,15 +78,21 @@ class _HIPConv2d(torch.autograd.Function):
             y = _cl(y[:, :kout])
         if bias is not None:
             y = y + bias.reshape(1, -1, 1, 1)
-        ctx.save_for_backward(x, w)
+        if bnx is not None and not cpad:
+            ctx.save_for_backward(x, w, bnx, bnscale, bnshift)
+        else:
+            ctx.save_for_backward(x, w)
         ctx.conf = (stride, padding, dilation, groups, bias is not None, cin,
                     kout, cpad, kpad)
+        ctx.bnact = bnact
+        ctx.bnslot = bnslot if (bnx is not None and not cpad) else None
         ctx.mark_non_differentiable(part)
         return y, part
 
     @staticmethod
     def backward(ctx, gy, _gpart):
-        x, w = ctx.saved_tensors
+        saved = ctx.saved_tensors
+        x, w = saved[0], saved[1]
         (stride, padding, dilation, groups, has_bias, cin, kout, cpad,
          kpad) = ctx.conf
         gy = _cl(gy)
@@ -96,9 +103,23 @@ class _HIPConv2d(torch.autograd.Function):
             wd = e.pad_channels(w, _pad8(cin)) if cpad else w
             if kpad:
                 wd = _pad_k(wd, _pad8(kout))
-            gx = e.conv2d_dgrad(gyp, wd, x.shape[2], x.shape[3],
-                                stride[0], stride[1], padding[0], padding[1],
-                                dilation[0], dilation[1], groups)
+            if ctx.bnslot is not None and len(saved) == 5:
+                # dgrad + BN-backward stats from the same epilogue: the
+                # consuming BN reads the partials IF this gx arrives there
+                # unmodified (data_ptr identity check in its backward)
+                bnx, bnscale, bnshift = saved[2], saved[3], saved[4]
+                gx, bpart = e.conv2d_dgrad_bn(
+                    gyp, wd, x.shape[2], x.shape[3], stride[0], stride[1],
+                    padding[0], padding[1], dilation[0], dilation[1], groups,
+                    bnx, bnscale, bnshift, ctx.bnact)
+                if bpart.numel():
+                    ctx.bnslot["parts"].append(bpart)
+                    ctx.bnslot["gx"] = gx
+            else:
+                gx = e.conv2d_dgrad(gyp, wd, x.shape[2], x.shape[3],
+                                    stride[0], stride[1], padding[0],
+                                    padding[1], dilation[0], dilation[1],
+                                    groups)
             if cpad:
                 gx = _cl(gx[:, :cin])
         if ctx.needs_input_grad[1]:
@@ -115,7 +136,8 @@ class _HIPConv2d(torch.autograd.Function):
                 gw = _cl(gw)
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
-        return gx, gw, gb, None, None, None, None, None
+        return (gx, gw, gb, None, None, None, None, None, None, None, None,
+                None, None)
 
 
 class _HIPDepthwiseConv2d(torch.autograd.Function):
@@ -204,8 +226,16 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
     if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
         emit = (bias is None and torch.is_grad_enabled()
                 and getattr(weight, "_emit_bn_partials", False))
+        info = (getattr(x, "_bn_bwd_info", None)
+                if torch.is_grad_enabled() else None)
+        if info is not None:
+            bnx, bnscale, bnshift, bnact, bnslot = info
+        else:
+            bnx = bnscale = bnshift = bnslot = None
+            bnact = 0
         y, part = _HIPConv2d.apply(x, weight, bias, stride, padding, dilation,
-                                   groups, emit)
+                                   groups, emit, bnx, bnscale, bnshift, bnact,
+                                   bnslot)
         if part.numel():
             y._bn_partials = part
         elif bias is None:
@@ -220,53 +250,56 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
 # BatchNorm (+residual add + activation) — K6/K8/K20 fused
 # ---------------------------------------------------------------------------
 class _HIPBatchNormAct(torch.autograd.Function):
-    """Training-mode fused BN: stats from the producing conv's epilogue
-    partials when available (F1 — no pass over x at all), else one per-channel
-    fp32 sums reduction kernel; then one normalize(+add)(+act) elementwise
-    kernel. Backward is one grad-stats reduction + one dx kernel, both
-    recomputing the act mask from x (F3a — the y stream is never re-read).
-    Running stats updated in fp32."""
+    """Training-mode fused BN. Stats (from conv-epilogue partials when
+    available — F1) are computed in the batch_norm_act wrapper; this Function
+    covers normalize(+add)(+act) forward and the backward chain, with the act
+    mask recomputed from x (F3a — the y stream is never re-read).
+
+    Backward grad-stats come from the CONSUMING conv's dgrad epilogue when it
+    emitted them (conv2d_dgrad_bn): the dgrad hands (partials, gx) through a
+    per-step slot dict, and they are used only if the incoming gy IS that gx
+    (data_ptr identity — any autograd accumulation allocates a new tensor, so
+    residual forks and multi-consumer graphs fall back automatically)."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, act, residual, part):
-        x = _cl(x)
+    def forward(ctx, x, weight, bias, mean, rstd, scale, shift, training,
+                act_id, residual, slot):
         e = ext()
-        act_id = _ACTS[act]
-        gamma = weight.float().contiguous()
-        beta = bias.float().contiguous()
-        rm, rv = running_mean, running_var
-        copy_back = False
-        if rm is not None and rm.dtype != torch.float32:
-            rm, rv = rm.float(), rv.float()
-            copy_back = True
-        # one fused kernel: sums -> mean/rstd/scale/shift + running update
-        mean, rstd, scale, shift = e.bn_stats(
-            x, gamma, beta, rm, rv, momentum, eps, training, part)
-        if copy_back:
-            with torch.no_grad():
-                running_mean.copy_(rm)
-                running_var.copy_(rv)
         res = _cl(residual) if residual is not None else None
         y = e.bn_apply_act(x, scale, shift, act_id, res)
-        ctx.save_for_backward(x, gamma, scale, shift, mean, rstd,
+        ctx.save_for_backward(x, weight, scale, shift, mean, rstd,
                               res if res is not None else x.new_empty(0))
         ctx.act_id = act_id
         ctx.training = training
         ctx.has_res = residual is not None
-        ctx.w_dtype = weight.dtype
+        ctx.slot = slot
         return y
 
     @staticmethod
     def backward(ctx, gy):
-        x, gamma, scale, shift, mean, rstd, res = ctx.saved_tensors
+        x, weight, scale, shift, mean, rstd, res = ctx.saved_tensors
         e = ext()
         gy = _cl(gy)
-        gx, gw, gb, gres = e.bn_bwd(
-            gy, x, res if ctx.has_res else None, mean, rstd, gamma,
-            scale, shift, ctx.act_id, ctx.training, ctx.has_res)
-        return (gx, gw.to(ctx.w_dtype), gb.to(ctx.w_dtype), None, None, None,
-                None, None, None, gres if ctx.has_res else None, None)
+        gamma = weight.float().contiguous()
+        slot = ctx.slot
+        part = None
+        if (slot is not None and slot.get("gx") is not None
+                and len(slot["parts"]) == 1
+                and gy.data_ptr() == slot["gx"].data_ptr()):
+            part = slot["parts"][0]
+        if part is not None:
+            c = x.shape[1]
+            sums = e.bn_reduce_partials(part)
+            gx, gw, gb, gres = e.bn_bwd_apply(
+                gy, x, res if ctx.has_res else None, mean, rstd, gamma,
+                scale, shift, sums, float(x.numel() // c), ctx.act_id,
+                ctx.training, ctx.has_res)
+        else:
+            gx, gw, gb, gres = e.bn_bwd(
+                gy, x, res if ctx.has_res else None, mean, rstd, gamma,
+                scale, shift, ctx.act_id, ctx.training, ctx.has_res)
+        return (gx, gw.to(weight.dtype), gb.to(weight.dtype), None, None,
+                None, None, None, None, gres if ctx.has_res else None, None)
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
@@ -278,9 +311,33 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
             if srcw is not None:
                 # flag the producing conv: it emits partials from now on
                 srcw._emit_bn_partials = True
-        return _HIPBatchNormAct.apply(x, weight, bias, running_mean,
-                                      running_var, training, momentum, eps,
-                                      act, residual, part)
+        e = ext()
+        act_id = _ACTS[act]
+        x = _cl(x)
+        rm, rv = running_mean, running_var
+        with torch.no_grad():
+            gamma = weight.float().contiguous()
+            beta = bias.float().contiguous()
+            copy_back = False
+            if rm is not None and rm.dtype != torch.float32:
+                rm, rv = rm.float(), rv.float()
+                copy_back = True
+            mean, rstd, scale, shift = e.bn_stats(
+                x, gamma, beta, rm, rv, momentum, eps, training, part)
+            if copy_back:
+                running_mean.copy_(rm)
+                running_var.copy_(rv)
+        slot = None
+        if (training and residual is None and act_id in (0, 1)
+                and torch.is_grad_enabled()):
+            slot = {"gx": None, "parts": []}
+        y = _HIPBatchNormAct.apply(x, weight, bias, mean, rstd, scale, shift,
+                                   training, act_id, residual, slot)
+        if slot is not None:
+            # consuming convs pick this up and emit BN-backward stats from
+            # their dgrad epilogue (see _HIPConv2d / conv2d_dgrad_bn)
+            y._bn_bwd_info = (x, scale, shift, act_id, slot)
+        return y
     y = F.batch_norm(x, running_mean, running_var, weight, bias, training,
                      momentum, eps)
     if residual is not None:
