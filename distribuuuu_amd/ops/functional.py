@@ -6,6 +6,8 @@ NHWC (PyTorch ``channels_last``): logical NCHW with [N,H,W,C] physical layout,
 weights [K,C,R,S] logical → [K,R,S,C] physical.
 """
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -327,9 +329,15 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
             if copy_back:
                 running_mean.copy_(rm)
                 running_var.copy_(rv)
+        # dgrad-side backward-stat emission is measured NET-NEGATIVE on
+        # ResNet-50 (the EMODE-2 epilogue's extra x stream + barrier
+        # serialization costs ~1.6 ms/step while removing only ~0.8 ms of
+        # bn_bwd_reduce — the residual BNs, the big tensors, can't fuse).
+        # Kept opt-in for shapes where it may win; tests force it on.
         slot = None
         if (training and residual is None and act_id in (0, 1)
-                and torch.is_grad_enabled()):
+                and torch.is_grad_enabled()
+                and os.environ.get("DISTRIBUUUU_BN_BWD_FUSE", "0") == "1"):
             slot = {"gx": None, "parts": []}
         y = _HIPBatchNormAct.apply(x, weight, bias, mean, rstd, scale, shift,
                                    training, act_id, residual, slot)
