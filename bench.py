@@ -119,10 +119,12 @@ def main():
     for _ in range(args.warmup):
         step()
 
-    # hipGraph capture of the whole train step (single-GPU): the inner loop
-    # replays one graph instead of ~900 eager launches.
-    use_graph = (world_size == 1 and has_gpu) if args.graph == -1 else bool(
-        args.graph)
+    # hipGraph capture of the whole train step: the inner loop replays one
+    # graph instead of ~900 eager launches. Multi-rank too — RCCL collectives
+    # (DDP buckets, SyncBN stats) are capture-legal on ROCm, and an
+    # eager-launch-bound rank would cap scaling; capture failure falls back
+    # to eager below.
+    use_graph = bool(has_gpu) if args.graph == -1 else bool(args.graph)
     graph = None
     if use_graph:
         try:
@@ -135,10 +137,20 @@ def main():
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 step()
+            graph.replay()
+            torch.cuda.synchronize()
         except Exception as exc:  # pragma: no cover - fall back to eager
             print(f"[bench] graph capture failed ({exc}); eager mode",
                   flush=True)
             graph = None
+        if world_size > 1:
+            # every rank must agree (a replaying rank paired with an eager
+            # one still matches collective order, but keep it uniform)
+            ok = torch.tensor([1.0 if graph is not None else 0.0],
+                              device=device)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 1.0:
+                graph = None
 
     if world_size > 1:
         dist.barrier()
