@@ -3,13 +3,11 @@
 // applied on the SOURCE address (glds writes lane-linear; platform guide
 // §5.4 rule 21), double-buffered, counted-barrier loop.
 //
-// Boundary handling: A-gather addresses are CLAMPED to a 16-byte zero
-// buffer when out of bounds (address select — no post-load cndmask, no
-// early waitcnt; the conv_smallk_pipe idiom), so the image is read directly
-// from HBM with no host-side pad or physical-order copy pass. The weight's
-// reduction span is zero-padded to a multiple of 64 (pad_weight_span) so
-// span-tail A garbage multiplies zero. M/K tile tails clamp their source
-// rows (garbage rows are computed but never stored).
+// Predication-free inner loop: the host pads the input image in HBM
+// (pad_image) and zero-pads the weight's reduction span to a multiple of 64
+// (pad_weight_span), so every gather address is in-bounds and span-tail
+// garbage in A multiplies a zero in B. M/K tile tails clamp their source rows
+// (garbage rows are computed but never stored).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4c;
@@ -24,14 +22,13 @@ constexpr int BUF_B = ATILE_B + BTILE_B;        // 48 KB per ring slot
 constexpr int NBUF = 3;                         // 144 KB ring
 
 struct Conv2Params {
-  const __hip_bfloat16* x;     // [N, H, W, Ct] — UNPADDED (address-clamped)
-  const __hip_bfloat16* w;     // span-padded [Kt, R, SPAN64]
-  const __hip_bfloat16* zbuf;  // 16 zero bytes: clamp target for OOB loads
-  __hip_bfloat16* y;           // [N, Ho, Wo, Kt]
-  int N, H, W, C, K;           // C/K are PER-GROUP (Cg/Kg)
-  int Ct, Kt;                  // totals (group stride in x/w/y)
+  const __hip_bfloat16* x;  // padded [N, Hp, Wp, Ct] (+slack)
+  const __hip_bfloat16* w;  // span-padded [Kt, R, SPAN64]
+  __hip_bfloat16* y;        // [N, Ho, Wo, Kt]
+  int N, Hp, Wp, C, K;      // C/K are PER-GROUP (Cg/Kg)
+  int Ct, Kt;               // totals (group stride in x/w/y)
   int R, SPAN64, Cg, S;
-  int sh, sw, ph, pw, dh, dw;
+  int sh, sw, dh, dw;
   int Ho, Wo;
   int M, nspan, ksteps;
   int tiles_m;
@@ -71,7 +68,7 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
   // A: 4 slots/thread (32 KB); B: 2 slots/thread (16 KB). Stored slot
   // s = tid + it*512; logical slot sl = s ^ (((s>>5)&1)<<1);
   // row = sl>>3, span-col16 = sl&7.
-  int a_n[4], a_h[4], a_w[4], a_sl[4];
+  int a_n[4], a_hwbase[4], a_sl[4];
   const int HoWo = p.Ho * p.Wo;
 #pragma unroll
   for (int it = 0; it < 4; ++it) {
@@ -83,8 +80,7 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
     const int n = m / HoWo;
     const int rem = m - n * HoWo;
     a_n[it] = n;
-    a_h[it] = (rem / p.Wo) * p.sh - p.ph;
-    a_w[it] = (rem % p.Wo) * p.sw - p.pw;
+    a_hwbase[it] = ((rem / p.Wo) * p.sh) * p.Wp + (rem % p.Wo) * p.sw;
   }
   int b_base[2], b_sl[2];
 #pragma unroll
@@ -107,13 +103,11 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
       const int span = span0 + (a_sl[it] & 7) * 8;
       const int s_ = span / p.Cg;   // may reach S at the padded tail
       const int c = span - s_ * p.Cg;
-      const int h = a_h[it] + r * p.dh;
-      const int w_ = a_w[it] + s_ * p.dw;
-      const bool ok = h >= 0 && h < p.H && w_ >= 0 && w_ < p.W;
       const __hip_bfloat16* src =
-          ok ? p.x + (((int64_t)a_n[it] * p.H + h) * p.W + w_) * p.Ct +
-                   g * p.Cg + c
-             : p.zbuf;
+          p.x +
+          ((int64_t)a_n[it] * p.Hp * p.Wp +
+           (a_hwbase[it] + r * p.dh * p.Wp + s_ * p.dw)) * p.Ct +
+          g * p.Cg + c;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)src,
           (__attribute__((address_space(3))) uint32_t*)(base + it * 8192),
@@ -357,11 +351,34 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   const int SC = S * Cg;
   const int SPAN64 = (SC + BK2 - 1) / BK2 * BK2;
 
-  // the kernel clamps out-of-bounds A gathers to a zero buffer, so the
-  // image is consumed DIRECTLY — no host pad pass, no physical-order copy
-  static at::Tensor zbuf;  // 16 zero bytes; allocated outside any capture
-  if (!zbuf.defined() || zbuf.device() != x.device())
-    zbuf = at::zeros({8}, x.options());
+  // padded input (+ slack for span-tail over-read)
+  at::Tensor xin = x;
+  int Hp = H, Wp = W;
+  if (ph > 0 || pw > 0) {
+    Hp = H + 2 * ph;
+    Wp = W + 2 * pw;
+    auto xp = at::empty({(int64_t)N * Hp * Wp * Ct + SPAN64 + 64},
+                        x.options());
+    int64_t total = (int64_t)N * Hp * Wp * Ct;
+    // span-tail reads of the LAST pixel land in the slack: it must be finite
+    // (tail A values multiply zero-padded B weights, but Inf*0 = NaN)
+    xp.narrow(0, total, SPAN64 + 64).zero_();
+    hipLaunchKernelGGL((pad_image_kernel<__hip_bfloat16>),
+                       dim3((Wp * Ct / 8 + 255) / 256, N * Hp), dim3(256), 0,
+                       cur_stream(),
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (__hip_bfloat16*)xp.data_ptr(), N, H, W, Ct, Hp, Wp, ph,
+                       pw);
+    xin = xp;
+  } else if (SPAN64 != SC) {
+    // slack for span-tail reads past the logical end
+    auto xp = at::empty({(int64_t)N * H * W * Ct + SPAN64 + 64}, x.options());
+    // copy in PHYSICAL (NHWC) order; zero the slack (read by span tails)
+    xp.narrow(0, 0, (int64_t)N * H * W * Ct)
+        .copy_(x.permute({0, 2, 3, 1}).reshape({-1}));
+    xp.narrow(0, (int64_t)N * H * W * Ct, SPAN64 + 64).zero_();
+    xin = xp;
+  }  // else: no padding, no span tail -> use x directly
   // span-padded weight
   at::Tensor wp = w;
   if (SPAN64 != SC) {
@@ -376,14 +393,13 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   Conv2Params p;
-  p.x = (const __hip_bfloat16*)x.data_ptr();
+  p.x = (const __hip_bfloat16*)xin.data_ptr();
   p.w = (const __hip_bfloat16*)wp.data_ptr();
-  p.zbuf = (const __hip_bfloat16*)zbuf.data_ptr();
   p.y = (__hip_bfloat16*)y.data_ptr();
-  p.N = N; p.H = H; p.W = W; p.C = C; p.K = K;
+  p.N = N; p.Hp = Hp; p.Wp = Wp; p.C = C; p.K = K;
   p.Ct = Ct; p.Kt = Kt;
   p.R = R; p.SPAN64 = SPAN64; p.Cg = C; p.S = S;
-  p.sh = sh; p.sw = sw; p.ph = ph; p.pw = pw; p.dh = dh; p.dw = dw;
+  p.sh = sh; p.sw = sw; p.dh = dh; p.dw = dw;
   p.Ho = Ho; p.Wo = Wo;
   p.M = N * Ho * Wo;
   p.nspan = SPAN64 / BK2;
