@@ -28,18 +28,32 @@ DEV_INLINE bf16x8a zero8() {
 
 constexpr int QT = 64;  // q rows per block
 
+// q/k/v/o rows address as base + ((n*L + l) * pix + hd * D): with heads=1
+// and pix=D this is the dense [B, L, D] layout; with heads>1 it reads the
+// qkv convs' channels_last output IN PLACE ([N, L, heads*D] physical — no
+// chunk/transpose/contiguous copies) and writes o back the same way.
+// `scale` folds the q pre-scale into the logits (and rel tables).
 struct MhsaParams {
-  const __hip_bfloat16* q;   // [B, L, D]
-  const __hip_bfloat16* k;   // [B, L, D]
-  const __hip_bfloat16* vt;  // [B, D, L]
+  const __hip_bfloat16* q;
+  const __hip_bfloat16* k;
+  const __hip_bfloat16* vt;  // [B, D, L] (transposed operand, dense)
   const float* rw;           // [B, L, 2W-1]
   const float* rh;           // [B, L, 2H-1]
-  __hip_bfloat16* o;         // [B, L, D]
+  __hip_bfloat16* o;
   __hip_bfloat16* pout;      // optional [B, L, L] softmax probs (backward)
   int B, L, D, H, W;
+  int heads;
+  int64_t qpix, kpix, vpix, opix;  // elements per pixel in each tensor
+  float scale;
   int ltiles16;  // ceil(L/16)
   int lpad;      // LDS row width for S (multiple of 16 + pad)
 };
+
+DEV_INLINE int64_t strided_row(int b, int l, int L, int heads, int64_t pix,
+                               int D) {
+  const int n = b / heads, hd = b - n * heads;
+  return ((int64_t)n * L + l) * pix + (int64_t)hd * D;
+}
 
 // Logits live in REGISTERS (LT tiles x 4 rows per lane); LDS holds only the
 // bf16 P tile for the O pass. The round-1 version kept a [QT][lpad] fp32
@@ -64,7 +78,8 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
   {
     const int qr = qrow_base + il;
     const bool ok = qr < p.L;
-    const __hip_bfloat16* qp = p.q + ((int64_t)b * p.L + (ok ? qr : 0)) * p.D;
+    const __hip_bfloat16* qp =
+        p.q + strided_row(b, ok ? qr : 0, p.L, p.heads, p.qpix, p.D);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       if (c < nd && ok)
@@ -84,7 +99,7 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
       const int kr = nt * 16 + il;
       const bool kok = kr < p.L;
       const __hip_bfloat16* kp =
-          p.k + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
+          p.k + strided_row(b, kok ? kr : 0, p.L, p.heads, p.kpix, p.D);
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         if (c >= nd) break;
@@ -109,7 +124,7 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
 #pragma unroll
     for (int nt = 0; nt < LT; ++nt) {
       const int kpos = nt * 16 + il;
-      float v = sacc[nt][rr];
+      float v = sacc[nt][rr] * p.scale;
       if (qok && kpos < p.L) {
         const int dy = (kpos % p.W) - wi + p.W - 1;
         const int dx = (kpos / p.W) - hi + p.H - 1;
@@ -200,7 +215,7 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
     for (int rr = 0; rr < 4; ++rr) {
       const int qr = qrow_base + kq * 4 + rr;
       if (qr < p.L)
-        p.o[((int64_t)b * p.L + qr) * p.D + dt * 16 + il] =
+        p.o[strided_row(b, qr, p.L, p.heads, p.opix, p.D) + dt * 16 + il] =
             from_f32<__hip_bfloat16>(acc[rr]);
     }
   }
@@ -222,17 +237,20 @@ __global__ __launch_bounds__(256) void mhsa_fwd_kernel(MhsaParams p) {
 //   per-block register accumulation + one atomic flush.
 // ---------------------------------------------------------------------------
 struct MhsaBwdQParams {
-  const __hip_bfloat16* dO;  // [B, L, D]
+  const __hip_bfloat16* dO;  // strided rows (opix)
   const __hip_bfloat16* P;   // [B, L, L]
-  const __hip_bfloat16* v;   // [B, L, D] (rows; B-operand of dP)
-  const __hip_bfloat16* kt;  // [B, D, L] (B-operand of dQ)
+  const __hip_bfloat16* v;   // strided rows (vpix; B-operand of dP)
+  const __hip_bfloat16* kt;  // [B, D, L] (B-operand of dQ, dense)
   const __hip_bfloat16* rw;  // rel_w [2W-1, D]
   const __hip_bfloat16* rh;  // rel_h [2H-1, D]
-  __hip_bfloat16* dq;        // [B, L, D]
+  __hip_bfloat16* dq;        // strided rows (qpix)
   __hip_bfloat16* ds;        // [B, L, L]
   float* drw;                // [B, L, 2W-1]
   float* drh;                // [B, L, 2H-1]
   int B, L, D, H, W;
+  int heads;
+  int64_t qpix, vpix, opix;
+  float scale;               // dq = scale * (dS kT + rel fold)
   int ltiles16, lpad;
 };
 
@@ -258,7 +276,7 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
     const int qr = qrow_base + il;
     const bool ok = qr < p.L;
     const __hip_bfloat16* dop =
-        p.dO + ((int64_t)b * p.L + (ok ? qr : 0)) * p.D;
+        p.dO + strided_row(b, ok ? qr : 0, p.L, p.heads, p.opix, p.D);
 #pragma unroll
     for (int c = 0; c < 4; ++c)
       dofrag[c] = (c < nd && ok)
@@ -275,7 +293,7 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
       const int kr = nt * 16 + il;
       const bool kok = kr < p.L;
       const __hip_bfloat16* vp =
-          p.v + ((int64_t)b * p.L + (kok ? kr : 0)) * p.D;
+          p.v + strided_row(b, kok ? kr : 0, p.L, p.heads, p.vpix, p.D);
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         if (c >= nd) break;
@@ -395,8 +413,8 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
           out += rws[m] * to_f32(p.rw[m * p.D + d]);
         for (int m = 0; m < 2 * p.H - 1; ++m)
           out += rhs[m] * to_f32(p.rh[m * p.D + d]);
-        p.dq[((int64_t)b * p.L + qr) * p.D + d] =
-            from_f32<__hip_bfloat16>(out);
+        p.dq[strided_row(b, qr, p.L, p.heads, p.qpix, p.D) + d] =
+            from_f32<__hip_bfloat16>(out * p.scale);
       }
     }
   }
@@ -405,12 +423,15 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
 // ---- dK = dS^T q, dV = P^T dO ---------------------------------------------
 struct MhsaBwdKVParams {
   const __hip_bfloat16* ds;  // [B, L, L]
-  const __hip_bfloat16* q;   // [B, L, D]
+  const __hip_bfloat16* q;   // strided rows (qpix)
   const __hip_bfloat16* P;   // [B, L, L]
-  const __hip_bfloat16* dO;  // [B, L, D]
-  __hip_bfloat16* dk;        // [B, L, D]
-  __hip_bfloat16* dv;        // [B, L, D]
+  const __hip_bfloat16* dO;  // strided rows (opix)
+  __hip_bfloat16* dk;        // strided rows (kpix)
+  __hip_bfloat16* dv;        // strided rows (vpix)
   int B, L, D;
+  int heads;
+  int64_t qpix, kpix, vpix, opix;
+  float scale;               // dk = scale * (dS^T q)
 };
 
 __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
@@ -433,11 +454,15 @@ __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
       const int r = tid >> 4, c8 = (tid & 15) << 3;
       const int qi = ch * 16 + r;
       const bool ok = qi < p.L && c8 < p.D;
-      const int64_t rowoff = ((int64_t)b * p.L + (qi < p.L ? qi : 0)) * p.D;
+      const int qic = qi < p.L ? qi : 0;
       *reinterpret_cast<bf16x8a*>(&q16[r][c8]) =
-          ok ? *reinterpret_cast<const bf16x8a*>(p.q + rowoff + c8) : zero8();
+          ok ? *reinterpret_cast<const bf16x8a*>(
+                   p.q + strided_row(b, qic, p.L, p.heads, p.qpix, p.D) + c8)
+             : zero8();
       *reinterpret_cast<bf16x8a*>(&do16[r][c8]) =
-          ok ? *reinterpret_cast<const bf16x8a*>(p.dO + rowoff + c8) : zero8();
+          ok ? *reinterpret_cast<const bf16x8a*>(
+                   p.dO + strided_row(b, qic, p.L, p.heads, p.opix, p.D) + c8)
+             : zero8();
     }
     if (tid < 128) {
       const int r = tid >> 3, c8 = (tid & 7) << 3;
@@ -481,11 +506,13 @@ __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
     __syncthreads();
   }
   if (kj < p.L && d0 < p.D) {
-    __hip_bfloat16* dkp = p.dk + ((int64_t)b * p.L + kj) * p.D + d0;
-    __hip_bfloat16* dvp = p.dv + ((int64_t)b * p.L + kj) * p.D + d0;
+    __hip_bfloat16* dkp =
+        p.dk + strided_row(b, kj, p.L, p.heads, p.kpix, p.D) + d0;
+    __hip_bfloat16* dvp =
+        p.dv + strided_row(b, kj, p.L, p.heads, p.vpix, p.D) + d0;
 #pragma unroll
     for (int u = 0; u < 32; ++u) {
-      dkp[u] = from_f32<__hip_bfloat16>(acck[u]);
+      dkp[u] = from_f32<__hip_bfloat16>(acck[u] * p.scale);
       dvp[u] = from_f32<__hip_bfloat16>(accv[u]);
     }
   }
@@ -494,10 +521,13 @@ __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
 // ---- grw[m, d] = sum_rows dRW[row, m] * q[row, d] -------------------------
 struct MhsaRelGradParams {
   const float* dr;          // [rows, M] (dRW or dRH)
-  const __hip_bfloat16* q;  // [rows, D]
+  const __hip_bfloat16* q;  // strided rows
   float* out;               // [M, D]
   int64_t rows;
   int M, D, rows_per_block;
+  int heads, L;
+  int64_t qpix;
+  float scale;              // grw = scale * (dRW^T q)
 };
 
 __global__ __launch_bounds__(256) void mhsa_rel_grad_kernel(
@@ -517,24 +547,29 @@ __global__ __launch_bounds__(256) void mhsa_rel_grad_kernel(
   const int64_t r0 = (int64_t)blockIdx.x * p.rows_per_block;
   const int64_t r1 = min(r0 + p.rows_per_block, p.rows);
   for (int64_t r = r0; r < r1; ++r) {
-    const float qv = to_f32(p.q[r * p.D + d]);
+    const int bb = (int)(r / p.L), ll = (int)(r - (int64_t)bb * p.L);
+    const float qv =
+        to_f32(p.q[strided_row(bb, ll, p.L, p.heads, p.qpix, p.D) + d]);
     const float* dr = p.dr + r * p.M + m0;
 #pragma unroll 7
     for (int m = 0; m < mn; ++m) acc[m] += dr[m] * qv;
   }
   for (int m = 0; m < mn; ++m)
-    atomicAdd(&p.out[(int64_t)(m0 + m) * p.D + d], acc[m]);
+    atomicAdd(&p.out[(int64_t)(m0 + m) * p.D + d], acc[m] * p.scale);
 }
 
 // ---- per-row rel-logit tables: out[row, m] = sum_d q[row, d] rel[m, d] ----
 // fp32 accumulate AND fp32 output (a bf16 GEMM output rounds the table and
 // measurably shifts the softmax when logits are large).
 struct RelTabParams {
-  const __hip_bfloat16* q;    // [rows, D]
+  const __hip_bfloat16* q;    // strided rows (qpix/heads; dense when heads=1)
   const __hip_bfloat16* rel;  // [M, D]
   float* out;                 // [rows, M]
   int64_t rows;
   int M, D;
+  int heads, L;
+  int64_t qpix;
+  float scale;                // tables of the PRE-SCALED q
 };
 
 __global__ __launch_bounds__(256) void mhsa_rel_tables_kernel(
@@ -550,16 +585,20 @@ __global__ __launch_bounds__(256) void mhsa_rel_tables_kernel(
     if (threadIdx.x < 16 * (p.D / 8)) {
       const int rr = threadIdx.x / (p.D / 8);
       const int c8 = (threadIdx.x % (p.D / 8)) * 8;
-      if (rr < 8 && row0 + rr < p.rows)
+      if (rr < 8 && row0 + rr < p.rows) {
+        const int64_t ro = row0 + rr;
+        const int bb = (int)(ro / p.L), ll = (int)(ro - (int64_t)bb * p.L);
         *reinterpret_cast<bf16x8a*>(&q_s[rr][c8]) =
-            *reinterpret_cast<const bf16x8a*>(p.q + (row0 + rr) * p.D + c8);
+            *reinterpret_cast<const bf16x8a*>(
+                p.q + strided_row(bb, ll, p.L, p.heads, p.qpix, p.D) + c8);
+      }
     }
     __syncthreads();
     if (m < p.M && row0 + rl < p.rows) {
       float acc = 0.f;
       for (int dd = 0; dd < p.D; ++dd)
         acc += to_f32(q_s[rl][dd]) * to_f32(rel_s[m * p.D + dd]);
-      p.out[(row0 + rl) * p.M + m] = acc;
+      p.out[(row0 + rl) * p.M + m] = acc * p.scale;
     }
     __syncthreads();
   }
@@ -567,20 +606,27 @@ __global__ __launch_bounds__(256) void mhsa_rel_tables_kernel(
 
 }  // namespace
 
-// rw = q @ rel^T, fp32 out: [rows, M]
-at::Tensor mhsa_rel_tables(at::Tensor q2d, at::Tensor rel) {
-  CHECK_GPU(q2d);
-  const int64_t rows = q2d.size(0);
-  const int D = q2d.size(1), M = rel.size(0);
+// rw = (scale*q) @ rel^T, fp32 out [rows, M]; q rows address as
+// base + ((row/L)*L + row%L ... ) * qpix + (b%heads)*D (dense when heads=1,
+// qpix=D, L=1).
+at::Tensor mhsa_rel_tables(at::Tensor q, at::Tensor rel, int64_t rows,
+                           int64_t heads, int64_t L, int64_t qpix,
+                           double scale) {
+  CHECK_GPU(q);
+  const int D = rel.size(1), M = rel.size(0);
   TORCH_CHECK(M <= 32 && D <= 128 && D % 8 == 0, "rel tables: M<=32 D<=128");
-  auto out = at::empty({rows, (int64_t)M}, q2d.options().dtype(at::kFloat));
+  auto out = at::empty({rows, (int64_t)M}, q.options().dtype(at::kFloat));
   RelTabParams p;
-  p.q = (const __hip_bfloat16*)q2d.data_ptr();
+  p.q = (const __hip_bfloat16*)q.data_ptr();
   p.rel = (const __hip_bfloat16*)rel.data_ptr();
   p.out = out.data_ptr<float>();
   p.rows = rows;
   p.M = M;
   p.D = D;
+  p.heads = (int)heads;
+  p.L = (int)L;
+  p.qpix = qpix;
+  p.scale = (float)scale;
   hipLaunchKernelGGL(mhsa_rel_tables_kernel,
                      dim3((int)std::min<int64_t>(ceil_div(rows, 8), 2048)),
                      dim3(256), 0, cur_stream(), p);
@@ -589,22 +635,29 @@ at::Tensor mhsa_rel_tables(at::Tensor q2d, at::Tensor rel) {
 
 // Backward: dO + saved (P, q, k, v) -> dq (incl. rel-pos fold), dk, dv,
 // grw, grh. kt/vt are the transposed operands ([B, D, L]).
+// Strided/NHWC backward: q/dO/v (and the dq/dk/dv outputs) live in the qkv
+// convs' channels_last layouts; dqk_out is one [N, 2*heads*D, H, W] tensor
+// receiving dq and dk halves (exactly what the qk conv's backward wants).
+// heads=1 + pix=D degenerates to the dense layout.
 std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
                                  at::Tensor kt, at::Tensor v,
                                  at::Tensor rel_w, at::Tensor rel_h,
-                                 int64_t H, int64_t W) {
+                                 int64_t H, int64_t W, int64_t heads,
+                                 int64_t qpix, int64_t kqoff, int64_t vpix,
+                                 int64_t opix, double scale,
+                                 at::Tensor dqk_out, at::Tensor dv_out) {
   CHECK_GPU(dO);
-  const int B = q.size(0), L = q.size(1), D = q.size(2);
+  const int B = P.size(0), L = P.size(1);
+  const int D = rel_w.size(1);
   const int MW = 2 * W - 1, MH = 2 * H - 1;
   TORCH_CHECK(MW <= 32 && MH <= 32, "mhsa_bwd: rel table > 32");
-  auto dq = at::empty_like(q);
   auto ds = at::empty_like(P);
-  auto dk = at::empty_like(q);
-  auto dv = at::empty_like(q);
-  auto fopts = q.options().dtype(at::kFloat);
+  auto fopts = P.options().dtype(at::kFloat);
   auto drw = at::empty({(int64_t)B * L, MW}, fopts);
   auto drh = at::empty({(int64_t)B * L, MH}, fopts);
   const int lpad = ((L + 31) / 32) * 32;
+  __hip_bfloat16* dq_ptr = (__hip_bfloat16*)dqk_out.data_ptr();
+  __hip_bfloat16* dk_ptr = dq_ptr + kqoff;
   MhsaBwdQParams p;
   p.dO = (const __hip_bfloat16*)dO.data_ptr();
   p.P = (const __hip_bfloat16*)P.data_ptr();
@@ -612,11 +665,14 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
   p.kt = (const __hip_bfloat16*)kt.data_ptr();
   p.rw = (const __hip_bfloat16*)rel_w.data_ptr();
   p.rh = (const __hip_bfloat16*)rel_h.data_ptr();
-  p.dq = (__hip_bfloat16*)dq.data_ptr();
+  p.dq = dq_ptr;
   p.ds = (__hip_bfloat16*)ds.data_ptr();
   p.drw = drw.data_ptr<float>();
   p.drh = drh.data_ptr<float>();
   p.B = B; p.L = L; p.D = D; p.H = H; p.W = W;
+  p.heads = (int)heads;
+  p.qpix = qpix; p.vpix = vpix; p.opix = opix;
+  p.scale = (float)scale;
   p.ltiles16 = (L + 15) / 16;
   p.lpad = lpad;
   TORCH_CHECK(p.ltiles16 <= 16, "mhsa_bwd: L must be <= 256");
@@ -640,9 +696,12 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
   kv.q = (const __hip_bfloat16*)q.data_ptr();
   kv.P = (const __hip_bfloat16*)P.data_ptr();
   kv.dO = (const __hip_bfloat16*)dO.data_ptr();
-  kv.dk = (__hip_bfloat16*)dk.data_ptr();
-  kv.dv = (__hip_bfloat16*)dv.data_ptr();
+  kv.dk = dk_ptr;
+  kv.dv = (__hip_bfloat16*)dv_out.data_ptr();
   kv.B = B; kv.L = L; kv.D = D;
+  kv.heads = (int)heads;
+  kv.qpix = qpix; kv.kpix = qpix; kv.vpix = vpix; kv.opix = opix;
+  kv.scale = (float)scale;
   dim3 gkv(B, (L + 63) / 64);
   hipLaunchKernelGGL(mhsa_bwd_kv_kernel, gkv, dim3(256), 0, cur_stream(), kv);
 
@@ -659,23 +718,41 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
     rp.M = which == 0 ? MW : MH;
     rp.D = D;
     rp.rows_per_block = rpb;
+    rp.heads = (int)heads;
+    rp.L = L;
+    rp.qpix = qpix;
+    rp.scale = (float)scale;
     hipLaunchKernelGGL(mhsa_rel_grad_kernel,
                        dim3((int)ceil_div(rows, rpb)), dim3(256), 0,
                        cur_stream(), rp);
   }
-  return {dq, dk, dv, grw, grh};
+  return {dqk_out, dv_out, grw, grh};
 }
 
-// q pre-scaled; returns O [B, L, D]
+// Geometry from vt [B, D, L] (the one dense operand). heads=1, pix=D is the
+// dense [B, L, D] layout; heads>1 reads q/k from the qkv convs'
+// channels_last output in place and writes o as channels_last
+// [N, heads*D, H, W]. q scale is folded into the logits.
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
                     at::Tensor rh, int64_t H, int64_t W,
-                    c10::optional<at::Tensor> pout) {
+                    c10::optional<at::Tensor> pout, int64_t heads,
+                    int64_t qpix, int64_t kpix, double scale) {
   CHECK_GPU(q);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "mhsa: bf16 only");
-  const int B = q.size(0), L = q.size(1), D = q.size(2);
+  const int B = vt.size(0), D = vt.size(1), L = vt.size(2);
   TORCH_CHECK(D % 32 == 0 && D <= 128, "mhsa: D must be <=128, %32==0");
   TORCH_CHECK(L == H * W, "L != H*W");
-  auto o = at::empty_like(q);
+  const int N = B / (int)heads;
+  at::Tensor o;
+  int64_t opix;
+  if (heads > 1) {
+    o = at::empty({N, heads * D, H, W},
+                  q.options().memory_format(at::MemoryFormat::ChannelsLast));
+    opix = heads * D;
+  } else {
+    o = at::empty({B, (int64_t)L, (int64_t)D}, q.options());
+    opix = D;
+  }
   MhsaParams p;
   p.q = (const __hip_bfloat16*)q.data_ptr();
   p.k = (const __hip_bfloat16*)k.data_ptr();
@@ -685,6 +762,9 @@ at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
   p.o = (__hip_bfloat16*)o.data_ptr();
   p.pout = pout.has_value() ? (__hip_bfloat16*)pout->data_ptr() : nullptr;
   p.B = B; p.L = L; p.D = D; p.H = H; p.W = W;
+  p.heads = (int)heads;
+  p.qpix = qpix; p.kpix = kpix; p.vpix = 0; p.opix = opix;
+  p.scale = (float)scale;
   p.ltiles16 = (L + 15) / 16;
   p.lpad = ((L + 31) / 32) * 32;  // multiple of 32 for the P fragment chunks
   TORCH_CHECK(p.ltiles16 <= 16, "mhsa: L must be <= 256");

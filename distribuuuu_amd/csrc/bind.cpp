@@ -83,12 +83,18 @@ at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
 // attention.hip
 at::Tensor mhsa_fwd(at::Tensor q, at::Tensor k, at::Tensor vt, at::Tensor rw,
                     at::Tensor rh, int64_t H, int64_t W,
-                    c10::optional<at::Tensor> pout);
+                    c10::optional<at::Tensor> pout, int64_t heads,
+                    int64_t qpix, int64_t kpix, double scale);
 std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
                                  at::Tensor kt, at::Tensor v,
                                  at::Tensor rel_w, at::Tensor rel_h,
-                                 int64_t H, int64_t W);
-at::Tensor mhsa_rel_tables(at::Tensor q2d, at::Tensor rel);
+                                 int64_t H, int64_t W, int64_t heads,
+                                 int64_t qpix, int64_t kqoff, int64_t vpix,
+                                 int64_t opix, double scale,
+                                 at::Tensor dqk_out, at::Tensor dv_out);
+at::Tensor mhsa_rel_tables(at::Tensor q, at::Tensor rel, int64_t rows,
+                           int64_t heads, int64_t L, int64_t qpix,
+                           double scale);
 // augment.hip
 at::Tensor aug_crop_flip_norm(at::Tensor raw, at::Tensor meta, int64_t S,
                               std::vector<double> mean,

@@ -17,7 +17,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import AdaptiveAvgPool2d, AvgPool2d, BatchNorm2d, Conv2d, Linear
-from ..ops.attention import mhsa_relpos
+from ..ops.attention import mhsa_relpos, mhsa_relpos_nhwc
 from .resnet import resnet50
 
 
@@ -54,6 +54,16 @@ class MHSA(nn.Module):
         n, _, hh, ww = x.shape
         qk = self.to_qk(x)
         v = self.to_v(x)
+        if (self.rel_pos_emb and self.dim_qk == self.dim_v
+                and qk.is_cuda and qk.dtype == torch.bfloat16
+                and self.dim_qk % 32 == 0 and self.dim_qk <= 128):
+            from ..ops.dispatch import use_hip
+            if use_hip(qk, "mhsa_fwd"):
+                # NHWC in-place path: the qkv convs' channels_last outputs
+                # feed the kernels directly (no chunk/permute copies)
+                return mhsa_relpos_nhwc(qk, v, self.rel_h, self.rel_w,
+                                        self.heads, self.dim_qk, self.dim_v,
+                                        hh, ww, self.scale)
         q, k = qk.chunk(2, dim=1)
         # [N, heads, L, d]
         q = q.reshape(n, self.heads, self.dim_qk, hh * ww).transpose(2, 3)

@@ -95,14 +95,15 @@ class _HIPMHSARelPos(torch.autograd.Function):
         # per-row relative-logit tables RW = q@rel_w^T, RH = q@rel_h^T —
         # dedicated kernel with fp32 accumulate AND fp32 output (a bf16
         # GEMM output rounds the tables and shifts large-logit softmaxes)
-        rw = e.mhsa_rel_tables(qf.reshape(-1, d), rel_w.contiguous())
-        rh = e.mhsa_rel_tables(qf.reshape(-1, d), rel_h.contiguous())
+        rows = n * heads * l
+        rw = e.mhsa_rel_tables(qf, rel_w.contiguous(), rows, 1, 1, d, 1.0)
+        rh = e.mhsa_rel_tables(qf, rel_h.contiguous(), rows, 1, 1, d, 1.0)
         rw = rw.reshape(n * heads, l, -1)
         rh = rh.reshape(n * heads, l, -1)
         need_grad = any(t.requires_grad for t in (q, k, v, rel_h, rel_w))
         pout = (torch.empty(n * heads, l, l, dtype=q.dtype, device=q.device)
                 if need_grad else None)
-        out = e.mhsa_fwd(qf, kf, vt, rw, rh, h, w, pout)
+        out = e.mhsa_fwd(qf, kf, vt, rw, rh, h, w, pout, 1, d, d, 1.0)
         ctx.save_for_backward(q, k, v, rel_h, rel_w,
                               pout if pout is not None else q.new_empty(0))
         ctx.hw = (h, w)
@@ -118,10 +119,14 @@ class _HIPMHSARelPos(torch.autograd.Function):
         kt = k.reshape(bh, l, d).transpose(1, 2).contiguous()
         vf = v.reshape(bh, l, d).contiguous()
         dO = gout.reshape(bh, l, d).contiguous()
-        dq, dk, dv, grw, grh = ext().mhsa_bwd(
-            dO, P, qf, kt, vf, rel_w.contiguous(), rel_h.contiguous(), h, w)
-        return (dq.reshape(n, heads, l, d), dk.reshape(n, heads, l, d),
-                dv.reshape(n, heads, l, d), grh.to(rel_h.dtype),
+        # dense layout: dq/dk share one [2, bh, l, d] buffer (kqoff splits it)
+        dqk = torch.empty(2, bh, l, d, dtype=q.dtype, device=q.device)
+        dvb = torch.empty(bh, l, d, dtype=q.dtype, device=q.device)
+        dqk, dvb, grw, grh = ext().mhsa_bwd(
+            dO, P, qf, kt, vf, rel_w.contiguous(), rel_h.contiguous(), h, w,
+            1, d, bh * l * d, d, d, 1.0, dqk, dvb)
+        return (dqk[0].reshape(n, heads, l, d), dqk[1].reshape(n, heads, l, d),
+                dvb.reshape(n, heads, l, d), grh.to(rel_h.dtype),
                 grw.to(rel_w.dtype), None, None)
 
 
@@ -134,3 +139,71 @@ def mhsa_relpos(q, k, v, rel_h, rel_w, h, w):
         fallback_warn("mhsa_relpos",
                       f"dtype {q.dtype} head_dim {q.shape[-1]}")
     return _torch_mhsa(q, k, v, rel_h, rel_w, h, w)
+
+
+class _HIPMHSARelPosNHWC(torch.autograd.Function):
+    """NHWC in-place attention I/O: consumes the qkv convs' channels_last
+    outputs DIRECTLY (physical [N, L, heads*D]) via per-pixel-strided kernel
+    addressing — no chunk/transpose/contiguous copies on q, k, dO, dq, dk,
+    dv, no separate q-scale pass (the scale folds into the logits/tables),
+    and the output/grads come back as channels_last 4-D, exactly what the
+    surrounding 1x1 convs produce and consume. Only the transposed v/k
+    operands (vt, kt — MFMA B-operands must be l-contiguous) remain ATen
+    permute-copies."""
+
+    @staticmethod
+    def forward(ctx, qk, v, rel_h, rel_w, heads, dqk, dv, h, w, scale):
+        e = ext()
+        n = qk.shape[0]
+        l = h * w
+        qkc = qk.contiguous(memory_format=torch.channels_last)
+        vc = v.contiguous(memory_format=torch.channels_last)
+        qpix = 2 * heads * dqk
+        vpix = heads * dv
+        # vt [B, dv, L] from physical [N, L, heads, dv] (one copy)
+        vt = (vc.permute(0, 2, 3, 1).reshape(n, l, heads, dv)
+              .permute(0, 2, 3, 1).reshape(n * heads, dv, l).contiguous())
+        rows = n * heads * l
+        rw = e.mhsa_rel_tables(qkc, rel_w.contiguous(), rows, heads, l, qpix,
+                               scale)
+        rh = e.mhsa_rel_tables(qkc, rel_h.contiguous(), rows, heads, l, qpix,
+                               scale)
+        need_grad = any(t.requires_grad for t in (qk, v, rel_h, rel_w))
+        pout = (torch.empty(n * heads, l, l, dtype=qk.dtype, device=qk.device)
+                if need_grad else None)
+        kview = (qkc.permute(0, 2, 3, 1).reshape(n * l, qpix)
+                 [:, heads * dqk:])  # data_ptr offset only; kernel strides
+        out = e.mhsa_fwd(qkc, kview, vt, rw, rh, h, w, pout, heads, qpix,
+                         qpix, scale)
+        ctx.save_for_backward(qkc, vc, rel_h, rel_w,
+                              pout if pout is not None else qk.new_empty(0))
+        ctx.conf = (heads, dqk, dv, h, w, scale)
+        return out  # [N, heads*dv, H, W] channels_last
+
+    @staticmethod
+    def backward(ctx, gout):
+        qkc, vc, rel_h, rel_w, P = ctx.saved_tensors
+        heads, dqk, dv, h, w, scale = ctx.conf
+        n = qkc.shape[0]
+        l = h * w
+        qpix = 2 * heads * dqk
+        vpix = heads * dv
+        gc = gout.contiguous(memory_format=torch.channels_last)
+        # kt [B, dqk, L] from the k half of qk (one copy)
+        kt = (qkc.permute(0, 2, 3, 1).reshape(n, l, 2 * heads, dqk)
+              [:, :, heads:, :].permute(0, 2, 3, 1)
+              .reshape(n * heads, dqk, l).contiguous())
+        dqk_out = torch.empty_like(qkc)
+        dv_out = torch.empty_like(vc)
+        dqk_out, dv_out, grw, grh = ext().mhsa_bwd(
+            gc, P, qkc, kt, vc, rel_w.contiguous(), rel_h.contiguous(), h, w,
+            heads, qpix, heads * dqk, vpix, vpix, scale, dqk_out, dv_out)
+        return (dqk_out, dv_out, grh.to(rel_h.dtype), grw.to(rel_w.dtype),
+                None, None, None, None, None, None)
+
+
+def mhsa_relpos_nhwc(qk, v, rel_h, rel_w, heads, dqk, dv, h, w, scale):
+    """qk: [N, 2*heads*dqk, H, W] cl; v: [N, heads*dv, H, W] cl ->
+    [N, heads*dv, H, W] cl."""
+    return _HIPMHSARelPosNHWC.apply(qk, v, rel_h, rel_w, heads, dqk, dv, h,
+                                    w, scale)

@@ -469,3 +469,49 @@ def test_conv_bn_conv_bwd_stats_fusion(monkeypatch):
         assert (got2 - ref).abs().max().item() < 6e-2 * max(scale, 1.0)
         # fused and unfused paths agree tightly with each other
         assert (got1 - got2).abs().max().item() < 2e-2 * max(scale, 1.0)
+
+
+def test_mhsa_nhwc_module_path_vs_dense():
+    """The NHWC in-place attention path (qkv conv outputs consumed directly)
+    must match the dense [B, L, D] path bit-for-near on fwd and all grads."""
+    from distribuuuu_amd.models.botnet import MHSA
+    from distribuuuu_amd.ops import attention as A
+
+    _ext()
+    torch.manual_seed(0)
+    heads, d, h, w = 4, 128, 14, 14
+    m = MHSA(256, (h, w), heads=heads, dim_qk=d, dim_v=d).to("cuda").to(
+        torch.bfloat16).to(memory_format=torch.channels_last)
+    m.rel_h.data = m.rel_h.data.float().to(torch.bfloat16)
+    x = torch.randn(2, 256, h, w, device="cuda", dtype=torch.bfloat16
+                    ).contiguous(memory_format=torch.channels_last)
+    x1 = x.clone().requires_grad_(True)
+    out1 = m(x1)  # NHWC path (gate satisfied)
+    g = torch.randn_like(out1)
+    out1.backward(g)
+    grads1 = [m.to_qk.weight.grad.clone(), m.to_v.weight.grad.clone(),
+              m.rel_h.grad.clone(), m.rel_w.grad.clone(), x1.grad.clone()]
+    for p_ in (m.to_qk.weight, m.to_v.weight, m.rel_h, m.rel_w):
+        p_.grad = None
+
+    # dense path: replicate the module's pre-NHWC composition
+    x2 = x.clone().requires_grad_(True)
+    qk = m.to_qk(x2)
+    v = m.to_v(x2)
+    q, k = qk.chunk(2, dim=1)
+    n = x.shape[0]
+    q = q.reshape(n, heads, d, h * w).transpose(2, 3)
+    k = k.reshape(n, heads, d, h * w).transpose(2, 3)
+    v = v.reshape(n, heads, d, h * w).transpose(2, 3)
+    out2 = A.mhsa_relpos(q * m.scale, k, v, m.rel_h, m.rel_w, h, w)
+    out2 = out2.transpose(2, 3).reshape(n, heads * d, h, w)
+    out2.backward(g)
+    grads2 = [m.to_qk.weight.grad, m.to_v.weight.grad, m.rel_h.grad,
+              m.rel_w.grad, x2.grad]
+
+    assert torch.allclose(out1.float(), out2.float(), atol=3e-2, rtol=3e-2), \
+        (out1.float() - out2.float()).abs().max().item()
+    for g1, g2 in zip(grads1, grads2):
+        scale = g2.float().abs().max().item()
+        err = (g1.float() - g2.float()).abs().max().item()
+        assert err < 4e-2 * max(scale, 1.0), (err, scale)
