@@ -760,6 +760,11 @@ struct BnBwdEmit {
   int act;
 };
 
+at::Tensor conv2d_fwd_v2_into(at::Tensor x, at::Tensor w, at::Tensor y,
+                              int64_t Ho, int64_t Wo, int64_t groups,
+                              int64_t osh, int64_t osw, int64_t oh0,
+                              int64_t ow0);
+
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
@@ -772,6 +777,22 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                                 groups, osh, osw, oh0, ow0,
                                 bemit ? nullptr : part_out);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv2d_fwd: bf16/fp32 only");
+  {
+    // deep pad-free windows (parity-decomposed / strided dgrads) ride the
+    // v2 glds ring with the scatter epilogue instead of the v1 tile kernel
+    static const bool v2i_off = []() {
+      const char* e = getenv("DISTRIBUUUU_CONV_V2INTO");
+      return e && e[0] == '0';
+    }();
+    const int Cgi = w.size(1), Ri = w.size(2), Si = w.size(3);
+    const int Kgi = (int)(w.size(0) / groups);
+    if (!v2i_off && part_out == nullptr && bemit == nullptr && ph == 0 &&
+        pw == 0 && sh == 1 && sw == 1 && dh == 1 && dw == 1 &&
+        (Si * Cgi) % 64 == 0 && Kgi >= 96 &&
+        (int64_t)Ri * Si * Cgi >= 512 && Cgi % 8 == 0 &&
+        Ho + Ri - 1 <= x.size(2) && Wo + Si - 1 <= x.size(3))
+      return conv2d_fwd_v2_into(x, w, y, Ho, Wo, groups, osh, osw, oh0, ow0);
+  }
   check_nhwc(x, "x");
   check_nhwc(w, "w");
   const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
@@ -847,7 +868,11 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
   // R*S*Cg=2088)
   const int v2Kg = Kt / (int)groups;
   const int v2Cg = C_ / (int)groups;
-  if (!v2_off && v2Kg >= 192 && (int64_t)R * S * v2Cg >= 512 &&
+  static const int v2_mink = []() {
+    const char* e = getenv("DISTRIBUUUU_V2_MINK");
+    return e ? atoi(e) : 192;
+  }();
+  if (!v2_off && v2Kg >= v2_mink && (int64_t)R * S * v2Cg >= 512 &&
       v2Cg % 8 == 0 && x.scalar_type() == at::kBFloat16)
     return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, groups, part_out,
                           bemit);

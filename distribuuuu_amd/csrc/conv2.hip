@@ -30,6 +30,7 @@ struct Conv2Params {
   int R, SPAN64, Cg, S;
   int sh, sw, dh, dw;
   int Ho, Wo;
+  int HoA, WoA, osh, osw, oh0, ow0;  // output scatter (dgrad parity/stride)
   int M, nspan, ksteps;
   int tiles_m;
   float* part;  // EMIT: [tiles_m*4, 2*Kt] BN sum/sumsq partials
@@ -203,8 +204,10 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
     if (m < p.M) {
       const int n = m / HoWo;
       const int rem = m - n * HoWo;
-      const int64_t obase = ((int64_t)n * HoWo + rem) * p.Kt + g * p.K +
-                            tile_n * BN2 + wn * 64;
+      const int64_t obase =
+          (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh + p.oh0) * p.WoA +
+           (rem % p.Wo) * p.osw + p.ow0) * p.Kt + g * p.K +
+          tile_n * BN2 + wn * 64;
       const int k0 = tile_n * BN2 + wn * 64 + ec;
       union {
         __hip_bfloat16 b[16];
@@ -238,7 +241,9 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
       if (valid) {
         const int n = m / HoWo;
         const int rem = m - n * HoWo;
-        const int64_t obase = ((int64_t)n * HoWo + rem) * p.Kt + g * p.K;
+        const int64_t obase =
+            (((int64_t)n * p.HoA + (rem / p.Wo) * p.osh + p.oh0) * p.WoA +
+             (rem % p.Wo) * p.osw + p.ow0) * p.Kt + g * p.K;
         if (k0 + 16 <= p.K) {
           xv.q[0] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0]);
           xv.q[1] = *reinterpret_cast<const uint4*>(&p.bnx[obase + k0 + 8]);
@@ -401,6 +406,7 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   p.R = R; p.SPAN64 = SPAN64; p.Cg = C; p.S = S;
   p.sh = sh; p.sw = sw; p.dh = dh; p.dw = dw;
   p.Ho = Ho; p.Wo = Wo;
+  p.HoA = Ho; p.WoA = Wo; p.osh = 1; p.osw = 1; p.oh0 = 0; p.ow0 = 0;
   p.M = N * Ho * Wo;
   p.nspan = SPAN64 / BK2;
   p.ksteps = R * p.nspan;
@@ -437,4 +443,41 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
 at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                          int64_t ph, int64_t pw, int64_t dh, int64_t dw) {
   return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, 1, nullptr, nullptr);
+}
+
+// Scatter-output variant for the pad-free dgrad routes (parity-decomposed
+// stride-2 and strided 1x1): writes into the caller's y canvas at
+// (ho*osh + oh0, wo*osw + ow0). Host-checked: ph==pw==0, dh==dw==1 and a
+// 64-aligned reduction span (no pad/copy passes at all).
+at::Tensor conv2d_fwd_v2_into(at::Tensor x, at::Tensor w, at::Tensor y,
+                              int64_t Ho, int64_t Wo, int64_t groups,
+                              int64_t osh, int64_t osw, int64_t oh0,
+                              int64_t ow0) {
+  const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int C = Cg, K = Kt / (int)groups;
+  const int SC = S * Cg;
+  TORCH_CHECK(SC % BK2 == 0, "v2_into: span must be 64-aligned");
+  Conv2Params p;
+  p.x = (const __hip_bfloat16*)x.data_ptr();
+  p.w = (const __hip_bfloat16*)w.data_ptr();
+  p.y = (__hip_bfloat16*)y.data_ptr();
+  p.N = N; p.Hp = H; p.Wp = W; p.C = C; p.K = K;
+  p.Ct = Ct; p.Kt = y.size(1);
+  p.R = R; p.SPAN64 = SC; p.Cg = C; p.S = S;
+  p.sh = 1; p.sw = 1; p.dh = 1; p.dw = 1;
+  p.Ho = Ho; p.Wo = Wo;
+  p.HoA = y.size(2); p.WoA = y.size(3);
+  p.osh = osh; p.osw = osw; p.oh0 = oh0; p.ow0 = ow0;
+  p.M = N * Ho * Wo;
+  p.nspan = SC / BK2;
+  p.ksteps = R * p.nspan;
+  p.tiles_m = (p.M + BM2 - 1) / BM2;
+  p.part = nullptr;
+  p.bnx = nullptr; p.bnscale = nullptr; p.bnshift = nullptr; p.bnact = 0;
+  TORCH_CHECK(Ho + R - 1 <= H && Wo + S - 1 <= W, "v2_into: window OOB");
+  dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2, groups);
+  hipLaunchKernelGGL(conv_igemm_v2_kernel<0>, grid, dim3(512), 0,
+                     cur_stream(), p);
+  return y;
 }
