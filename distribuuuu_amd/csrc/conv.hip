@@ -279,10 +279,21 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
     __builtin_amdgcn_wave_barrier();
   }
   if (EMODE == 1 || EMODE == 2) {
-    const int kbase = tile_n * BN + wn * 64;
-    bn_partial_store(p.part,
-                     (int64_t)(tile_m * 2 + wm) * 2 * p.Kt + g * p.Kg + kbase,
-                     p.Kt, lane, min(64, p.Kg - kbase), ps, pq);
+    // cross-wave (wm) combine via LDS halves the partial rows
+    float* xarea = reinterpret_cast<float*>(&ldsB[0][0][0]);
+    if (wm == 1) {
+      xarea[(wn * 2 + 0) * 64 + lane] = ps;
+      xarea[(wn * 2 + 1) * 64 + lane] = pq;
+    }
+    __syncthreads();
+    if (wm == 0) {
+      ps += xarea[(wn * 2 + 0) * 64 + lane];
+      pq += xarea[(wn * 2 + 1) * 64 + lane];
+      const int kbase = tile_n * BN + wn * 64;
+      bn_partial_store(p.part,
+                       (int64_t)tile_m * 2 * p.Kt + g * p.Kg + kbase,
+                       p.Kt, lane, min(64, p.Kg - kbase), ps, pq);
+    }
   }
 }
 
@@ -372,9 +383,24 @@ __global__ __launch_bounds__(256) void conv_gemm_small_kernel(
             (int)min((int64_t)16, p.M - (mbase + mi * 16)), lane);
       __builtin_amdgcn_wave_barrier();
     }
-    if (EMIT)
-      bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
-                       p.Kt, lane, min(64, p.Kt - k0), ps, pq);
+    if (EMIT) {
+      __shared__ float xarea[3][2][64];
+      if (wid > 0) {
+        xarea[wid - 1][0][lane] = ps;
+        xarea[wid - 1][1][lane] = pq;
+      }
+      __syncthreads();
+      if (wid == 0) {
+#pragma unroll
+        for (int j = 0; j < 3; ++j) {
+          ps += xarea[j][0][lane];
+          pq += xarea[j][1][lane];
+        }
+        bn_partial_store(p.part, (int64_t)blockIdx.x * 2 * p.Kt + k0,
+                         p.Kt, lane, min(64, p.Kt - k0), ps, pq);
+      }
+      __syncthreads();  // xarea reused next k0 chunk
+    }
   }
 }
 
@@ -450,9 +476,24 @@ __global__ __launch_bounds__(256, 3) void conv_gemm_smallc_kernel(
             (int)min((int64_t)16, p.M - (mbase + mi * 16)), lane);
       __builtin_amdgcn_wave_barrier();
     }
-    if (EMIT)
-      bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 2 * p.Kt + k0,
-                       p.Kt, lane, min(64, p.Kt - k0), ps, pq);
+    if (EMIT) {
+      __shared__ float xarea[3][2][64];
+      if (wid > 0) {
+        xarea[wid - 1][0][lane] = ps;
+        xarea[wid - 1][1][lane] = pq;
+      }
+      __syncthreads();
+      if (wid == 0) {
+#pragma unroll
+        for (int j = 0; j < 3; ++j) {
+          ps += xarea[j][0][lane];
+          pq += xarea[j][1][lane];
+        }
+        bn_partial_store(p.part, (int64_t)blockIdx.x * 2 * p.Kt + k0,
+                         p.Kt, lane, min(64, p.Kt - k0), ps, pq);
+      }
+      __syncthreads();  // xarea reused next k0 chunk
+    }
   }
 }
 
@@ -611,9 +652,23 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(SmallConvParams p) {
                            lane);
     __builtin_amdgcn_wave_barrier();
   }
-  if (EMIT)
-    bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane,
-                     64, ps, pq);
+  if (EMIT) {
+    __shared__ float xarea[3][2][64];
+    if (wid > 0) {
+      xarea[wid - 1][0][lane] = ps;
+      xarea[wid - 1][1][lane] = pq;
+    }
+    __syncthreads();
+    if (wid == 0) {
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        ps += xarea[j][0][lane];
+        pq += xarea[j][1][lane];
+      }
+      bn_partial_store(p.part, (int64_t)blockIdx.x * 128, 64, lane, 64, ps,
+                       pq);
+    }
+  }
 }
 
 
@@ -732,9 +787,23 @@ __global__ __launch_bounds__(256) void conv_smallk_pipe_kernel(
                            lane);
     __builtin_amdgcn_wave_barrier();
   }
-  if (EMIT)
-    bn_partial_store(p.part, (int64_t)(blockIdx.x * 4 + wid) * 128, 64, lane,
-                     64, ps, pq);
+  if (EMIT) {
+    __shared__ float xarea[3][2][64];
+    if (wid > 0) {
+      xarea[wid - 1][0][lane] = ps;
+      xarea[wid - 1][1][lane] = pq;
+    }
+    __syncthreads();
+    if (wid == 0) {
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        ps += xarea[j][0][lane];
+        pq += xarea[j][1][lane];
+      }
+      bn_partial_store(p.part, (int64_t)blockIdx.x * 128, 64, lane, 64, ps,
+                       pq);
+    }
+  }
 }
 
 }  // namespace
@@ -822,7 +891,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
   p.bnact = 0;
   dim3 grid(p.tiles_m, (Kg + BN - 1) / BN, groups);
   if (bemit != nullptr) {
-    *part_out = at::empty({(int64_t)p.tiles_m * 2, (int64_t)2 * Kt},
+    *part_out = at::empty({(int64_t)p.tiles_m, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
     p.bnx = bemit->x;
@@ -832,7 +901,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
     hipLaunchKernelGGL(conv_igemm_fwd_kernel<2>, grid, dim3(256), 0,
                        cur_stream(), p);
   } else if (part_out != nullptr) {
-    *part_out = at::empty({(int64_t)p.tiles_m * 2, (int64_t)2 * Kt},
+    *part_out = at::empty({(int64_t)p.tiles_m, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
     hipLaunchKernelGGL(conv_igemm_fwd_kernel<1>, grid, dim3(256), 0,
@@ -895,7 +964,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     const dim3 sg((int)((M + 255) / 256));
     if (bemit != nullptr) part_out = nullptr;  // small 1x1: no bwd emission
     if (part_out != nullptr) {
-      *part_out = at::empty({(int64_t)sg.x * 4, (int64_t)2 * Kt},
+      *part_out = at::empty({(int64_t)sg.x, (int64_t)2 * Kt},
                             x.options().dtype(at::kFloat));
       sp.part = part_out->data_ptr<float>();
       if (C_ == 64)
@@ -955,7 +1024,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     const dim3 skg((int)((sp.M + 255) / 256));
     if (bemit != nullptr) part_out = nullptr;  // smallk: no bwd emission
     if (part_out != nullptr) {
-      *part_out = at::empty({(int64_t)skg.x * 4, 128},
+      *part_out = at::empty({(int64_t)skg.x, 128},
                             x.options().dtype(at::kFloat));
       sp.part = part_out->data_ptr<float>();
       if (C_ == 64)

@@ -282,10 +282,24 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
     __builtin_amdgcn_wave_barrier();
   }
   if (EMODE != 0) {
-    const int kbase = tile_n * BN2 + wn * 64;
-    bn_partial_store(p.part,
-                     (int64_t)(tile_m * 4 + wm) * 2 * p.Kt + g * p.K + kbase,
-                     p.Kt, lane, min(64, p.K - kbase), ps, pq);
+    // cross-wave (wm) combine: partial rows = tiles_m (not *4)
+    float* xarea = reinterpret_cast<float*>(smem) + 8 * (16 * 68);
+    if (wm > 0) {
+      xarea[((wn * 3 + (wm - 1)) * 2 + 0) * 64 + lane] = ps;
+      xarea[((wn * 3 + (wm - 1)) * 2 + 1) * 64 + lane] = pq;
+    }
+    __syncthreads();
+    if (wm == 0) {
+#pragma unroll
+      for (int j = 0; j < 3; ++j) {
+        ps += xarea[((wn * 3 + j) * 2 + 0) * 64 + lane];
+        pq += xarea[((wn * 3 + j) * 2 + 1) * 64 + lane];
+      }
+      const int kbase = tile_n * BN2 + wn * 64;
+      bn_partial_store(p.part,
+                       (int64_t)tile_m * 2 * p.Kt + g * p.K + kbase,
+                       p.Kt, lane, min(64, p.K - kbase), ps, pq);
+    }
   }
 }
 
@@ -418,7 +432,7 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
   p.bnact = 0;
   dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2, groups);
   if (bemit != nullptr) {
-    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * Kt},
+    *part_out = at::empty({(int64_t)p.tiles_m, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
     p.bnx = bemit->x;
@@ -428,7 +442,7 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
     hipLaunchKernelGGL(conv_igemm_v2_kernel<2>, grid, dim3(512), 0,
                        cur_stream(), p);
   } else if (part_out != nullptr) {
-    *part_out = at::empty({(int64_t)p.tiles_m * 4, (int64_t)2 * Kt},
+    *part_out = at::empty({(int64_t)p.tiles_m, (int64_t)2 * Kt},
                           x.options().dtype(at::kFloat));
     p.part = part_out->data_ptr<float>();
     hipLaunchKernelGGL(conv_igemm_v2_kernel<1>, grid, dim3(512), 0,
