@@ -917,6 +917,11 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                           int64_t ph, int64_t pw, int64_t dh, int64_t dw,
                           int64_t groups, at::Tensor* part_out,
                           const BnBwdEmit* bemit = nullptr);
+at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
+                              int64_t Cg_, int64_t R_, int64_t S_, int64_t sh,
+                              int64_t sw, int64_t ph, int64_t pw, int64_t dh,
+                              int64_t dw, int64_t groups,
+                              at::Tensor* part_out, const BnBwdEmit* bemit);
 
 static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
                                   int64_t sw, int64_t ph, int64_t pw,
@@ -1131,6 +1136,52 @@ __global__ void dilate_nhwc_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 }  // namespace
 
+namespace {
+// fused dgrad weight transform: flipT + span-pad in ONE pass. Output flat
+// [Ct, R, SPAN64] rows for the v2 ring (sp < S*Kg: w[gKg+k][R-1-r][S-1-s][c],
+// else 0) — replaces the separate flip write + pad read/write passes.
+template <typename T>
+__global__ void weight_flip_t_span_kernel(const T* __restrict__ w,
+                                          T* __restrict__ o, int Kg, int Cg,
+                                          int R, int S, int G, int SPAN64) {
+  const int64_t total = (int64_t)G * Cg * R * SPAN64;
+  const int SKg = S * Kg;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int sp = i % SPAN64;
+    int64_t t = i / SPAN64;
+    const int r = t % R;
+    t /= R;
+    const int c = t % Cg;
+    const int g = t / Cg;
+    T v = from_f32<T>(0.f);
+    if (sp < SKg) {
+      const int sidx = sp / Kg;
+      const int k = sp - sidx * Kg;
+      v = w[((((int64_t)(g * Kg + k) * R) + (R - 1 - r)) * S +
+             (S - 1 - sidx)) * Cg + c];
+    }
+    o[i] = v;
+  }
+}
+}  // namespace
+
+at::Tensor weight_flip_t_span(at::Tensor w, int64_t groups, int64_t SPAN64) {
+  CHECK_GPU(w);
+  check_nhwc(w, "w");
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Kg = Kt / groups;
+  auto o = at::empty({(int64_t)Cg * groups * R * SPAN64}, w.options());
+  const int64_t total = o.numel();
+  DISPATCH_FLOAT_AND_BF16(w.scalar_type(), "weight_flip_t_span", [&] {
+    hipLaunchKernelGGL((weight_flip_t_span_kernel<scalar_t>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)w.data_ptr(), (scalar_t*)o.data_ptr(),
+                       Kg, Cg, R, S, (int)groups, (int)SPAN64);
+  });
+  return o;
+}
+
 // weight transform for dgrad: [Kt,Cg,R,S]cl -> [Ct,Kg,R,S]cl flipped
 at::Tensor weight_flip_t(at::Tensor w, int64_t groups) {
   CHECK_GPU(w);
@@ -1228,6 +1279,31 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
   const int N = gy.size(0);
   const int Cg = w.size(1), R = w.size(2), S = w.size(3);
   const int Ct = Cg * groups;
+  {
+    // same-size dgrad on the v2 ring: ONE fused flip+span-pad weight pass
+    // (the generic path below would flip then pad separately)
+    const int Kt0 = gy.size(1);
+    const int dKg0 = Ct / (int)groups;
+    const int dCg0 = Kt0 / (int)groups;
+    static const int v2mink2 = []() {
+      const char* e = getenv("DISTRIBUUUU_V2_MINK");
+      return e ? atoi(e) : 192;
+    }();
+    static const bool v2off2 = []() {
+      const char* e = getenv("DISTRIBUUUU_CONV_V2");
+      return e && e[0] == '0';
+    }();
+    if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
+        dw * (S - 1) == 2 * pw && Kt0 % (8 * (int)groups) == 0 && !v2off2 &&
+        dKg0 >= v2mink2 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
+        gy.scalar_type() == at::kBFloat16) {
+      const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
+      auto wsp = weight_flip_t_span(w, groups, SPAN64);
+      return conv2d_fwd_v2_flat(gy, wsp, Ct, dCg0, R, S, 1, 1,
+                                dh * (R - 1) - ph, dw * (S - 1) - pw, dh, dw,
+                                groups, part_out, bemit);
+    }
+  }
   auto wt = weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
   if (R == 1 && S == 1 && (sh > 1 || sw > 1) && ph == 0 && pw == 0) {
     // strided-output GEMM: gx[ho*sh, wo*sw] = gy[ho, wo] @ w^T, rest zero.
@@ -1271,8 +1347,8 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
   const int Kt_ = gy.size(1);
   if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
       dw * (S - 1) == 2 * pw && (groups == 1 || Kt_ % (8 * groups) == 0)) {
-    // same-size conv: the plain fwd path (incl. the v2 ring kernel) applies;
-    // grouped too — wt is [Ct, Kg, R, S], a valid grouped conv weight
+    // same-size conv: the plain fwd path applies (the v2-eligible shapes
+    // were already taken above with the fused weight transform)
     return conv2d_fwd_impl(gy, wt, 1, 1, dh * (R - 1) - ph,
                            dw * (S - 1) - pw, dh, dw, groups, part_out,
                            bemit);

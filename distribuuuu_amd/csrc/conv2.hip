@@ -358,22 +358,26 @@ struct BnBwdEmit {
   int act;
 };
 
-at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
-                          int64_t ph, int64_t pw, int64_t dh, int64_t dw,
-                          int64_t groups, at::Tensor* part_out,
-                          const BnBwdEmit* bemit) {
+// Core launcher taking the weight ALREADY in span-padded flat [Kt, R,
+// SPAN64] form (either from pad_weight_span or the fused dgrad
+// flip+span transform in conv.hip).
+at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
+                              int64_t Cg_, int64_t R_, int64_t S_, int64_t sh,
+                              int64_t sw, int64_t ph, int64_t pw, int64_t dh,
+                              int64_t dw, int64_t groups,
+                              at::Tensor* part_out, const BnBwdEmit* bemit) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
   check_nhwc(x, "x");
-  check_nhwc(w, "w");
   const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
-  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Kt = Kt_, Cg = Cg_, R = R_, S = S_;
   TORCH_CHECK(Cg * groups == Ct, "v2: channel/group mismatch");
   const int C = Cg, K = Kt / groups;
   const int Ho = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   const int Wo = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
   const int SC = S * Cg;
   const int SPAN64 = (SC + BK2 - 1) / BK2 * BK2;
+  TORCH_CHECK(wp.numel() == (int64_t)Kt * R * SPAN64, "v2_flat: weight size");
 
   // padded input when ph/pw > 0 (+ slack for span-tail over-read); the
   // unpadded unaligned-span case uses x DIRECTLY with tail-address clamping
@@ -398,17 +402,6 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                        pw);
     xin = xp;
   }
-  // span-padded weight
-  at::Tensor wp = w;
-  if (SPAN64 != SC) {
-    wp = at::empty({(int64_t)Kt * R * SPAN64}, w.options());
-    int64_t total = (int64_t)Kt * R * SPAN64;
-    hipLaunchKernelGGL((pad_weight_span_kernel<__hip_bfloat16>),
-                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
-                       (const __hip_bfloat16*)w.data_ptr(),
-                       (__hip_bfloat16*)wp.data_ptr(), Kt, R, SC, SPAN64);
-  }
-
   auto y = at::empty({N, Kt, Ho, Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   Conv2Params p;
@@ -465,6 +458,27 @@ at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                          cur_stream(), p);
   }
   return y;
+}
+
+at::Tensor conv2d_fwd_v2p(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
+                          int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                          int64_t groups, at::Tensor* part_out,
+                          const BnBwdEmit* bemit) {
+  check_nhwc(w, "w");
+  const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int SC = S * Cg;
+  const int SPAN64 = (SC + BK2 - 1) / BK2 * BK2;
+  at::Tensor wp = w;
+  if (SPAN64 != SC) {
+    wp = at::empty({(int64_t)Kt * R * SPAN64}, w.options());
+    int64_t total = (int64_t)Kt * R * SPAN64;
+    hipLaunchKernelGGL((pad_weight_span_kernel<__hip_bfloat16>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const __hip_bfloat16*)w.data_ptr(),
+                       (__hip_bfloat16*)wp.data_ptr(), Kt, R, SC, SPAN64);
+  }
+  return conv2d_fwd_v2_flat(x, wp, Kt, Cg, R, S, sh, sw, ph, pw, dh, dw,
+                            groups, part_out, bemit);
 }
 
 at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
