@@ -5,6 +5,9 @@
 at::Tensor relu_fwd(at::Tensor x);
 at::Tensor relu_bwd(at::Tensor gy, at::Tensor y);
 at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b);
+at::Tensor se_scale_fwd(at::Tensor x, at::Tensor s);
+std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
+                                     at::Tensor s);
 // bn.hip
 std::vector<at::Tensor> bn_sums(at::Tensor x);
 at::Tensor bn_reduce_partials(at::Tensor part);
@@ -107,6 +110,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("se_scale_fwd", &se_scale_fwd);
+  m.def("se_scale_bwd", &se_scale_bwd);
   m.def("bn_sums", &bn_sums);
   m.def("bn_reduce_partials", &bn_reduce_partials);
   m.def("bn_stats", &bn_stats, py::arg("x"), py::arg("gamma"), py::arg("beta"),

@@ -107,3 +107,157 @@ at::Tensor relu_bwd(at::Tensor gy, at::Tensor y) {
   });
   return gx;
 }
+
+// ---------------------------------------------------------------------------
+// SE channel scaling (K5): y[n, :, c] = x * s[n, c], and the fused backward
+// gx = gy * s, gs[n, c] = sum_hw(gy * x) in ONE pass (ATen needed two
+// broadcast muls plus a separate reduction). One block per image n: the
+// per-(n, c) scales hoist once and the gs reduction never crosses blocks.
+// ---------------------------------------------------------------------------
+namespace {
+
+DEV_INLINE int p2f_se(int v) { return 1 << (31 - __builtin_clz(v)); }
+
+template <typename T>
+__global__ __launch_bounds__(256) void se_scale_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ s, T* __restrict__ y,
+    int HW, int C) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  const int n = blockIdx.x;
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2f_se(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  if (rl >= nrl) return;
+  const P* xp = reinterpret_cast<const P*>(x) + (int64_t)n * HW * cpacks;
+  P* yp = reinterpret_cast<P*>(y) + (int64_t)n * HW * cpacks;
+  const P* sp = reinterpret_cast<const P*>(s) + (int64_t)n * cpacks;
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sv[V];
+    const P sq = sp[cp];
+#pragma unroll
+    for (int j = 0; j < V; ++j) sv[j] = to_f32(sq.v[j]);
+    const int64_t rstep = (int64_t)nrl * cpacks;
+    const P* xq = xp + (int64_t)rl * cpacks + cp;
+    P* yq = yp + (int64_t)rl * cpacks + cp;
+    for (int row = rl; row < HW; row += nrl) {
+      P px = xq[0];
+#pragma unroll
+      for (int j = 0; j < V; ++j)
+        px.v[j] = from_f32<T>(to_f32(px.v[j]) * sv[j]);
+      yq[0] = px;
+      xq += rstep;
+      yq += rstep;
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void se_scale_bwd_kernel(
+    const T* __restrict__ gy, const T* __restrict__ x,
+    const T* __restrict__ s, T* __restrict__ gx, T* __restrict__ gs, int HW,
+    int C) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  __shared__ float red[256 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
+  const int n = blockIdx.x;
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2f_se(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const P* gp = reinterpret_cast<const P*>(gy) + (int64_t)n * HW * cpacks;
+  const P* xp = reinterpret_cast<const P*>(x) + (int64_t)n * HW * cpacks;
+  P* oxp = reinterpret_cast<P*>(gx) + (int64_t)n * HW * cpacks;
+  const P* sp = reinterpret_cast<const P*>(s) + (int64_t)n * cpacks;
+  P* gsp = reinterpret_cast<P*>(gs) + (int64_t)n * cpacks;
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float acc[V] = {};
+    if (active) {
+      float sv[V];
+      const P sq = sp[cp];
+#pragma unroll
+      for (int j = 0; j < V; ++j) sv[j] = to_f32(sq.v[j]);
+      const int64_t rstep = (int64_t)nrl * cpacks;
+      const P* gq = gp + (int64_t)rl * cpacks + cp;
+      const P* xq = xp + (int64_t)rl * cpacks + cp;
+      P* oq = oxp + (int64_t)rl * cpacks + cp;
+      for (int row = rl; row < HW; row += nrl) {
+        P pg = gq[0], px = xq[0], ox;
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float g = to_f32(pg.v[j]);
+          acc[j] += g * to_f32(px.v[j]);
+          ox.v[j] = from_f32<T>(g * sv[j]);
+        }
+        oq[0] = ox;
+        gq += rstep;
+        xq += rstep;
+        oq += rstep;
+      }
+    }
+    float* slot = &red[(rl * ncp + cp0) * V];
+    if (active)
+#pragma unroll
+      for (int j = 0; j < V; ++j) slot[j] = acc[j];
+    __syncthreads();
+    for (int st = nrl >> 1; st > 0; st >>= 1) {
+      if (active && rl < st) {
+        const float* other = &red[((rl + st) * ncp + cp0) * V];
+#pragma unroll
+        for (int j = 0; j < V; ++j) slot[j] += other[j];
+      }
+      __syncthreads();
+    }
+    if (active && rl == 0) {
+      P o;
+#pragma unroll
+      for (int j = 0; j < V; ++j) o.v[j] = from_f32<T>(slot[j]);
+      gsp[cp] = o;
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+at::Tensor se_scale_fwd(at::Tensor x, at::Tensor s) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  auto y = at::empty_like(x);
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "se_scale_fwd", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "se_scale: C % ", V);
+    hipLaunchKernelGGL((se_scale_fwd_kernel<scalar_t>), dim3(N), dim3(256), 0,
+                       cur_stream(), (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)s.data_ptr(),
+                       (scalar_t*)y.data_ptr(), HW, C);
+  });
+  return y;
+}
+
+std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
+                                     at::Tensor s) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int N = x.size(0), C = x.size(1);
+  const int HW = x.size(2) * x.size(3);
+  auto gx = at::empty_like(x);
+  auto gs = at::empty({N, C, 1, 1}, x.options());
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "se_scale_bwd", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "se_scale: C % ", V);
+    hipLaunchKernelGGL((se_scale_bwd_kernel<scalar_t>), dim3(N), dim3(256), 0,
+                       cur_stream(), (const scalar_t*)gy.data_ptr(),
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)s.data_ptr(),
+                       (scalar_t*)gx.data_ptr(), (scalar_t*)gs.data_ptr(),
+                       HW, C);
+  });
+  return {gx, gs};
+}

@@ -13,6 +13,8 @@ import math
 
 import torch.nn as nn
 
+from ..ops import functional as DF
+
 from ..ops import AdaptiveAvgPool2d, BatchNorm2d, Conv2d, Dropout, Linear, SiLU, Sigmoid
 
 
@@ -28,7 +30,7 @@ class SqueezeExcite(nn.Module):
     def forward(self, x):
         s = self.avg_pool(x)
         s = self.gate(self.conv_expand(self.act(self.conv_reduce(s))))
-        return x * s
+        return DF.se_scale(x, s)
 
 
 class MBConv(nn.Module):

@@ -12,6 +12,7 @@ The 3x3 grouped conv is the flagship group-conv MFMA path of BASELINE.json confi
 import numpy as np
 import torch.nn as nn
 
+from ..ops import functional as DF  # noqa: E501
 from ..ops import AdaptiveAvgPool2d, BatchNorm2d, Conv2d, Linear, ReLU, Sigmoid
 
 
@@ -46,7 +47,7 @@ class SE(nn.Module):
 
     def forward(self, x):
         s = self.gate(self.fc2(self.act(self.fc1(self.avg_pool(x)))))
-        return x * s
+        return DF.se_scale(x, s)
 
 
 class RegBlock(nn.Module):

@@ -386,6 +386,34 @@ def add_relu(a, b):
     return F.relu(a + b)
 
 
+class _HIPSEScale(torch.autograd.Function):
+    """SE gate application y = x * s[n, c] with fused backward
+    (gx = gy * s and gs = sum_hw(gy * x) in one kernel pass — ATen needed
+    two broadcast muls plus a reduction)."""
+
+    @staticmethod
+    def forward(ctx, x, s):
+        x = _cl(x)
+        s = _cl(s)
+        y = ext().se_scale_fwd(x, s)
+        ctx.save_for_backward(x, s)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, s = ctx.saved_tensors
+        gx, gs = ext().se_scale_bwd(_cl(gy), x, s)
+        return gx, gs
+
+
+def se_scale(x, s):
+    """x [N, C, H, W] * s [N, C, 1, 1] (the SE excite gate)."""
+    if (use_hip(x, "se_scale_fwd") and x.shape[1] % 8 == 0
+            and x.dtype == s.dtype):
+        return _HIPSEScale.apply(x, s)
+    return x * s
+
+
 class _HIPReLU(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):

@@ -515,3 +515,28 @@ def test_mhsa_nhwc_module_path_vs_dense():
         scale = g2.float().abs().max().item()
         err = (g1.float() - g2.float()).abs().max().item()
         assert err < 4e-2 * max(scale, 1.0), (err, scale)
+
+
+@pytest.mark.parametrize("shape", [(4, 232, 28, 28), (2, 1392, 14, 14)])
+def test_se_scale_fwd_bwd(shape):
+    """Fused SE gate (y = x*s[n,c]; bwd gx = gy*s, gs = sum_hw gy*x) vs ATen."""
+    from distribuuuu_amd.ops import functional as DF
+
+    _ext()
+    torch.manual_seed(0)
+    n, c, h, w = shape
+    x = _cl(torch.randn(n, c, h, w, device="cuda",
+                        dtype=torch.bfloat16)).requires_grad_(True)
+    s = torch.rand(n, c, 1, 1, device="cuda",
+                   dtype=torch.bfloat16).requires_grad_(True)
+    y = DF.se_scale(x, s)
+    gy = _cl(torch.randn_like(y))
+    y.backward(gy)
+    xr = x.detach().float().requires_grad_(True)
+    sr = s.detach().float().requires_grad_(True)
+    (xr * sr).backward(gy.float())
+    assert torch.allclose(y.float(), (x.detach().float() * s.detach().float()),
+                          atol=2e-2, rtol=2e-2)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=2e-2, rtol=2e-2)
+    scale = sr.grad.abs().max().item()
+    assert (s.grad.float() - sr.grad).abs().max().item() < 3e-2 * max(scale, 1)
