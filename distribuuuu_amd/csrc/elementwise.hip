@@ -121,10 +121,12 @@ DEV_INLINE int p2f_se(int v) { return 1 << (31 - __builtin_clz(v)); }
 template <typename T>
 __global__ __launch_bounds__(256) void se_scale_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ s, T* __restrict__ y,
-    int HW, int C) {
+    int HW, int C, int rows_per_chunk) {
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
-  const int n = blockIdx.x;
+  const int n = blockIdx.y;
+  const int row_lo = blockIdx.x * rows_per_chunk;
+  const int row_hi = min(row_lo + rows_per_chunk, HW);
   const int cpacks = C / V;
   const int ncp = min(cpacks, (int)blockDim.x);
   const int nrl = p2f_se(blockDim.x / ncp);
@@ -140,9 +142,9 @@ __global__ __launch_bounds__(256) void se_scale_fwd_kernel(
 #pragma unroll
     for (int j = 0; j < V; ++j) sv[j] = to_f32(sq.v[j]);
     const int64_t rstep = (int64_t)nrl * cpacks;
-    const P* xq = xp + (int64_t)rl * cpacks + cp;
-    P* yq = yp + (int64_t)rl * cpacks + cp;
-    for (int row = rl; row < HW; row += nrl) {
+    const P* xq = xp + (int64_t)(row_lo + rl) * cpacks + cp;
+    P* yq = yp + (int64_t)(row_lo + rl) * cpacks + cp;
+    for (int row = row_lo + rl; row < row_hi; row += nrl) {
       P px = xq[0];
 #pragma unroll
       for (int j = 0; j < V; ++j)
@@ -154,15 +156,19 @@ __global__ __launch_bounds__(256) void se_scale_fwd_kernel(
   }
 }
 
+// gs accumulates with fp32 atomics across row-chunk blocks (same
+// determinism tradeoff as the split-m wgrad; ~1e-3 relative run-to-run)
 template <typename T>
 __global__ __launch_bounds__(256) void se_scale_bwd_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
-    const T* __restrict__ s, T* __restrict__ gx, T* __restrict__ gs, int HW,
-    int C) {
+    const T* __restrict__ s, T* __restrict__ gx, float* __restrict__ gs,
+    int HW, int C, int rows_per_chunk) {
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   __shared__ float red[256 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
-  const int n = blockIdx.x;
+  const int n = blockIdx.y;
+  const int row_lo = blockIdx.x * rows_per_chunk;
+  const int row_hi = min(row_lo + rows_per_chunk, HW);
   const int cpacks = C / V;
   const int ncp = min(cpacks, (int)blockDim.x);
   const int nrl = p2f_se(blockDim.x / ncp);
@@ -173,7 +179,7 @@ __global__ __launch_bounds__(256) void se_scale_bwd_kernel(
   const P* xp = reinterpret_cast<const P*>(x) + (int64_t)n * HW * cpacks;
   P* oxp = reinterpret_cast<P*>(gx) + (int64_t)n * HW * cpacks;
   const P* sp = reinterpret_cast<const P*>(s) + (int64_t)n * cpacks;
-  P* gsp = reinterpret_cast<P*>(gs) + (int64_t)n * cpacks;
+  float* gsp = gs + (int64_t)n * C;
   for (int cp = cp0; cp < cpacks; cp += ncp) {
     float acc[V] = {};
     if (active) {
@@ -182,10 +188,10 @@ __global__ __launch_bounds__(256) void se_scale_bwd_kernel(
 #pragma unroll
       for (int j = 0; j < V; ++j) sv[j] = to_f32(sq.v[j]);
       const int64_t rstep = (int64_t)nrl * cpacks;
-      const P* gq = gp + (int64_t)rl * cpacks + cp;
-      const P* xq = xp + (int64_t)rl * cpacks + cp;
-      P* oq = oxp + (int64_t)rl * cpacks + cp;
-      for (int row = rl; row < HW; row += nrl) {
+      const P* gq = gp + (int64_t)(row_lo + rl) * cpacks + cp;
+      const P* xq = xp + (int64_t)(row_lo + rl) * cpacks + cp;
+      P* oq = oxp + (int64_t)(row_lo + rl) * cpacks + cp;
+      for (int row = row_lo + rl; row < row_hi; row += nrl) {
         P pg = gq[0], px = xq[0], ox;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
@@ -213,10 +219,9 @@ __global__ __launch_bounds__(256) void se_scale_bwd_kernel(
       __syncthreads();
     }
     if (active && rl == 0) {
-      P o;
 #pragma unroll
-      for (int j = 0; j < V; ++j) o.v[j] = from_f32<T>(slot[j]);
-      gsp[cp] = o;
+      for (int j = 0; j < V; ++j)
+        atomicAdd(&gsp[cp * V + j], slot[j]);
     }
     __syncthreads();
   }
@@ -230,13 +235,16 @@ at::Tensor se_scale_fwd(at::Tensor x, at::Tensor s) {
   const int N = x.size(0), C = x.size(1);
   const int HW = x.size(2) * x.size(3);
   auto y = at::empty_like(x);
+  const int chunks = std::max<int>(1, (int)ceil_div(768, N));
+  const int rpc = (int)ceil_div(HW, chunks);
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "se_scale_fwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % V == 0, "se_scale: C % ", V);
-    hipLaunchKernelGGL((se_scale_fwd_kernel<scalar_t>), dim3(N), dim3(256), 0,
+    hipLaunchKernelGGL((se_scale_fwd_kernel<scalar_t>),
+                       dim3((int)ceil_div(HW, rpc), N), dim3(256), 0,
                        cur_stream(), (const scalar_t*)x.data_ptr(),
                        (const scalar_t*)s.data_ptr(),
-                       (scalar_t*)y.data_ptr(), HW, C);
+                       (scalar_t*)y.data_ptr(), HW, C, rpc);
   });
   return y;
 }
@@ -248,16 +256,19 @@ std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
   const int N = x.size(0), C = x.size(1);
   const int HW = x.size(2) * x.size(3);
   auto gx = at::empty_like(x);
-  auto gs = at::empty({N, C, 1, 1}, x.options());
+  auto gsf = at::zeros({N, C}, x.options().dtype(at::kFloat));
+  const int chunks = std::max<int>(1, (int)ceil_div(768, N));
+  const int rpc = (int)ceil_div(HW, chunks);
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "se_scale_bwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % V == 0, "se_scale: C % ", V);
-    hipLaunchKernelGGL((se_scale_bwd_kernel<scalar_t>), dim3(N), dim3(256), 0,
+    hipLaunchKernelGGL((se_scale_bwd_kernel<scalar_t>),
+                       dim3((int)ceil_div(HW, rpc), N), dim3(256), 0,
                        cur_stream(), (const scalar_t*)gy.data_ptr(),
                        (const scalar_t*)x.data_ptr(),
                        (const scalar_t*)s.data_ptr(),
-                       (scalar_t*)gx.data_ptr(), (scalar_t*)gs.data_ptr(),
-                       HW, C);
+                       (scalar_t*)gx.data_ptr(), gsf.data_ptr<float>(),
+                       HW, C, rpc);
   });
-  return {gx, gs};
+  return {gx, gsf.to(x.scalar_type()).reshape({N, C, 1, 1})};
 }
