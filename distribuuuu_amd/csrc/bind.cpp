@@ -6,6 +6,7 @@ at::Tensor relu_fwd(at::Tensor x);
 at::Tensor relu_bwd(at::Tensor gy, at::Tensor y);
 at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b);
 at::Tensor se_scale_fwd(at::Tensor x, at::Tensor s);
+at::Tensor dropout_fwd(at::Tensor x, double p, int64_t seed);
 std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
                                      at::Tensor s);
 // bn.hip
@@ -111,6 +112,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd);
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("se_scale_fwd", &se_scale_fwd);
+  m.def("dropout_fwd", &dropout_fwd);
   m.def("se_scale_bwd", &se_scale_bwd);
   m.def("bn_sums", &bn_sums);
   m.def("bn_reduce_partials", &bn_reduce_partials);

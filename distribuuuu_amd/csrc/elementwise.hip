@@ -272,3 +272,45 @@ std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
   });
   return {gx, gsf.to(x.scalar_type()).reshape({N, C, 1, 1})};
 }
+
+// ---------------------------------------------------------------------------
+// Dropout (K19): counter-based RNG (SplitMix64 on (seed, index)) so the mask
+// is a pure function of (seed, position) — no state, replay-stable under
+// hipGraph, and backward regenerates the mask instead of storing it.
+// ---------------------------------------------------------------------------
+namespace {
+
+DEV_INLINE float u01_from(uint64_t z) {
+  z += 0x9e3779b97f4a7c15ULL;
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ULL;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebULL;
+  z = z ^ (z >> 31);
+  return (float)(z >> 40) * (1.0f / 16777216.0f);  // 24-bit mantissa
+}
+
+template <typename T, bool BWD>
+__global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ y,
+                               int64_t total, float p, float inv_keep,
+                               uint64_t seed) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const bool keep = u01_from(seed ^ (uint64_t)i) >= p;
+    y[i] = keep ? from_f32<T>(to_f32(x[i]) * inv_keep) : from_f32<T>(0.f);
+  }
+}
+
+}  // namespace
+
+at::Tensor dropout_fwd(at::Tensor x, double p, int64_t seed) {
+  CHECK_GPU(x);
+  auto y = at::empty_like(x);
+  const int64_t total = x.numel();
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "dropout_fwd", [&] {
+    hipLaunchKernelGGL((dropout_kernel<scalar_t, false>),
+                       dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),
+                       (const scalar_t*)x.data_ptr(),
+                       (scalar_t*)y.data_ptr(), total, (float)p,
+                       1.f / (1.f - (float)p), (uint64_t)seed);
+  });
+  return y;
+}

