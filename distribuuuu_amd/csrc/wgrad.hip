@@ -264,6 +264,198 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(WgradParams p) {
 }
 
 // ---------------------------------------------------------------------------
+// 128x128 TR-STAGED wgrad: the 64x64 kernel issues 16 tr-read pairs for 8
+// MFMAs per k-step per wave (LDS-read bound at ~350 TF); this tile gives
+// each wave a 64x64 output (16 tr-read pairs for 32 MFMAs) while KEEPING
+// the full per-thread boundary predication the glds ring lacks — so dense
+// 3x3 shapes qualify with no pad pass. Staging uses the same
+// [32 m][16 col] 1056-B subtiles and b128 writes.
+// ---------------------------------------------------------------------------
+constexpr int T128_BM = 128;            // k rows per tile
+constexpr int T128_BN = 128;            // rsc cols per tile
+constexpr int T128_TILE = 16 * SUBT;    // one operand tile (16.5 KB)
+
+struct WgradT128Params {
+  const __hip_bfloat16* x;   // [N,H,W,Ct]
+  const __hip_bfloat16* gy;  // [N,Ho,Wo,Kt]
+  float* acc;                // [Kt, R*S*Cg] zeroed
+  int N, H, W, Ct, Kt;
+  int R, S, Cg, Kg;
+  int sh, sw, ph, pw, dh, dw;
+  int Ho, Wo;
+  int M, RSC;
+  int ktiles, ntiles, chunks, csteps;
+  unsigned long long magicHoWo, magicWo;
+};
+
+__global__ __launch_bounds__(256) void conv_wgrad_tr128_kernel(
+    WgradT128Params p) {
+  const int g = blockIdx.z;
+  const int ktile = blockIdx.x % p.ktiles;
+  const int ntile = blockIdx.x / p.ktiles;
+  const int chunk = blockIdx.y;
+
+  __shared__ __align__(16) char smem[2 * 2 * T128_TILE];
+  auto ldsA = [&](int buf) -> char* { return smem + buf * 2 * T128_TILE; };
+  auto ldsB = [&](int buf) -> char* {
+    return smem + buf * 2 * T128_TILE + T128_TILE;
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int il = lane & 15, kq = lane >> 4;
+
+  // staging: slot s = tid + it*256 (it<4): m = s>>4 (0..63), col8 = (s&15)*8
+  // subtile layout [m/32][col/16] of [32][16], stride SUBT
+  int s_m[4], s_col[4], s_off[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int sl = tid + it * 256;
+    s_m[it] = sl >> 4;
+    s_col[it] = (sl & 15) << 3;
+    s_off[it] = ((s_m[it] >> 5) * 8 + (s_col[it] >> 4)) * SUBT +
+                (s_m[it] & 31) * 32 + (s_col[it] & 15) * 2;
+  }
+  const int SCg = p.S * p.Cg;
+  // B (x) rsc decomposition per slot (fixed)
+  int b_r[4], b_s[4], b_c[4];
+  bool b_ok[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int rsc = ntile * T128_BN + s_col[it];
+    b_ok[it] = rsc < p.RSC;
+    const int rr = b_ok[it] ? rsc / SCg : 0;
+    const int rem = (b_ok[it] ? rsc : 0) - rr * SCg;
+    b_r[it] = rr;
+    b_s[it] = rem / p.Cg;
+    b_c[it] = rem - b_s[it] * p.Cg;
+  }
+  int a_k[4];
+  bool a_ok[4];
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int k = ktile * T128_BM + s_col[it];
+    a_ok[it] = k < p.Kg;
+    a_k[it] = g * p.Kg + (a_ok[it] ? k : 0);
+  }
+
+  const int m0 = chunk * (p.csteps * WBK);
+  const int HoWo = p.Ho * p.Wo;
+  const int ksteps = min(p.csteps, (int)((p.M - m0 + WBK - 1) / WBK));
+
+  uint4 regA[4], regB[4];
+  auto stage_load = [&](int ks) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int m = m0 + ks * WBK + s_m[it];
+      const bool m_ok = m < p.M;
+      const int mm = m_ok ? m : 0;
+      const int n = magic_div(mm, p.magicHoWo);
+      const int rem = mm - n * HoWo;
+      const int ho = magic_div(rem, p.magicWo);
+      const int wo = rem - ho * p.Wo;
+      regA[it] = (m_ok && a_ok[it])
+                     ? *reinterpret_cast<const uint4*>(
+                           p.gy + ((int64_t)mm * p.Kt) + a_k[it])
+                     : uint4{0, 0, 0, 0};
+      const int h = ho * p.sh - p.ph + b_r[it] * p.dh;
+      const int w_ = wo * p.sw - p.pw + b_s[it] * p.dw;
+      regB[it] =
+          (m_ok && b_ok[it] && h >= 0 && h < p.H && w_ >= 0 && w_ < p.W)
+              ? *reinterpret_cast<const uint4*>(
+                    p.x + (((int64_t)n * p.H + h) * p.W + w_) * p.Ct +
+                    g * p.Cg + b_c[it])
+              : uint4{0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&](int buf) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      *reinterpret_cast<uint4*>(ldsA(buf) + s_off[it]) = regA[it];
+      *reinterpret_cast<uint4*>(ldsB(buf) + s_off[it]) = regB[it];
+    }
+  };
+
+  f32x4w accv[4][4] = {};
+  stage_load(0);
+  stage_write(0);
+  __syncthreads();
+  if (ksteps > 1) stage_load(1);
+
+  int cur = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    const unsigned abase =
+        (unsigned)(unsigned long long)(ldsA(cur));
+    const unsigned bbase =
+        (unsigned)(unsigned long long)(ldsB(cur));
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {
+      TrFrag fa[4], fb[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int sb = (mc * 8 + wm * 4 + mi) * SUBT;
+        tr_read_issue(abase + tr_addr(sb, lane, 0),
+                      abase + tr_addr(sb, lane, 1), fa[mi]);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int sb = (mc * 8 + wn * 4 + ni) * SUBT;
+        tr_read_issue(bbase + tr_addr(sb, lane, 0),
+                      bbase + tr_addr(sb, lane, 1), fb[ni]);
+      }
+      asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        if (ni == 1) {
+          asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+        if (ni == 2) {
+          asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+        if (ni == 3) {
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+          accv[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[mi].v, fb[ni].v, accv[mi][ni], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      stage_write(cur ^ 1);
+      if (ks + 2 < ksteps) stage_load(ks + 2);
+      __syncthreads();
+    }
+    cur ^= 1;
+  }
+
+  // epilogue: fp32 atomic accumulate (D: col=lane&15, row=(lane>>4)*4+rr)
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int k = ktile * T128_BM + wm * 64 + mi * 16 + kq * 4 + rr;
+      if (k >= p.Kg) continue;
+      const int64_t rowbase = (int64_t)(g * p.Kg + k) * p.RSC;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int col = ntile * T128_BN + wn * 64 + ni * 16 + il;
+        if (col < p.RSC) atomicAdd(&p.acc[rowbase + col], accv[mi][ni][rr]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Ring-staged wgrad (the conv-v2 structure applied to wgrad): packed 1 KiB
 // [32 m][16 col] subtiles filled by global_load_lds (lane-linear), tr_b16
 // fragment reads, 3-slot LDS ring with counted vmcnt across raw barriers.
@@ -811,6 +1003,58 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                        0, cur_stream(), q.acc, (__hip_bfloat16*)gw.data_ptr(),
                        total);
     return gw;
+  }
+
+  // 128x128 tr-staged tile (A/B via DISTRIBUUUU_WGRAD_T128: 1 force,
+  // 0 off, default auto-gate from the probe measurements)
+  {
+    const char* e = getenv("DISTRIBUUUU_WGRAD_T128");
+    const int t128 = e ? atoi(e) : -1;
+    const int64_t t128_blocks =
+        (int64_t)((Kg + 127) / 128) * (((int64_t)R * S * Cg + 127) / 128) *
+        ((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK)) * groups;
+    // measured (tools/probes/wgrad_dense_ab.py): wins/ties every dense 3x3
+    // with Kg >= 128 (512-deep: -13%), loses at Kg = 64 (half-empty tiles)
+    const bool t128_gate = Kg >= 128 && (int64_t)R * S * Cg >= 128;
+    if (t128 == 1 || (t128 == -1 && t128_gate)) {
+      WgradT128Params q;
+      q.x = (const __hip_bfloat16*)x.data_ptr();
+      q.gy = (const __hip_bfloat16*)gy.data_ptr();
+      q.N = N; q.H = H; q.W = W; q.Ct = Ct; q.Kt = Kt;
+      q.R = R; q.S = S; q.Cg = Cg; q.Kg = Kg;
+      q.sh = sh; q.sw = sw; q.ph = ph; q.pw = pw; q.dh = dh; q.dw = dw;
+      q.Ho = Ho; q.Wo = Wo;
+      q.M = (int)M64;
+      q.RSC = R * S * Cg;
+      q.ktiles = (Kg + T128_BM - 1) / T128_BM;
+      q.ntiles = (q.RSC + T128_BN - 1) / T128_BN;
+      q.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
+      q.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
+      int csteps = CHUNK_STEPS;
+      int chunks = (int)((M64 + (int64_t)csteps * WBK - 1) / (csteps * WBK));
+      while (csteps > 8 &&
+             (int64_t)q.ktiles * q.ntiles * chunks * groups < 512) {
+        csteps /= 2;
+        chunks = (int)((M64 + (int64_t)csteps * WBK - 1) / (csteps * WBK));
+      }
+      q.csteps = csteps;
+      q.chunks = chunks;
+      auto accb = at::zeros({(int64_t)Kt, q.RSC},
+                            x.options().dtype(at::kFloat));
+      q.acc = accb.data_ptr<float>();
+      dim3 grid(q.ktiles * q.ntiles, chunks, groups);
+      hipLaunchKernelGGL(conv_wgrad_tr128_kernel, grid, dim3(256), 0,
+                         cur_stream(), q);
+      auto gw = at::empty({Kt, Cg, (int64_t)R, (int64_t)S},
+                          x.options().memory_format(
+                              at::MemoryFormat::ChannelsLast));
+      const int64_t tot = (int64_t)Kt * q.RSC;
+      hipLaunchKernelGGL(cast_acc_kernel, dim3(grid_1d(tot, 256)), dim3(256),
+                         0, cur_stream(), q.acc,
+                         (__hip_bfloat16*)gw.data_ptr(), tot);
+      return gw;
+    }
+    (void)t128_blocks;
   }
 
   WgradParams p;

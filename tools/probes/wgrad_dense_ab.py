@@ -62,3 +62,30 @@ for cnt, c, h, w, k, r, s in SHAPES:
           f"ring128 {res['1']:8.1f} us   xerr {err:.2e}/{scl:.1e}")
 print(f"step totals: 64x64 {tot['0']:.2f} ms   ring128 {tot['1']:.2f} ms")
 os.environ.pop("DISTRIBUUUU_WGRAD_128", None)
+
+
+# --- T128 tr-staged variant A/B (env DISTRIBUUUU_WGRAD_T128) ---
+print("\n== 64x64 vs tr-staged 128x128 ==")
+tot = {"0": 0.0, "1": 0.0}
+for cnt, c, h, w, k, r, s in SHAPES:
+    ho = (h + 2 - r) // s + 1
+    x = torch.randn(N, c, h, w, device="cuda",
+                    dtype=torch.bfloat16).contiguous(memory_format=cl)
+    gy = torch.randn(N, k, ho, ho, device="cuda",
+                     dtype=torch.bfloat16).contiguous(memory_format=cl)
+    res = {}
+    for mode in ("0", "1"):
+        os.environ["DISTRIBUUUU_WGRAD_T128"] = mode
+        t = bench(lambda: e.conv2d_wgrad(gy, x, r, r, s, s, 1, 1, 1, 1, 1))
+        res[mode] = t * 1e6
+        tot[mode] += cnt * t * 1e3
+    os.environ["DISTRIBUUUU_WGRAD_T128"] = "0"
+    gw0 = e.conv2d_wgrad(gy, x, r, r, s, s, 1, 1, 1, 1, 1).float()
+    os.environ["DISTRIBUUUU_WGRAD_T128"] = "1"
+    gw1 = e.conv2d_wgrad(gy, x, r, r, s, s, 1, 1, 1, 1, 1).float()
+    err = (gw0 - gw1).abs().max().item()
+    scl = gw0.abs().max().item()
+    print(f"{c:4d}x{h:2d} s{s} x{cnt}:  64x64 {res['0']:8.1f} us   "
+          f"t128 {res['1']:8.1f} us   xerr {err:.2e}/{scl:.1e}")
+print(f"step totals: 64x64 {tot['0']:.2f} ms   t128 {tot['1']:.2f} ms")
+os.environ.pop("DISTRIBUUUU_WGRAD_T128", None)
