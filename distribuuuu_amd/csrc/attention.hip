@@ -260,8 +260,12 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
   const int q0 = blockIdx.y * QT;
   extern __shared__ __align__(16) char smem[];
   __hip_bfloat16* DS = reinterpret_cast<__hip_bfloat16*>(smem);
-  float* RWs = reinterpret_cast<float*>(smem + QT * p.lpad * 2);  // [QT][32]
-  float* RHs = RWs + QT * 32;                                     // [QT][32]
+  // rel scratch in bf16: the LDS copy only feeds the dq fold (the fp32
+  // copies for the rel-grad kernel go to global); 8 KB saved puts the
+  // block at 4 waves-groups/CU instead of 3
+  __hip_bfloat16* RWs =
+      reinterpret_cast<__hip_bfloat16*>(smem + QT * p.lpad * 2);  // [QT][32]
+  __hip_bfloat16* RHs = RWs + QT * 32;                            // [QT][32]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -363,7 +367,7 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
       float s = 0.f;
       if (qr < p.L && wj >= 0 && wj < p.W)
         for (int hj = 0; hj < p.H; ++hj) s += to_f32(dsrow[hj * p.W + wj]);
-      RWs[(wid * 16 + r) * 32 + lane] = s;
+      RWs[(wid * 16 + r) * 32 + lane] = from_f32<__hip_bfloat16>(s);
       if (qr < p.L) p.drw[((int64_t)b * p.L + qr) * (2 * p.W - 1) + lane] = s;
     }
     if (lane < 2 * p.H - 1) {
@@ -371,7 +375,7 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
       float s = 0.f;
       if (qr < p.L && hj >= 0 && hj < p.H)
         for (int wj = 0; wj < p.W; ++wj) s += to_f32(dsrow[hj * p.W + wj]);
-      RHs[(wid * 16 + r) * 32 + lane] = s;
+      RHs[(wid * 16 + r) * 32 + lane] = from_f32<__hip_bfloat16>(s);
       if (qr < p.L) p.drh[((int64_t)b * p.L + qr) * (2 * p.H - 1) + lane] = s;
     }
   }
@@ -407,12 +411,12 @@ __global__ __launch_bounds__(256) void mhsa_bwd_q_kernel(MhsaBwdQParams p) {
       if (qr < p.L) {
         const int d = dt * 16 + il;
         float out = acc[rr];
-        const float* rws = &RWs[(wid * 16 + rloc) * 32];
-        const float* rhs = &RHs[(wid * 16 + rloc) * 32];
+        const __hip_bfloat16* rws = &RWs[(wid * 16 + rloc) * 32];
+        const __hip_bfloat16* rhs = &RHs[(wid * 16 + rloc) * 32];
         for (int m = 0; m < 2 * p.W - 1; ++m)
-          out += rws[m] * to_f32(p.rw[m * p.D + d]);
+          out += to_f32(rws[m]) * to_f32(p.rw[m * p.D + d]);
         for (int m = 0; m < 2 * p.H - 1; ++m)
-          out += rhs[m] * to_f32(p.rh[m * p.D + d]);
+          out += to_f32(rhs[m]) * to_f32(p.rh[m * p.D + d]);
         p.dq[strided_row(b, qr, p.L, p.heads, p.qpix, p.D) + d] =
             from_f32<__hip_bfloat16>(out * p.scale);
       }
@@ -676,7 +680,7 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
   p.ltiles16 = (L + 15) / 16;
   p.lpad = lpad;
   TORCH_CHECK(p.ltiles16 <= 16, "mhsa_bwd: L must be <= 256");
-  const int smem_bytes = QT * lpad * 2 + QT * 32 * 8;
+  const int smem_bytes = QT * lpad * 2 + QT * 32 * 4;
   dim3 grid(B, (L + QT - 1) / QT);
   if (p.ltiles16 <= 4)
     hipLaunchKernelGGL(mhsa_bwd_q_kernel<4>, grid, dim3(256), smem_bytes,
