@@ -307,14 +307,18 @@ __global__ __launch_bounds__(256) void conv_wgrad_tr128_kernel(
   const int wm = wid >> 1, wn = wid & 1;
   const int il = lane & 15, kq = lane >> 4;
 
-  // staging: slot s = tid + it*256 (it<4): m = s>>4 (0..63), col8 = (s&15)*8
-  // subtile layout [m/32][col/16] of [32][16], stride SUBT
+  // staging: slot s = tid + it*256 (it<4): col8 = (s&7)*8 + (s>>9)*64,
+  // m = (s>>3)&63 — 8 consecutive lanes span 4 subtiles at +4-dword steps
+  // and each next 8-lane group rotates +8 dwords (m+1), so every 8-lane
+  // b128 write phase hits distinct banks (the 64x64 kernel's property; the
+  // first s>>4 mapping wrapped banks at lane 8 and cost ~2e9 conflict
+  // cycles in the PMC capture)
   int s_m[4], s_col[4], s_off[4];
 #pragma unroll
   for (int it = 0; it < 4; ++it) {
     const int sl = tid + it * 256;
-    s_m[it] = sl >> 4;
-    s_col[it] = (sl & 15) << 3;
+    s_m[it] = (sl >> 3) & 63;
+    s_col[it] = ((sl & 7) << 3) + ((sl >> 9) << 6);
     s_off[it] = ((s_m[it] >> 5) * 8 + (s_col[it] >> 4)) * SUBT +
                 (s_m[it] & 31) * 32 + (s_col[it] & 15) * 2;
   }
