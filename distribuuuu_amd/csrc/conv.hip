@@ -53,9 +53,11 @@ struct ConvParams {
   int bnact;                // 0 none, 1 relu
 };
 
-// EMODE: 0 plain, 1 forward BN sum/sumsq partials, 2 backward masked stats
-template <int EMODE>
+// EMODE: 0 plain, 1 forward BN sum/sumsq partials, 2 backward masked stats.
+// ACC: epilogue accumulates into y (residual-fork dgrad into gres).
+template <int EMODE, bool ACC = false>
 __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
+  static_assert(!(ACC && EMODE != 0), "ACC only with plain epilogue");
   const int g = blockIdx.z;
   // XCD-aware swizzle over m-tiles (T1; bijective form)
   int tile_m = blockIdx.x, tile_n = blockIdx.y;
@@ -208,7 +210,28 @@ __global__ __launch_bounds__(256) void conv_igemm_fwd_kernel(ConvParams p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (k0 + 16 <= p.Kg) {
+      if (ACC) {
+        if (k0 + 16 <= p.Kg) {
+          union {
+            __hip_bfloat16 b[16];
+            uint4 q[2];
+          } old;
+          old.q[0] = *reinterpret_cast<const uint4*>(&p.y[obase + k0]);
+          old.q[1] = *reinterpret_cast<const uint4*>(&p.y[obase + k0 + 8]);
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            u.b[j] = from_f32<__hip_bfloat16>(to_f32(u.b[j]) +
+                                              to_f32(old.b[j]));
+          *reinterpret_cast<uint4*>(&p.y[obase + k0]) = u.q[0];
+          *reinterpret_cast<uint4*>(&p.y[obase + k0 + 8]) = u.q[1];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            if (k0 + j < p.Kg)
+              p.y[obase + k0 + j] = from_f32<__hip_bfloat16>(
+                  to_f32(u.b[j]) + to_f32(p.y[obase + k0 + j]));
+        }
+      } else if (k0 + 16 <= p.Kg) {
         *reinterpret_cast<uint4*>(&p.y[obase + k0]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + k0 + 8]) = u.q[1];
       } else {
@@ -832,14 +855,15 @@ struct BnBwdEmit {
 at::Tensor conv2d_fwd_v2_into(at::Tensor x, at::Tensor w, at::Tensor y,
                               int64_t Ho, int64_t Wo, int64_t groups,
                               int64_t osh, int64_t osw, int64_t oh0,
-                              int64_t ow0);
+                              int64_t ow0, bool acc = false);
 
 at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
                            int64_t Wo, int64_t sh, int64_t sw, int64_t ph,
                            int64_t pw, int64_t dh, int64_t dw, int64_t groups,
                            int64_t osh = 1, int64_t osw = 1, int64_t oh0 = 0,
                            int64_t ow0 = 0, at::Tensor* part_out = nullptr,
-                           const BnBwdEmit* bemit = nullptr) {
+                           const BnBwdEmit* bemit = nullptr,
+                           bool acc = false) {
   CHECK_GPU(x);
   if (x.scalar_type() == at::kFloat)
     return conv2d_fwd_into_fp32(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw,
@@ -860,7 +884,8 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
         (Si * Cgi) % 64 == 0 && Kgi >= 96 &&
         (int64_t)Ri * Si * Cgi >= 512 && Cgi % 8 == 0 &&
         Ho + Ri - 1 <= x.size(2) && Wo + Si - 1 <= x.size(3))
-      return conv2d_fwd_v2_into(x, w, y, Ho, Wo, groups, osh, osw, oh0, ow0);
+      return conv2d_fwd_v2_into(x, w, y, Ho, Wo, groups, osh, osw, oh0, ow0,
+                                acc);
   }
   check_nhwc(x, "x");
   check_nhwc(w, "w");
@@ -906,6 +931,9 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
     p.part = part_out->data_ptr<float>();
     hipLaunchKernelGGL(conv_igemm_fwd_kernel<1>, grid, dim3(256), 0,
                        cur_stream(), p);
+  } else if (acc) {
+    hipLaunchKernelGGL((conv_igemm_fwd_kernel<0, true>), grid, dim3(256), 0,
+                       cur_stream(), p);
   } else {
     hipLaunchKernelGGL(conv_igemm_fwd_kernel<0>, grid, dim3(256), 0,
                        cur_stream(), p);
@@ -921,7 +949,8 @@ at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
                               int64_t Cg_, int64_t R_, int64_t S_, int64_t sh,
                               int64_t sw, int64_t ph, int64_t pw, int64_t dh,
                               int64_t dw, int64_t groups,
-                              at::Tensor* part_out, const BnBwdEmit* bemit);
+                              at::Tensor* part_out, const BnBwdEmit* bemit,
+                              at::Tensor* acc_into = nullptr);
 
 static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
                                   int64_t sw, int64_t ph, int64_t pw,
@@ -1399,4 +1428,86 @@ std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor gy, at::Tensor w,
                               &part, &em);
   if (!part.defined()) part = at::empty({0, 0}, bnscale.options());
   return {gx, part};
+}
+
+// Residual-fork gradient accumulation (docs/ARCHITECTURE.md "Round-2
+// status"): compute this conv's dgrad and ADD it into `into` — the other
+// fork branch's already-computed gradient — inside the dgrad epilogue
+// (+1 read stream) instead of autograd's separate 2-read/1-write
+// elementwise add over the whole gx. Returns {tensor, flag}: flag 1 means
+// `into` now holds the sum; flag 0 means the dispatched route has no ACC
+// epilogue and the returned tensor is a FRESH gx (caller must add).
+std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
+    at::Tensor gy, at::Tensor w, int64_t H, int64_t W, int64_t sh, int64_t sw,
+    int64_t ph, int64_t pw, int64_t dh, int64_t dw, int64_t groups,
+    at::Tensor into) {
+  CHECK_GPU(gy);
+  const int N = gy.size(0);
+  const int Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Ct = Cg * groups;
+  const int Kt0 = gy.size(1);
+  const int dKg0 = Ct / (int)groups;
+  const int dCg0 = Kt0 / (int)groups;
+  const bool same_size = sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
+                         dw * (S - 1) == 2 * pw;
+  static const int v2mink3 = []() {
+    const char* e = getenv("DISTRIBUUUU_V2_MINK");
+    return e ? atoi(e) : 192;
+  }();
+  static const bool v2off3 = []() {
+    const char* e = getenv("DISTRIBUUUU_CONV_V2");
+    return e && e[0] == '0';
+  }();
+  if (gy.scalar_type() == at::kBFloat16 && groups == 1 && dCg0 % 8 == 0 &&
+      into.scalar_type() == at::kBFloat16 &&
+      into.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+      into.sizes() == at::IntArrayRef({(int64_t)N, (int64_t)Ct, H, W})) {
+    if (same_size && !v2off3 && Kt0 % 8 == 0 && dKg0 >= v2mink3 &&
+        (int64_t)R * S * dCg0 >= 512) {
+      const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
+      auto wsp = weight_flip_t_span(w, 1, SPAN64);
+      conv2d_fwd_v2_flat(gy, wsp, Ct, dCg0, R, S, 1, 1, dh * (R - 1) - ph,
+                         dw * (S - 1) - pw, dh, dw, 1, nullptr, nullptr,
+                         &into);
+      return {into, 1};
+    }
+    if (same_size) {
+      auto wt = weight_flip_t(w, 1);
+      conv2d_fwd_into(gy, wt, into, H, W, 1, 1, dh * (R - 1) - ph,
+                      dw * (S - 1) - pw, dh, dw, 1, 1, 1, 0, 0, nullptr,
+                      nullptr, true);
+      return {into, 1};
+    }
+    if (R == 1 && S == 1 && (sh > 1 || sw > 1) && ph == 0 && pw == 0) {
+      // strided 1x1 proj: scatter-accumulate at (ho*sh, wo*sw); the gap
+      // positions receive no main-branch gradient, so `into` already holds
+      // their final values.
+      auto wt = weight_flip_t(w, 1);
+      conv2d_fwd_into(gy, wt, into, gy.size(2), gy.size(3), 1, 1, 0, 0, 1, 1,
+                      1, sh, sw, 0, 0, nullptr, nullptr, true);
+      return {into, 1};
+    }
+    if (R == 3 && S == 3 && sh == 2 && sw == 2 && dh == 1 && dw == 1 &&
+        ph == 1 && pw == 1) {
+      // parity decomposition (see conv2d_dgrad_impl): the four parity
+      // classes tile gx disjointly, so each sub-conv scatter-accumulates
+      // its own positions exactly once.
+      auto wt = weight_flip_t(w, 1);
+      for (int h0 = 0; h0 < 2; ++h0) {
+        const int64_t hu = (H - h0 + 1) >> 1;
+        auto wr = (h0 == 0) ? wt.slice(2, 1, 2) : wt.slice(2, 0, 3, 2);
+        for (int w0 = 0; w0 < 2; ++w0) {
+          const int64_t wu = (W - w0 + 1) >> 1;
+          auto wsub = ((w0 == 0) ? wr.slice(3, 1, 2) : wr.slice(3, 0, 3, 2))
+                          .contiguous(at::MemoryFormat::ChannelsLast);
+          conv2d_fwd_into(gy, wsub, into, hu, wu, 1, 1, 0, 0, 1, 1, 1, 2, 2,
+                          h0, w0, nullptr, nullptr, true);
+        }
+      }
+      return {into, 1};
+    }
+  }
+  auto gx = conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
+                              nullptr, nullptr);
+  return {gx, 0};
 }

@@ -51,8 +51,12 @@ DEV_INLINE int swz(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
 // CLAMP: span-tail reads of the last pixels clamp to the final in-bounds
 // 16-byte address (garbage x zero-padded weight) so unaligned-span shapes
 // read x DIRECTLY instead of taking a physical-order copy with slack.
-template <int EMODE, bool CLAMP = false>
+// ACC: the epilogue ACCUMULATES into y instead of overwriting it — dgrads
+// of residual-forked tensors add straight into the BN's gres buffer and
+// the fork's separate gradient-sum kernel disappears.
+template <int EMODE, bool CLAMP = false, bool ACC = false>
 __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
+  static_assert(!(ACC && EMODE != 0), "ACC only with plain epilogue");
   const int g = blockIdx.z;
   int tile_m = blockIdx.x, tile_n = blockIdx.y;
   {  // XCD-aware bijective remap over m-tiles (T1)
@@ -221,7 +225,28 @@ __global__ __launch_bounds__(512) void conv_igemm_v2_kernel(Conv2Params p) {
 #pragma unroll
       for (int j = 0; j < 16; ++j)
         u.b[j] = from_f32<__hip_bfloat16>(slab[er * 68 + ec + j]);
-      if (k0 + 16 <= p.K) {
+      if (ACC) {
+        if (k0 + 16 <= p.K) {
+          union {
+            __hip_bfloat16 b[16];
+            uint4 q[2];
+          } old;
+          old.q[0] = *reinterpret_cast<const uint4*>(&p.y[obase + ec]);
+          old.q[1] = *reinterpret_cast<const uint4*>(&p.y[obase + ec + 8]);
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            u.b[j] = from_f32<__hip_bfloat16>(to_f32(u.b[j]) +
+                                              to_f32(old.b[j]));
+          *reinterpret_cast<uint4*>(&p.y[obase + ec]) = u.q[0];
+          *reinterpret_cast<uint4*>(&p.y[obase + ec + 8]) = u.q[1];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j)
+            if (k0 + j < p.K)
+              p.y[obase + ec + j] = from_f32<__hip_bfloat16>(
+                  to_f32(u.b[j]) + to_f32(p.y[obase + ec + j]));
+        }
+      } else if (k0 + 16 <= p.K) {
         *reinterpret_cast<uint4*>(&p.y[obase + ec]) = u.q[0];
         *reinterpret_cast<uint4*>(&p.y[obase + ec + 8]) = u.q[1];
       } else {
@@ -365,7 +390,8 @@ at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
                               int64_t Cg_, int64_t R_, int64_t S_, int64_t sh,
                               int64_t sw, int64_t ph, int64_t pw, int64_t dh,
                               int64_t dw, int64_t groups,
-                              at::Tensor* part_out, const BnBwdEmit* bemit) {
+                              at::Tensor* part_out, const BnBwdEmit* bemit,
+                              at::Tensor* acc_into = nullptr) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "v2: bf16 only");
   check_nhwc(x, "x");
@@ -402,8 +428,19 @@ at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
                        pw);
     xin = xp;
   }
-  auto y = at::empty({N, Kt, Ho, Wo},
-                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  at::Tensor y;
+  if (acc_into != nullptr) {
+    TORCH_CHECK(part_out == nullptr && bemit == nullptr,
+                "v2_flat: acc excludes partial emission");
+    TORCH_CHECK(acc_into->sizes() ==
+                    at::IntArrayRef({(int64_t)N, (int64_t)Kt, (int64_t)Ho,
+                                     (int64_t)Wo}),
+                "v2_flat: acc_into shape");
+    y = *acc_into;
+  } else {
+    y = at::empty({N, Kt, Ho, Wo},
+                  x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  }
   Conv2Params p;
   p.x = (const __hip_bfloat16*)xin.data_ptr();
   p.w = (const __hip_bfloat16*)wp.data_ptr();
@@ -449,6 +486,13 @@ at::Tensor conv2d_fwd_v2_flat(at::Tensor x, at::Tensor wp, int64_t Kt_,
     else
       hipLaunchKernelGGL(conv_igemm_v2_kernel<1>, grid, dim3(512), 0,
                          cur_stream(), p);
+  } else if (acc_into != nullptr) {
+    if (clamp_tail)
+      hipLaunchKernelGGL((conv_igemm_v2_kernel<0, true, true>), grid,
+                         dim3(512), 0, cur_stream(), p);
+    else
+      hipLaunchKernelGGL((conv_igemm_v2_kernel<0, false, true>), grid,
+                         dim3(512), 0, cur_stream(), p);
   } else {
     if (clamp_tail)
       hipLaunchKernelGGL((conv_igemm_v2_kernel<0, true>), grid, dim3(512), 0,
@@ -493,7 +537,7 @@ at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
 at::Tensor conv2d_fwd_v2_into(at::Tensor x, at::Tensor w, at::Tensor y,
                               int64_t Ho, int64_t Wo, int64_t groups,
                               int64_t osh, int64_t osw, int64_t oh0,
-                              int64_t ow0) {
+                              int64_t ow0, bool acc = false) {
   const int N = x.size(0), Ct = x.size(1), H = x.size(2), W = x.size(3);
   const int Kt = w.size(0), Cg = w.size(1), R = w.size(2), S = w.size(3);
   const int C = Cg, K = Kt / (int)groups;
@@ -519,7 +563,11 @@ at::Tensor conv2d_fwd_v2_into(at::Tensor x, at::Tensor w, at::Tensor y,
   p.bnx = nullptr; p.bnscale = nullptr; p.bnshift = nullptr; p.bnact = 0;
   TORCH_CHECK(Ho + R - 1 <= H && Wo + S - 1 <= W, "v2_into: window OOB");
   dim3 grid(p.tiles_m, (K + BN2 - 1) / BN2, groups);
-  hipLaunchKernelGGL(conv_igemm_v2_kernel<0>, grid, dim3(512), 0,
-                     cur_stream(), p);
+  if (acc)
+    hipLaunchKernelGGL((conv_igemm_v2_kernel<0, false, true>), grid, dim3(512),
+                       0, cur_stream(), p);
+  else
+    hipLaunchKernelGGL(conv_igemm_v2_kernel<0>, grid, dim3(512), 0,
+                       cur_stream(), p);
   return y;
 }

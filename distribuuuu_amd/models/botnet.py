@@ -17,6 +17,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import AdaptiveAvgPool2d, AvgPool2d, BatchNorm2d, Conv2d, Linear
+from ..ops import functional as DF
 from ..ops.attention import mhsa_relpos, mhsa_relpos_nhwc
 from .resnet import resnet50
 
@@ -109,8 +110,9 @@ class BoTBlock(nn.Module):
         nn.init.zeros_(self.bn3.weight)
 
     def forward(self, x):
-        identity = x if self.shortcut is None else self.shortcut(x)
-        out = self.bn1(self.conv1(x))
+        xm, xs = DF.fork(x)
+        identity = xs if self.shortcut is None else self.shortcut(xs)
+        out = self.bn1(self.conv1(xm))
         out = self.mhsa(out)
         if self.pool is not None:
             out = self.pool(out)

@@ -52,13 +52,19 @@ class MBConv(nn.Module):
         self.project_bn = BatchNorm2d(c_out, eps=1e-3)
 
     def forward(self, x):
-        out = x if self.expand is None else self.expand(x)
+        if not self.use_residual:
+            out = x if self.expand is None else self.expand(x)
+            out = self.dw_bn(self.dw(out))
+            out = self.se(out)
+            return self.project_bn(self.project(out))
+        # residual block: fuse the add into project_bn (same output) and
+        # fork the input so the expand conv's dgrad can accumulate into
+        # the shortcut gradient (see ops.functional.fork)
+        xm, xs = DF.fork(x)
+        out = xm if self.expand is None else self.expand(xm)
         out = self.dw_bn(self.dw(out))
         out = self.se(out)
-        out = self.project_bn(self.project(out))
-        if self.use_residual:
-            out = out + x
-        return out
+        return self.project_bn(self.project(out), residual=xs)
 
 
 # (expand_ratio, channels, repeats, stride, kernel) — B0 table

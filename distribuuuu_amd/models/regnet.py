@@ -77,8 +77,9 @@ class RegBlock(nn.Module):
         nn.init.zeros_(self.c_bn.weight)
 
     def forward(self, x):
-        identity = x if self.proj is None else self.proj(x)
-        out = self.a_bn(self.a(x))
+        xm, xs = DF.fork(x)
+        identity = xs if self.proj is None else self.proj(xs)
+        out = self.a_bn(self.a(xm))
         out = self.b_bn(self.b(out))
         if self.se is not None:
             out = self.se(out)

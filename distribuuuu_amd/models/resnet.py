@@ -12,6 +12,7 @@ checkpoints load directly.
 
 import torch.nn as nn
 
+from ..ops import functional as DF
 from ..ops import AdaptiveAvgPool2d, BatchNorm2d, Conv2d, Linear, MaxPool2d
 
 
@@ -40,8 +41,9 @@ class BasicBlock(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
+        xm, xs = DF.fork(x)
+        identity = xs if self.downsample is None else self.downsample(xs)
+        out = self.bn1(self.conv1(xm))
         out = self.conv2(out)
         return self.bn2(out, residual=identity)
 
@@ -64,8 +66,9 @@ class Bottleneck(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
+        xm, xs = DF.fork(x)
+        identity = xs if self.downsample is None else self.downsample(xs)
+        out = self.bn1(self.conv1(xm))
         out = self.bn2(self.conv2(out))
         out = self.conv3(out)
         return self.bn3(out, residual=identity)

@@ -54,7 +54,7 @@ class _HIPConv2d(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, bias, stride, padding, dilation, groups, emit_part,
-                bnx, bnscale, bnshift, bnact, bnslot):
+                bnx, bnscale, bnshift, bnact, bnslot, fork_slot):
         x = _cl(x)
         w = _cl(w)
         e = ext()
@@ -88,6 +88,7 @@ class _HIPConv2d(torch.autograd.Function):
                     kout, cpad, kpad)
         ctx.bnact = bnact
         ctx.bnslot = bnslot if (bnx is not None and not cpad) else None
+        ctx.fork_slot = fork_slot
         ctx.mark_non_differentiable(part)
         return y, part
 
@@ -105,7 +106,18 @@ class _HIPConv2d(torch.autograd.Function):
             wd = e.pad_channels(w, _pad8(cin)) if cpad else w
             if kpad:
                 wd = _pad_k(wd, _pad8(kout))
-            if ctx.bnslot is not None and len(saved) == 5:
+            fslot = ctx.fork_slot
+            fbuf = fslot.get("g") if fslot is not None else None
+            if fbuf is not None and not cpad and fbuf.dtype == gyp.dtype:
+                # residual fork (see _Fork): the other branch's gradient is
+                # already in fbuf — accumulate this dgrad into it inside the
+                # epilogue (one extra read stream) instead of letting
+                # autograd run a separate whole-tensor add
+                gx, acc_done = e.conv2d_dgrad_acc(
+                    gyp, wd, x.shape[2], x.shape[3], stride[0], stride[1],
+                    padding[0], padding[1], dilation[0], dilation[1], groups,
+                    fbuf)
+            elif ctx.bnslot is not None and len(saved) == 5:
                 # dgrad + BN-backward stats from the same epilogue: the
                 # consuming BN reads the partials IF this gx arrives there
                 # unmodified (data_ptr identity check in its backward)
@@ -124,6 +136,8 @@ class _HIPConv2d(torch.autograd.Function):
                                     groups)
             if cpad:
                 gx = _cl(gx[:, :cin])
+            if fslot is not None and fbuf is None:
+                fslot["g"] = gx
         if ctx.needs_input_grad[1]:
             xw = e.pad_channels(x, _pad8(cin)) if cpad else x
             gw = e.conv2d_wgrad(gyp, xw, w.shape[2], w.shape[3],
@@ -139,7 +153,7 @@ class _HIPConv2d(torch.autograd.Function):
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
         return (gx, gw, gb, None, None, None, None, None, None, None, None,
-                None, None)
+                None, None, None)
 
 
 class _HIPDepthwiseConv2d(torch.autograd.Function):
@@ -235,9 +249,11 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
         else:
             bnx = bnscale = bnshift = bnslot = None
             bnact = 0
+        fslot = (getattr(x, "_fork_slot", None)
+                 if torch.is_grad_enabled() else None)
         y, part = _HIPConv2d.apply(x, weight, bias, stride, padding, dilation,
                                    groups, emit, bnx, bnscale, bnshift, bnact,
-                                   bnslot)
+                                   bnslot, fslot)
         if part.numel():
             y._bn_partials = part
         elif bias is None:
@@ -265,7 +281,7 @@ class _HIPBatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, mean, rstd, scale, shift, training,
-                act_id, residual, slot):
+                act_id, residual, slot, fork_slot):
         e = ext()
         res = _cl(residual) if residual is not None else None
         y = e.bn_apply_act(x, scale, shift, act_id, res)
@@ -275,6 +291,7 @@ class _HIPBatchNormAct(torch.autograd.Function):
         ctx.training = training
         ctx.has_res = residual is not None
         ctx.slot = slot
+        ctx.fork_slot = fork_slot
         return y
 
     @staticmethod
@@ -300,8 +317,14 @@ class _HIPBatchNormAct(torch.autograd.Function):
             gx, gw, gb, gres = e.bn_bwd(
                 gy, x, res if ctx.has_res else None, mean, rstd, gamma,
                 scale, shift, ctx.act_id, ctx.training, ctx.has_res)
+        if (ctx.has_res and ctx.fork_slot is not None
+                and "g" not in ctx.fork_slot):
+            # the shortcut branch of a residual fork: deposit gres so the
+            # main branch's final dgrad can accumulate into it (see _Fork)
+            ctx.fork_slot["g"] = gres
         return (gx, gw.to(weight.dtype), gb.to(weight.dtype), None, None,
-                None, None, None, None, gres if ctx.has_res else None, None)
+                None, None, None, None, gres if ctx.has_res else None, None,
+                None)
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
@@ -339,8 +362,12 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
                 and torch.is_grad_enabled()
                 and os.environ.get("DISTRIBUUUU_BN_BWD_FUSE", "0") == "1"):
             slot = {"gx": None, "parts": []}
+        res_fslot = (getattr(residual, "_fork_slot", None)
+                     if residual is not None and torch.is_grad_enabled()
+                     else None)
         y = _HIPBatchNormAct.apply(x, weight, bias, mean, rstd, scale, shift,
-                                   training, act_id, residual, slot)
+                                   training, act_id, residual, slot,
+                                   res_fslot)
         if slot is not None:
             # consuming convs pick this up and emit BN-backward stats from
             # their dgrad epilogue (see _HIPConv2d / conv2d_dgrad_bn)
@@ -351,6 +378,45 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
     if residual is not None:
         y = y + residual
     return _apply_act(y, act)
+
+
+class _Fork(torch.autograd.Function):
+    """Explicit residual fork: ``x1, x2 = fork(x)`` marks the two consumers
+    of a residual-block input so their backwards can chain gradient
+    accumulation. The reference (see torchvision-style blocks in
+    distribuuuu) leans on autograd's implicit fan-in add — a whole-tensor
+    2-read/1-write elementwise pass per block. Here the first branch to
+    finish deposits its gradient in a shared slot, the second accumulates
+    into that buffer inside its dgrad epilogue (conv2d_dgrad_acc) or
+    deposits gres (BN shortcut), and this backward detects pointer identity
+    and skips the add. Falls back to g1 + g2 whenever a route could not
+    accumulate."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return x.view_as(x), x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, g1, g2):
+        if g1 is None:
+            return g2
+        if g2 is None:
+            return g1
+        if g1.data_ptr() == g2.data_ptr():
+            return g1
+        return g1 + g2
+
+
+def fork(x):
+    """Split a residual-block input into (main, shortcut) fork handles."""
+    if not (x.is_cuda and torch.is_grad_enabled() and x.requires_grad
+            and x.dtype == torch.bfloat16 and use_hip(x, "conv2d_fwd")):
+        return x, x
+    x1, x2 = _Fork.apply(x)
+    slot = {}
+    x1._fork_slot = slot
+    x2._fork_slot = slot
+    return x1, x2
 
 
 def _apply_act(y, act):

@@ -421,3 +421,87 @@ def test_wgrad_ring128_grouped_kg_tail(monkeypatch):
     err = (gw.float() - wf.grad).abs().max().item()
     scale = wf.grad.abs().max().item()
     assert err < 3e-2 * max(scale, 1.0), (err, scale)
+
+
+@pytest.mark.gpu
+class TestDgradAcc:
+    """conv2d_dgrad_acc: dgrad accumulated into the fork partner's gradient
+    buffer inside the epilogue (docs/ARCHITECTURE.md, residual forks)."""
+
+    def _check(self, N, C, H, W, K, R, stride, pad, groups=1, want_acc=True):
+        e = _ext()
+        cl = torch.channels_last
+        ho = (H + 2 * pad - R) // stride + 1
+        gy = torch.randn(N, K, ho, ho, device="cuda",
+                         dtype=torch.bfloat16).contiguous(memory_format=cl)
+        w = torch.randn(K, C // groups, R, R, device="cuda",
+                        dtype=torch.bfloat16).contiguous(memory_format=cl)
+        w = w * 0.05
+        into = torch.randn(N, C, H, W, device="cuda",
+                           dtype=torch.bfloat16).contiguous(memory_format=cl)
+        base = into.clone()
+        gx_ref = e.conv2d_dgrad(gy, w, H, W, stride, stride, pad, pad, 1, 1,
+                                groups)
+        out, flag = e.conv2d_dgrad_acc(gy, w, H, W, stride, stride, pad, pad,
+                                       1, 1, groups, into)
+        assert bool(flag) == want_acc
+        if want_acc:
+            assert out.data_ptr() == into.data_ptr()
+            ref = base.float() + gx_ref.float()
+        else:
+            ref = gx_ref.float()
+        err = (out.float() - ref).abs().max().item()
+        scl = ref.abs().max().item() + 1e-6
+        assert err / scl < 3e-2, (err, scl)
+
+    def test_same_size_1x1_v1(self):
+        self._check(8, 256, 28, 28, 64, 1, 1, 0)
+
+    def test_same_size_3x3_v2(self):
+        self._check(8, 512, 14, 14, 512, 3, 1, 1)
+
+    def test_strided_1x1_proj(self):
+        self._check(8, 512, 28, 28, 1024, 1, 2, 0)
+
+    def test_parity_3x3_s2(self):
+        self._check(8, 64, 32, 32, 64, 3, 2, 1)
+
+    def test_grouped_falls_back(self):
+        self._check(8, 256, 14, 14, 256, 3, 1, 1, groups=32, want_acc=False)
+
+
+@pytest.mark.gpu
+def test_fork_grads_match_unforked():
+    """A residual stack trained through fork() must produce the same grads
+    as the plain two-consumer autograd graph (implicit fan-in add)."""
+    from distribuuuu_amd.models.resnet import Bottleneck
+    from distribuuuu_amd.ops import functional as DF
+
+    def run(seed, forked):
+        orig = DF.fork
+        if not forked:
+            DF.fork = lambda t: (t, t)
+        try:
+            torch.manual_seed(seed)
+            blk = torch.nn.Sequential(
+                Bottleneck(256, 64), Bottleneck(256, 64)).cuda().bfloat16()
+            blk = blk.to(memory_format=torch.channels_last)
+            x = torch.randn(4, 256, 28, 28, device="cuda",
+                            dtype=torch.bfloat16).contiguous(
+                                memory_format=torch.channels_last)
+            x.requires_grad_(True)
+            y = blk(x)
+            y.float().square().mean().backward()
+            return x.grad.float().cpu(), [
+                p.grad.float().cpu() for p in blk.parameters()
+                if p.grad is not None]
+        finally:
+            DF.fork = orig
+
+    gx_f, gp_f = run(7, True)
+    gx_p, gp_p = run(7, False)
+    s = gx_p.abs().max().item() + 1e-6
+    assert (gx_f - gx_p).abs().max().item() / s < 3e-2
+    for a, b in zip(gp_f, gp_p):
+        s = b.abs().max().item() + 1e-6
+        assert (a - b).abs().max().item() / s < 3e-2
