@@ -285,14 +285,27 @@ struct WgradT128Params {
   int Ho, Wo;
   int M, RSC;
   int ktiles, ntiles, chunks, csteps;
+  int xcd_remap;
   unsigned long long magicHoWo, magicWo;
 };
 
 __global__ __launch_bounds__(256) void conv_wgrad_tr128_kernel(
     WgradT128Params p) {
   const int g = blockIdx.z;
-  const int ktile = blockIdx.x % p.ktiles;
-  const int ntile = blockIdx.x / p.ktiles;
+  int tile = blockIdx.x;
+  {  // XCD-aware bijective remap (same form as conv2.hip T1): each XCD's
+     // dispatch subsequence (blockIdx.x % 8) gets a CONTIGUOUS tile range,
+     // decomposed ntile-fastest below, so the staged A (gy) chunk-slab is
+     // re-served from that XCD's L2 across its ntile walk instead of
+     // re-read from HBM ntiles times.
+    const int tiles = p.ktiles * p.ntiles;
+    const int q = tiles >> 3, r8 = tiles & 7;
+    const int xcd = tile % 8, idx = tile / 8;
+    if (p.xcd_remap)
+      tile = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q) + idx;
+  }
+  const int ktile = p.xcd_remap ? tile / p.ntiles : tile % p.ktiles;
+  const int ntile = p.xcd_remap ? tile % p.ntiles : tile / p.ktiles;
   const int chunk = blockIdx.y;
 
   __shared__ __align__(16) char smem[2 * 2 * T128_TILE];
@@ -882,10 +895,8 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
                        ((R * S * Cg) % 16 == 0);
   // measured: the ring variant trails the tr-staged kernel on most shapes
   // (its 8-MFMA k-steps don't cover the pipeline); keep it opt-in.
-  static const bool ring_on = []() {
-    const char* e = getenv("DISTRIBUUUU_WGRAD_RING");
-    return e && e[0] == '1';
-  }();
+  const char* ring_env = getenv("DISTRIBUUUU_WGRAD_RING");
+  const bool ring_on = ring_env && ring_env[0] == '1';
   if (ring_ok && ring_on) {
     at::Tensor xin = x;
     int Hp = H, Wp = W;
@@ -1032,6 +1043,10 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
       q.RSC = R * S * Cg;
       q.ktiles = (Kg + T128_BM - 1) / T128_BM;
       q.ntiles = (q.RSC + T128_BN - 1) / T128_BN;
+      {
+        const char* xr = getenv("DISTRIBUUUU_WGRAD_XCD");
+        q.xcd_remap = xr && xr[0] == '0' ? 0 : 1;
+      }
       q.magicHoWo = ((1ULL << 47) / ((unsigned long long)Ho * Wo)) + 1;
       q.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
       int csteps = CHUNK_STEPS;

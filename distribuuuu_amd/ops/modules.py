@@ -66,10 +66,29 @@ class BatchNorm2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # batch counter kept as a host int and folded into the buffer only
+        # when the state dict is read: the per-step `buffer += 1` was a
+        # 4.7 us GPU launch per BN layer (53/step on ResNet-50, ~0.25
+        # ms/step inside the captured graph) for a value nothing reads
+        # during training
+        self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        self._nbt_pending = 0
+        super()._load_from_state_dict(*args, **kwargs)
+
+    def _flush_nbt(self):
+        if self._nbt_pending:
+            self.num_batches_tracked += self._nbt_pending
+            self._nbt_pending = 0
 
     def forward(self, x, residual=None):
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt_pending += 1
         return DF.batch_norm_act(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, self.act, residual,

@@ -131,7 +131,7 @@ class SyncBatchNorm(BatchNorm2d):
         if (not self.training or not dist.is_initialized()
                 or dist.get_world_size(self.process_group) == 1):
             return super().forward(x, residual)
-        self.num_batches_tracked += 1
+        self._nbt_pending += 1
         rm, rv = self.running_mean, self.running_var
         return _SyncBNFunction.apply(
             x, self.weight, self.bias, rm, rv, self.momentum, self.eps,
