@@ -709,8 +709,10 @@ std::vector<at::Tensor> mhsa_bwd(at::Tensor dO, at::Tensor P, at::Tensor q,
   dim3 gkv(B, (L + 63) / 64);
   hipLaunchKernelGGL(mhsa_bwd_kv_kernel, gkv, dim3(256), 0, cur_stream(), kv);
 
-  auto grw = at::zeros({MW, D}, fopts);
-  auto grh = at::zeros({MH, D}, fopts);
+  auto grw = at::empty({MW, D}, fopts);
+  auto grh = at::empty({MH, D}, fopts);
+  hipMemsetAsync(grw.data_ptr(), 0, grw.numel() * 4, cur_stream());
+  hipMemsetAsync(grh.data_ptr(), 0, grh.numel() * 4, cur_stream());
   const int64_t rows = (int64_t)B * L;
   const int rpb = 128;
   for (int which = 0; which < 2; ++which) {

@@ -220,7 +220,7 @@ at::Tensor dwconv_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
   const int Ho = gy.size(2), Wo = gy.size(3);
   auto acc = at::empty({C, 1, (int64_t)R, (int64_t)S},
                        x.options().dtype(at::kFloat));
-  acc.zero_();
+  hipMemsetAsync(acc.data_ptr(), 0, acc.numel() * 4, cur_stream());
   const int64_t pixels = (int64_t)N * Ho * Wo;
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "dwconv_wgrad", [&] {
     constexpr int V = 16 / sizeof(scalar_t);

@@ -256,7 +256,8 @@ std::vector<at::Tensor> se_scale_bwd(at::Tensor gy, at::Tensor x,
   const int N = x.size(0), C = x.size(1);
   const int HW = x.size(2) * x.size(3);
   auto gx = at::empty_like(x);
-  auto gsf = at::zeros({N, C}, x.options().dtype(at::kFloat));
+  auto gsf = at::empty({N, C}, x.options().dtype(at::kFloat));
+  hipMemsetAsync(gsf.data_ptr(), 0, gsf.numel() * 4, cur_stream());
   const int chunks = std::max<int>(1, (int)ceil_div(768, N));
   const int rpc = (int)ceil_div(HW, chunks);
   DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "se_scale_bwd", [&] {

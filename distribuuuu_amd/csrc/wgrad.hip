@@ -925,7 +925,7 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
     q.ntiles = (q.RSC + WBN - 1) / WBN;
     const int chunks = (int)((M64 + CHUNK_STEPS * WBK - 1) / (CHUNK_STEPS * WBK));
     auto accbuf2 = at::empty({(int64_t)Kt, q.RSC}, x.options().dtype(at::kFloat));
-    accbuf2.zero_();
+    hipMemsetAsync(accbuf2.data_ptr(), 0, accbuf2.numel() * 4, cur_stream());
     q.acc = accbuf2.data_ptr<float>();
     dim3 grid2(q.ktiles * q.ntiles, chunks, groups);
     hipLaunchKernelGGL(conv_wgrad_ring_kernel, grid2, dim3(256), 0,
@@ -999,7 +999,7 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
     q.csteps = csteps;
     auto accbuf = at::empty({(int64_t)Kt, q.RSC},
                             x.options().dtype(at::kFloat));
-    accbuf.zero_();
+    hipMemsetAsync(accbuf.data_ptr(), 0, accbuf.numel() * 4, cur_stream());
     q.acc = accbuf.data_ptr<float>();
     static bool attr_done = false;
     if (!attr_done) {
@@ -1058,8 +1058,9 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
       }
       q.csteps = csteps;
       q.chunks = chunks;
-      auto accb = at::zeros({(int64_t)Kt, q.RSC},
+      auto accb = at::empty({(int64_t)Kt, q.RSC},
                             x.options().dtype(at::kFloat));
+      hipMemsetAsync(accb.data_ptr(), 0, accb.numel() * 4, cur_stream());
       q.acc = accb.data_ptr<float>();
       dim3 grid(q.ktiles * q.ntiles, chunks, groups);
       hipLaunchKernelGGL(conv_wgrad_tr128_kernel, grid, dim3(256), 0,
@@ -1092,7 +1093,7 @@ at::Tensor conv2d_wgrad(at::Tensor gy, at::Tensor x, int64_t R, int64_t S,
   p.magicWo = ((1ULL << 47) / (unsigned long long)Wo) + 1;
 
   auto accbuf = at::empty({(int64_t)Kt, p.RSC}, x.options().dtype(at::kFloat));
-  accbuf.zero_();
+  hipMemsetAsync(accbuf.data_ptr(), 0, accbuf.numel() * 4, cur_stream());
   p.acc = accbuf.data_ptr<float>();
   dim3 grid(p.ktiles * p.ntiles, p.chunks, groups);
   if (p.ktiles * p.ntiles <= 12)  // XCD-grouped 1-D grid (see kernel)
