@@ -15,6 +15,26 @@ from .dispatch import ext, fallback_warn, use_hip
 
 _ACTS = {"none": 0, "relu": 1, "silu": 2, "sigmoid": 3}
 
+_WGRAD_STREAM = None
+
+
+def _wgrad_stream():
+    """Side HIP stream for weight-gradient kernels. dgrad and wgrad of a
+    conv are independent consumers of the same gy; both kernel families are
+    latency- (not bandwidth-) bound on MI355X (profiles/wgrad_pmc_r2.md), so
+    co-residency on two streams fills each other's stall slots. Ordering is
+    by stream events (wait_stream both ways), which hipGraph capture turns
+    into graph edges, and the join happens before backward returns — so
+    DDP's reducer (which orders against the autograd stream) stays correct
+    at any world size. Env DISTRIBUUUU_WGRAD_STREAM=0 disables."""
+    global _WGRAD_STREAM
+    if _WGRAD_STREAM is None:
+        if os.environ.get("DISTRIBUUUU_WGRAD_STREAM", "1") == "0":
+            _WGRAD_STREAM = False
+        else:
+            _WGRAD_STREAM = torch.cuda.Stream()
+    return _WGRAD_STREAM
+
 
 def _cl(x):
     """Ensure channels_last physical layout for 4-D GPU tensors."""
@@ -37,6 +57,21 @@ def _pad_k(w, k8):
         memory_format=torch.channels_last)
     wp[:k] = w
     return wp
+
+
+def _conv_wgrad(e, gyp, x, w, stride, padding, dilation, groups, cin, kout,
+                cpad, kpad):
+    xw = e.pad_channels(x, _pad8(cin)) if cpad else x
+    gw = e.conv2d_wgrad(gyp, xw, w.shape[2], w.shape[3], stride[0], stride[1],
+                        padding[0], padding[1], dilation[0], dilation[1],
+                        groups)
+    if kpad:
+        gw = gw[:kout]
+    if cpad:
+        gw = gw[:, :cin]
+    if kpad or cpad:
+        gw = _cl(gw)
+    return gw
 
 
 class _HIPConv2d(torch.autograd.Function):
@@ -102,6 +137,16 @@ class _HIPConv2d(torch.autograd.Function):
         e = ext()
         gx = gw = gb = None
         gyp = e.pad_channels(gy, _pad8(kout)) if kpad else gy
+        side = _wgrad_stream() if ctx.needs_input_grad[1] else False
+        if side is not False:
+            # launch wgrad on the side stream FIRST so it co-runs with the
+            # dgrad enqueued on the main stream below (both are
+            # latency-bound; see _wgrad_stream)
+            main = torch.cuda.current_stream()
+            side.wait_stream(main)
+            with torch.cuda.stream(side):
+                gw = _conv_wgrad(e, gyp, x, w, stride, padding, dilation,
+                                 groups, cin, kout, cpad, kpad)
         if ctx.needs_input_grad[0]:
             wd = e.pad_channels(w, _pad8(cin)) if cpad else w
             if kpad:
@@ -138,18 +183,15 @@ class _HIPConv2d(torch.autograd.Function):
                 gx = _cl(gx[:, :cin])
             if fslot is not None and fbuf is None:
                 fslot["g"] = gx
-        if ctx.needs_input_grad[1]:
-            xw = e.pad_channels(x, _pad8(cin)) if cpad else x
-            gw = e.conv2d_wgrad(gyp, xw, w.shape[2], w.shape[3],
-                                stride[0], stride[1], padding[0],
-                                padding[1], dilation[0], dilation[1],
-                                groups)
-            if kpad:
-                gw = gw[:kout]
-            if cpad:
-                gw = gw[:, :cin]
-            if kpad or cpad:
-                gw = _cl(gw)
+        if side is not False:
+            main.wait_stream(side)
+            if not torch.cuda.is_current_stream_capturing():
+                gw.record_stream(main)
+                gyp.record_stream(side)
+                x.record_stream(side)
+        elif ctx.needs_input_grad[1]:
+            gw = _conv_wgrad(e, gyp, x, w, stride, padding, dilation, groups,
+                             cin, kout, cpad, kpad)
         if has_bias and ctx.needs_input_grad[2]:
             gb = gy.sum(dim=(0, 2, 3))
         return (gx, gw, gb, None, None, None, None, None, None, None, None,
