@@ -36,26 +36,6 @@ def _wgrad_stream():
     return _WGRAD_STREAM
 
 
-_WGRAD_PENDING = []  # input refs kept alive until the deferred join
-
-
-def _defer_wgrad_join():
-    """Without DDP nothing reads grads until optimizer.step, so the
-    main-stream join can happen ONCE at the end of the whole backward (an
-    autograd engine callback) instead of per conv — the side stream then
-    overlaps the entire backward chain, not just each op's own dgrad.
-    Under DDP the reducer consumes grads mid-backward, so join per op."""
-    import torch.distributed as dist
-    return not (dist.is_available() and dist.is_initialized()
-                and dist.get_world_size() > 1)
-
-
-def join_wgrad_stream():
-    if _WGRAD_STREAM and _WGRAD_PENDING:
-        torch.cuda.current_stream().wait_stream(_WGRAD_STREAM)
-        _WGRAD_PENDING.clear()
-
-
 def _cl(x):
     """Ensure channels_last physical layout for 4-D GPU tensors."""
     if x.dim() == 4:
@@ -204,20 +184,11 @@ class _HIPConv2d(torch.autograd.Function):
             if fslot is not None and fbuf is None:
                 fslot["g"] = gx
         if side is not False:
-            if _defer_wgrad_join():
-                # refs (not record_stream, which is a no-op under hipGraph
-                # capture) keep the side stream's inputs allocated until the
-                # end-of-backward join
-                if not _WGRAD_PENDING:
-                    torch.autograd.Variable._execution_engine.queue_callback(
-                        join_wgrad_stream)
-                _WGRAD_PENDING.append((gyp, x))
-            else:
-                main.wait_stream(side)
-                if not torch.cuda.is_current_stream_capturing():
-                    gw.record_stream(main)
-                    gyp.record_stream(side)
-                    x.record_stream(side)
+            main.wait_stream(side)
+            if not torch.cuda.is_current_stream_capturing():
+                gw.record_stream(main)
+                gyp.record_stream(side)
+                x.record_stream(side)
         elif ctx.needs_input_grad[1]:
             gw = _conv_wgrad(e, gyp, x, w, stride, padding, dilation, groups,
                              cin, kout, cpad, kpad)
