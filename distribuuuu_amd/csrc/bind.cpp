@@ -72,7 +72,18 @@ std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor gy, at::Tensor w,
 std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
     at::Tensor gy, at::Tensor w, int64_t H, int64_t W, int64_t sh, int64_t sw,
     int64_t ph, int64_t pw, int64_t dh, int64_t dw, int64_t groups,
-    at::Tensor into);
+    at::Tensor into, c10::optional<at::Tensor> pre_opt, int64_t pre_kind);
+
+std::vector<at::Tensor> conv2d_dgrad_prep(at::Tensor w, int64_t Kt0,
+                                          int64_t sh, int64_t sw, int64_t ph,
+                                          int64_t pw, int64_t dh, int64_t dw,
+                                          int64_t groups);
+
+at::Tensor conv2d_dgrad_pre(at::Tensor gy, at::Tensor w, int64_t H,
+                            int64_t W, int64_t sh, int64_t sw, int64_t ph,
+                            int64_t pw, int64_t dh, int64_t dw,
+                            int64_t groups, at::Tensor pre,
+                            int64_t pre_kind);
 at::Tensor gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor conv2d_fwd_v2(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                          int64_t ph, int64_t pw, int64_t dh, int64_t dw);
@@ -154,6 +165,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_dgrad_bn", &conv2d_dgrad_bn);
   m.def("conv2d_dgrad_acc", &conv2d_dgrad_acc);
+  m.def("conv2d_dgrad_prep", &conv2d_dgrad_prep);
+  m.def("conv2d_dgrad_pre", &conv2d_dgrad_pre);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("gemm_nt", &gemm_nt);
   m.def("conv2d_fwd_v2", &conv2d_fwd_v2);

@@ -1302,7 +1302,9 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
                                     int64_t ph, int64_t pw, int64_t dh,
                                     int64_t dw, int64_t groups,
                                     at::Tensor* part_out,
-                                    const BnBwdEmit* bemit) {
+                                    const BnBwdEmit* bemit,
+                                    const at::Tensor* pre = nullptr,
+                                    int64_t pre_kind = -1) {
   CHECK_GPU(gy);
   check_nhwc(gy, "gy");
   const int N = gy.size(0);
@@ -1327,13 +1329,17 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
         dKg0 >= v2mink2 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
         gy.scalar_type() == at::kBFloat16) {
       const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
-      auto wsp = weight_flip_t_span(w, groups, SPAN64);
+      auto wsp = (pre != nullptr && pre_kind == 1)
+                     ? *pre
+                     : weight_flip_t_span(w, groups, SPAN64);
       return conv2d_fwd_v2_flat(gy, wsp, Ct, dCg0, R, S, 1, 1,
                                 dh * (R - 1) - ph, dw * (S - 1) - pw, dh, dw,
                                 groups, part_out, bemit);
     }
   }
-  auto wt = weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
+  auto wt = (pre != nullptr && pre_kind == 0)
+                ? *pre
+                : weight_flip_t(w, groups);  // [Ct, Kg, R, S] cl
   if (R == 1 && S == 1 && (sh > 1 || sw > 1) && ph == 0 && pw == 0) {
     // strided-output GEMM: gx[ho*sh, wo*sw] = gy[ho, wo] @ w^T, rest zero.
     // The zeroed gap positions contribute g = 0 to the BN-backward sums, so
@@ -1406,6 +1412,51 @@ at::Tensor conv2d_dgrad(at::Tensor gy, at::Tensor w, int64_t H, int64_t W,
                            nullptr, nullptr);
 }
 
+// Forward-time dgrad weight-transform precompute (launched on a side
+// stream during the forward pass so backward's critical path skips the
+// flip kernels). Returns {transformed_weight, kind}: kind 1 = span-padded
+// flat flip for the same-size v2 ring, kind 0 = plain flipT. The gate
+// mirrors conv2d_dgrad_impl's v2 dispatch exactly.
+std::vector<at::Tensor> conv2d_dgrad_prep(at::Tensor w, int64_t Kt0,
+                                          int64_t sh, int64_t sw, int64_t ph,
+                                          int64_t pw, int64_t dh, int64_t dw,
+                                          int64_t groups) {
+  const int Cg = w.size(1), R = w.size(2), S = w.size(3);
+  const int Ct = Cg * (int)groups;
+  const int dKg0 = Ct / (int)groups;
+  const int dCg0 = (int)(Kt0 / groups);
+  static const int v2mink4 = []() {
+    const char* e = getenv("DISTRIBUUUU_V2_MINK");
+    return e ? atoi(e) : 192;
+  }();
+  static const bool v2off4 = []() {
+    const char* e = getenv("DISTRIBUUUU_CONV_V2");
+    return e && e[0] == '0';
+  }();
+  int64_t kind = 0;
+  at::Tensor out;
+  if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
+      dw * (S - 1) == 2 * pw && Kt0 % (8 * groups) == 0 && !v2off4 &&
+      dKg0 >= v2mink4 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
+      w.scalar_type() == at::kBFloat16) {
+    const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
+    out = weight_flip_t_span(w, groups, SPAN64);
+    kind = 1;
+  } else {
+    out = weight_flip_t(w, groups);
+  }
+  return {out, at::scalar_tensor(kind)};
+}
+
+at::Tensor conv2d_dgrad_pre(at::Tensor gy, at::Tensor w, int64_t H,
+                            int64_t W, int64_t sh, int64_t sw, int64_t ph,
+                            int64_t pw, int64_t dh, int64_t dw,
+                            int64_t groups, at::Tensor pre,
+                            int64_t pre_kind) {
+  return conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
+                           nullptr, nullptr, &pre, pre_kind);
+}
+
 // dgrad + BN-backward stat partials (docs/DESIGN_bn_conv_fusion.md, the
 // dgrad-side analogue of conv2d_fwd_bn): bnx/scale/shift/act describe the
 // BatchNorm whose gy this dgrad produces. Returns {gx, part}; part is an
@@ -1426,6 +1477,7 @@ std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor gy, at::Tensor w,
   at::Tensor part;
   auto gx = conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
                               &part, &em);
+  // (pre-transform not threaded here: the EMODE-2 path is opt-in/off)
   if (!part.defined()) part = at::empty({0, 0}, bnscale.options());
   return {gx, part};
 }
@@ -1440,7 +1492,9 @@ std::vector<at::Tensor> conv2d_dgrad_bn(at::Tensor gy, at::Tensor w,
 std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
     at::Tensor gy, at::Tensor w, int64_t H, int64_t W, int64_t sh, int64_t sw,
     int64_t ph, int64_t pw, int64_t dh, int64_t dw, int64_t groups,
-    at::Tensor into) {
+    at::Tensor into, c10::optional<at::Tensor> pre_opt = c10::nullopt,
+    int64_t pre_kind = -1) {
+  const at::Tensor* pre = pre_opt.has_value() ? &*pre_opt : nullptr;
   CHECK_GPU(gy);
   const int N = gy.size(0);
   const int Cg = w.size(1), R = w.size(2), S = w.size(3);
@@ -1465,14 +1519,17 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
     if (same_size && !v2off3 && Kt0 % 8 == 0 && dKg0 >= v2mink3 &&
         (int64_t)R * S * dCg0 >= 512) {
       const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
-      auto wsp = weight_flip_t_span(w, 1, SPAN64);
+      auto wsp = (pre != nullptr && pre_kind == 1)
+                     ? *pre
+                     : weight_flip_t_span(w, 1, SPAN64);
       conv2d_fwd_v2_flat(gy, wsp, Ct, dCg0, R, S, 1, 1, dh * (R - 1) - ph,
                          dw * (S - 1) - pw, dh, dw, 1, nullptr, nullptr,
                          &into);
       return {into, 1};
     }
     if (same_size) {
-      auto wt = weight_flip_t(w, 1);
+      auto wt = (pre != nullptr && pre_kind == 0) ? *pre
+                                                   : weight_flip_t(w, 1);
       conv2d_fwd_into(gy, wt, into, H, W, 1, 1, dh * (R - 1) - ph,
                       dw * (S - 1) - pw, dh, dw, 1, 1, 1, 0, 0, nullptr,
                       nullptr, true);
@@ -1482,7 +1539,8 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
       // strided 1x1 proj: scatter-accumulate at (ho*sh, wo*sw); the gap
       // positions receive no main-branch gradient, so `into` already holds
       // their final values.
-      auto wt = weight_flip_t(w, 1);
+      auto wt = (pre != nullptr && pre_kind == 0) ? *pre
+                                                   : weight_flip_t(w, 1);
       conv2d_fwd_into(gy, wt, into, gy.size(2), gy.size(3), 1, 1, 0, 0, 1, 1,
                       1, sh, sw, 0, 0, nullptr, nullptr, true);
       return {into, 1};
@@ -1492,7 +1550,8 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
       // parity decomposition (see conv2d_dgrad_impl): the four parity
       // classes tile gx disjointly, so each sub-conv scatter-accumulates
       // its own positions exactly once.
-      auto wt = weight_flip_t(w, 1);
+      auto wt = (pre != nullptr && pre_kind == 0) ? *pre
+                                                   : weight_flip_t(w, 1);
       for (int h0 = 0; h0 < 2; ++h0) {
         const int64_t hu = (H - h0 + 1) >> 1;
         auto wr = (h0 == 0) ? wt.slice(2, 1, 2) : wt.slice(2, 0, 3, 2);
@@ -1508,6 +1567,6 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
     }
   }
   auto gx = conv2d_dgrad_impl(gy, w, H, W, sh, sw, ph, pw, dh, dw, groups,
-                              nullptr, nullptr);
+                              nullptr, nullptr, pre, pre_kind);
   return {gx, 0};
 }

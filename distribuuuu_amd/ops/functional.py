@@ -115,6 +115,24 @@ class _HIPConv2d(torch.autograd.Function):
             y = _cl(y[:, :kout])
         if bias is not None:
             y = y + bias.reshape(1, -1, 1, 1)
+        wt_pre = wt_kind = wt_ev = None
+        if (torch.is_grad_enabled() and x.requires_grad and not cpad
+                and not kpad):
+            ts = _wgrad_stream()
+            if ts is not False:
+                # precompute the dgrad weight transform (flipT / span-flip)
+                # on the side stream NOW, overlapping forward compute —
+                # backward's critical path skips those kernels entirely
+                cur = torch.cuda.current_stream()
+                ts.wait_stream(cur)
+                with torch.cuda.stream(ts):
+                    wt_pre, kt = e.conv2d_dgrad_prep(
+                        w, kout, stride[0], stride[1], padding[0],
+                        padding[1], dilation[0], dilation[1], groups)
+                wt_kind = int(kt.item())
+                wt_ev = torch.cuda.Event()
+                wt_ev.record(ts)
+        ctx.wt_pre = (wt_pre, wt_kind, wt_ev)
         if bnx is not None and not cpad:
             ctx.save_for_backward(x, w, bnx, bnscale, bnshift)
         else:
@@ -151,6 +169,11 @@ class _HIPConv2d(torch.autograd.Function):
             wd = e.pad_channels(w, _pad8(cin)) if cpad else w
             if kpad:
                 wd = _pad_k(wd, _pad8(kout))
+            pre, pkind, pev = ctx.wt_pre
+            if pre is not None:
+                torch.cuda.current_stream().wait_event(pev)
+                if not torch.cuda.is_current_stream_capturing():
+                    pre.record_stream(torch.cuda.current_stream())
             fslot = ctx.fork_slot
             fbuf = fslot.get("g") if fslot is not None else None
             if fbuf is not None and not cpad and fbuf.dtype == gyp.dtype:
@@ -161,7 +184,7 @@ class _HIPConv2d(torch.autograd.Function):
                 gx, acc_done = e.conv2d_dgrad_acc(
                     gyp, wd, x.shape[2], x.shape[3], stride[0], stride[1],
                     padding[0], padding[1], dilation[0], dilation[1], groups,
-                    fbuf)
+                    fbuf, pre, pkind if pre is not None else -1)
             elif ctx.bnslot is not None and len(saved) == 5:
                 # dgrad + BN-backward stats from the same epilogue: the
                 # consuming BN reads the partials IF this gx arrives there
@@ -174,6 +197,11 @@ class _HIPConv2d(torch.autograd.Function):
                 if bpart.numel():
                     ctx.bnslot["parts"].append(bpart)
                     ctx.bnslot["gx"] = gx
+            elif pre is not None:
+                gx = e.conv2d_dgrad_pre(gyp, wd, x.shape[2], x.shape[3],
+                                        stride[0], stride[1], padding[0],
+                                        padding[1], dilation[0], dilation[1],
+                                        groups, pre, pkind)
             else:
                 gx = e.conv2d_dgrad(gyp, wd, x.shape[2], x.shape[3],
                                     stride[0], stride[1], padding[0],
