@@ -443,7 +443,7 @@ class TestDgradAcc:
         gx_ref = e.conv2d_dgrad(gy, w, H, W, stride, stride, pad, pad, 1, 1,
                                 groups)
         out, flag = e.conv2d_dgrad_acc(gy, w, H, W, stride, stride, pad, pad,
-                                       1, 1, groups, into)
+                                       1, 1, groups, into, None, -1)
         assert bool(flag) == want_acc
         if want_acc:
             assert out.data_ptr() == into.data_ptr()
