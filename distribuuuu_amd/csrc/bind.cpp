@@ -34,7 +34,11 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
                                at::Tensor scale, at::Tensor shift,
-                               int64_t act, bool training, bool need_gres);
+                               int64_t act, bool training, bool need_gres,
+                               c10::optional<at::Tensor> mask);
+std::vector<at::Tensor> bn_apply_act_mask(at::Tensor x, at::Tensor scale,
+                                          at::Tensor shift, int64_t act,
+                                          at::Tensor res);
 // pool.hip
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
                                     int64_t P);
@@ -138,6 +142,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("training") = true, py::arg("part") = py::none());
   m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
+  m.def("bn_apply_act_mask", &bn_apply_act_mask);
   m.def("bn_bwd_stats", &bn_bwd_stats, py::arg("gy"), py::arg("x"),
         py::arg("res"), py::arg("scale"), py::arg("shift"),
         py::arg("act"));
@@ -149,7 +154,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd", &bn_bwd, py::arg("gy"), py::arg("x"),
         py::arg("res"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
         py::arg("scale"), py::arg("shift"), py::arg("act"),
-        py::arg("training"), py::arg("need_gres"));
+        py::arg("training"), py::arg("need_gres"),
+        py::arg("mask") = py::none());
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("gap_fwd", &gap_fwd);

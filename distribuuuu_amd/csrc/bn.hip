@@ -242,11 +242,16 @@ __global__ void bn_eval_prep_kernel(const float* __restrict__ running_mean,
 // pointer increments (no per-iteration modulo or table gathers — the
 // previous channel-incrementing grid-stride form re-gathered the float4
 // tables every pack).
-template <typename T, bool HAS_RES>
+// EMITMASK (HAS_RES relu only): one act'(z) bit per element, a byte per
+// V-pack at [row * cpacks + cp] — the backward kernels then skip the res
+// stream AND the z recompute (docs/ARCHITECTURE.md round-2 status).
+template <typename T, bool HAS_RES, bool EMITMASK = false>
 __global__ __launch_bounds__(256, 2) void bn_apply_kernel(
     const T* __restrict__ x, const float* __restrict__ scale,
     const float* __restrict__ shift, const T* __restrict__ res,
-    T* __restrict__ y, int64_t rows, int C, int64_t rows_per_block, int act) {
+    T* __restrict__ y, int64_t rows, int C, int64_t rows_per_block, int act,
+    unsigned char* __restrict__ mask = nullptr) {
+  static_assert(!(EMITMASK && !HAS_RES), "mask emission is for residual BNs");
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   const int cpacks = C / V;
@@ -273,6 +278,7 @@ __global__ __launch_bounds__(256, 2) void bn_apply_kernel(
     const P* xq = xp + row * cpacks + cp;
     const P* rq = rp + row * cpacks + cp;
     P* yq = yp + row * cpacks + cp;
+    unsigned char* mq = EMITMASK ? mask + row * cpacks + cp : nullptr;
     for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
       P px4[4], pr4[4];
 #pragma unroll
@@ -284,28 +290,38 @@ __global__ __launch_bounds__(256, 2) void bn_apply_kernel(
       rq += 4 * rstep;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
+        unsigned int mb = 0;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float z = to_f32(px4[u].v[j]) * sc[j] + sh[j];
           if (HAS_RES) z += to_f32(pr4[u].v[j]);
+          if (EMITMASK && z > 0.f) mb |= 1u << j;
           px4[u].v[j] = from_f32<T>(act_apply(z, act));
         }
         yq[u * rstep] = px4[u];
+        if (EMITMASK) mq[u * rstep] = (unsigned char)mb;
       }
       yq += 4 * rstep;
+      if (EMITMASK) mq += 4 * rstep;
     }
     for (; row < row1; row += nrl) {
       P px = xq[0], pr;
       if (HAS_RES) pr = rq[0];
       xq += rstep;
       rq += rstep;
+      unsigned int mb = 0;
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         float z = to_f32(px.v[j]) * sc[j] + sh[j];
         if (HAS_RES) z += to_f32(pr.v[j]);
+        if (EMITMASK && z > 0.f) mb |= 1u << j;
         px.v[j] = from_f32<T>(act_apply(z, act));
       }
       yq[0] = px;
+      if (EMITMASK) {
+        mq[0] = (unsigned char)mb;
+        mq += rstep;
+      }
       yq += rstep;
     }
   }
@@ -314,12 +330,15 @@ __global__ __launch_bounds__(256, 2) void bn_apply_kernel(
 // ---- bwd stage 1: RAW per-channel sums sum(g), sum(g*x) --------------------
 // g = gy * act'; act' recomputes z = scale*x+shift(+res) for every act, so
 // the y stream is never read. No mean/rstd/gamma gathers on the hot path.
-template <typename T, bool HAS_RES, int ACT>
+template <typename T, bool HAS_RES, int ACT, bool MASKED = false>
 __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, float* __restrict__ part, int64_t rows,
-    int C, int64_t rows_per_block) {
+    int C, int64_t rows_per_block,
+    const unsigned char* __restrict__ mask = nullptr) {
+  static_assert(!(MASKED && !(HAS_RES && ACT == 1)),
+                "mask path covers residual relu BNs only");
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
@@ -353,24 +372,32 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
       const P* gq = gp + row * cpacks + cp;
       const P* xq = xp + row * cpacks + cp;
       const P* rq = rp + row * cpacks + cp;
+      const unsigned char* mq = MASKED ? mask + row * cpacks + cp : nullptr;
       for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
         P pg4[4], px4[4], pr4[4];
+        unsigned char mb4[4];
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
           pg4[u] = gq[u * rstep];
           px4[u] = xq[u * rstep];
-          if (HAS_RES && ACT != 0) pr4[u] = rq[u * rstep];
+          if (MASKED)
+            mb4[u] = mq[u * rstep];
+          else if (HAS_RES && ACT != 0)
+            pr4[u] = rq[u * rstep];
         }
         gq += 4 * rstep;
         xq += 4 * rstep;
         rq += 4 * rstep;
+        if (MASKED) mq += 4 * rstep;
 #pragma unroll
         for (int u = 0; u < 4; ++u)
 #pragma unroll
           for (int j = 0; j < V; ++j) {
             float xv = to_f32(px4[u].v[j]);
             float g = to_f32(pg4[u].v[j]);
-            if (ACT != 0) {
+            if (MASKED) {
+              g = (mb4[u] >> j) & 1 ? g : 0.f;
+            } else if (ACT != 0) {
               float z = xv * sc[j] + sh[j];
               if (HAS_RES) z += to_f32(pr4[u].v[j]);
               g *= act_grad(z, ACT);
@@ -381,15 +408,22 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_kernel(
       }
       for (; row < row1; row += nrl) {
         P pg = gq[0], px = xq[0], prr;
-        if (HAS_RES && ACT != 0) prr = rq[0];
+        unsigned char mb = 0;
+        if (MASKED)
+          mb = mq[0];
+        else if (HAS_RES && ACT != 0)
+          prr = rq[0];
         gq += rstep;
         xq += rstep;
         rq += rstep;
+        if (MASKED) mq += rstep;
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float xv = to_f32(px.v[j]);
           float g = to_f32(pg.v[j]);
-          if (ACT != 0) {
+          if (MASKED) {
+            g = (mb >> j) & 1 ? g : 0.f;
+          } else if (ACT != 0) {
             float z = xv * sc[j] + sh[j];
             if (HAS_RES) z += to_f32(prr.v[j]);
             g *= act_grad(z, ACT);
@@ -491,14 +525,17 @@ __global__ void bn_bwd_finalize_kernel(
 // channel pack, so the 3-5 per-channel coefficient tables are loaded ONCE
 // per column instead of per grid-stride iteration — the flat grid-stride
 // form re-gathered them every pack and cost ~0.5 ms/step on ResNet-50.
-template <typename T, bool HAS_RES, int ACT>
+template <typename T, bool HAS_RES, int ACT, bool MASKED = false>
 __global__ __launch_bounds__(256, 2) void bn_bwd_dx_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ res, const float* __restrict__ scale,
     const float* __restrict__ shift, const float* __restrict__ P1c,
     const float* __restrict__ P2c, const float* __restrict__ P3c,
     T* __restrict__ gx, T* __restrict__ gres, int64_t rows, int C,
-    int64_t rows_per_block) {
+    int64_t rows_per_block,
+    const unsigned char* __restrict__ mask = nullptr) {
+  static_assert(!(MASKED && !(HAS_RES && ACT == 1)),
+                "mask path covers residual relu BNs only");
   constexpr int V = 16 / sizeof(T);
   using P = Pack<T, V>;
   const int cpacks = C / V;
@@ -533,19 +570,25 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_dx_kernel(
     const P* gq = gp + row * cpacks + cp;
     const P* xq = xp + row * cpacks + cp;
     const P* rq = rp + row * cpacks + cp;
+    const unsigned char* mq = MASKED ? mask + row * cpacks + cp : nullptr;
     P* oq = oxp + row * cpacks + cp;
     P* orq = orp + row * cpacks + cp;
     for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
       P pg4[4], px4[4], pr4[4];
+      unsigned char mb4[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         pg4[u] = gq[u * rstep];
         px4[u] = xq[u * rstep];
-        if (HAS_RES && ACT != 0) pr4[u] = rq[u * rstep];
+        if (MASKED)
+          mb4[u] = mq[u * rstep];
+        else if (HAS_RES && ACT != 0)
+          pr4[u] = rq[u * rstep];
       }
       gq += 4 * rstep;
       xq += 4 * rstep;
       rq += 4 * rstep;
+      if (MASKED) mq += 4 * rstep;
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         P ox, orr;
@@ -553,7 +596,9 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_dx_kernel(
         for (int j = 0; j < V; ++j) {
           float xv = to_f32(px4[u].v[j]);
           float g = to_f32(pg4[u].v[j]);
-          if (ACT != 0) {
+          if (MASKED) {
+            g = (mb4[u] >> j) & 1 ? g : 0.f;
+          } else if (ACT != 0) {
             float z = xv * sc[j] + sh[j];
             if (HAS_RES) z += to_f32(pr4[u].v[j]);
             g *= act_grad(z, ACT);
@@ -569,16 +614,23 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_dx_kernel(
     }
     for (; row < row1; row += nrl) {
       P pg = gq[0], px = xq[0], pr;
-      if (HAS_RES && ACT != 0) pr = rq[0];
+      unsigned char mb = 0;
+      if (MASKED)
+        mb = mq[0];
+      else if (HAS_RES && ACT != 0)
+        pr = rq[0];
       gq += rstep;
       xq += rstep;
       rq += rstep;
+      if (MASKED) mq += rstep;
       P ox, orr;
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         float xv = to_f32(px.v[j]);
         float g = to_f32(pg.v[j]);
-        if (ACT != 0) {
+        if (MASKED) {
+          g = (mb >> j) & 1 ? g : 0.f;
+        } else if (ACT != 0) {
           float z = xv * sc[j] + sh[j];
           if (HAS_RES) z += to_f32(pr.v[j]);
           g *= act_grad(z, ACT);
@@ -613,12 +665,18 @@ void launch_bwd_reduce(const scalar_t* gy, const scalar_t* x,
                        const scalar_t* resp,
                        const float* scale, const float* shift, float* part,
                        int64_t rows, int C, int64_t rpb, int act, int rgrid,
-                       hipStream_t stream) {
+                       hipStream_t stream,
+                       const unsigned char* maskp = nullptr) {
 #define BR_CASE(HR, A)                                                      \
   hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, HR, A>), dim3(rgrid),  \
                      dim3(256), 0, stream, gy, x, resp, scale, shift,       \
                      part, rows, C, rpb)
-  if (resp != nullptr) {
+  if (maskp != nullptr) {
+    TORCH_CHECK(act == 1 && resp != nullptr, "mask path is residual relu");
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<scalar_t, true, 1, true>),
+                       dim3(rgrid), dim3(256), 0, stream, gy, x, resp, scale,
+                       shift, part, rows, C, rpb, maskp);
+  } else if (resp != nullptr) {
     switch (act) {
       case 0: BR_CASE(true, 0); break;
       case 1: BR_CASE(true, 1); break;
@@ -643,7 +701,8 @@ void launch_dx(const at::Tensor& gy, const at::Tensor& x,
                const at::Tensor& shift, const at::Tensor& P1,
                const at::Tensor& P2, const at::Tensor& P3, at::Tensor& gx,
                scalar_t* gresp, int64_t rows, int C, int64_t rpb, int act,
-               int rgrid, hipStream_t stream) {
+               int rgrid, hipStream_t stream,
+               const unsigned char* maskp = nullptr) {
 #define DX_CASE(HR, A)                                                       \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, HR, A>), dim3(rgrid),       \
                      dim3(256), 0, stream, (const scalar_t*)gy.data_ptr(),   \
@@ -652,7 +711,17 @@ void launch_dx(const at::Tensor& gy, const at::Tensor& x,
                      P1.data_ptr<float>(), P2.data_ptr<float>(),             \
                      P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,  \
                      rows, C, rpb)
-  if (resp != nullptr) {
+  if (maskp != nullptr) {
+    TORCH_CHECK(act == 1 && resp != nullptr, "mask path is residual relu");
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, true, 1, true>),
+                       dim3(rgrid), dim3(256), 0, stream,
+                       (const scalar_t*)gy.data_ptr(),
+                       (const scalar_t*)x.data_ptr(), resp,
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       P1.data_ptr<float>(), P2.data_ptr<float>(),
+                       P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), gresp,
+                       rows, C, rpb, maskp);
+  } else if (resp != nullptr) {
     switch (act) {
       case 0: DX_CASE(true, 0); break;
       case 1: DX_CASE(true, 1); break;
@@ -847,6 +916,38 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
   return y;
 }
 
+// Residual relu BN apply that ALSO emits the act' bitmask (one byte per
+// V-pack) so the backward kernels can drop the res stream and the z
+// recompute. Returns {y, mask}.
+std::vector<at::Tensor> bn_apply_act_mask(at::Tensor x, at::Tensor scale,
+                                          at::Tensor shift, int64_t act,
+                                          at::Tensor res) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  TORCH_CHECK(act == 1, "mask emission: relu only");
+  const int C = x.size(1);
+  auto y = at::empty_like(x);
+  at::Tensor mask;
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_apply_act_mask", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    const int64_t rows = x.numel() / C;
+    const int cpacks = C / V;
+    mask = at::empty({rows * cpacks}, x.options().dtype(at::kByte));
+    const int nrl = std::max(256 / cpacks, 1);
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
+    hipLaunchKernelGGL((bn_apply_kernel<scalar_t, true, true>), dim3(rgrid),
+                       dim3(256), 0, cur_stream(),
+                       (const scalar_t*)x.data_ptr(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       (const scalar_t*)res.data_ptr(),
+                       (scalar_t*)y.data_ptr(), rows, C, rpb, (int)act,
+                       (unsigned char*)mask.data_ptr());
+  });
+  return {y, mask};
+}
+
 // Local raw grad-stat sums [2C] = [sum(g), sum(g*x)] (stage 1 + stage 2);
 // used standalone by SyncBN (which all-reduces the result across ranks).
 at::Tensor bn_bwd_stats(at::Tensor gy, at::Tensor x,
@@ -935,7 +1036,8 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                                c10::optional<at::Tensor> res, at::Tensor mean,
                                at::Tensor rstd, at::Tensor gamma,
                                at::Tensor scale, at::Tensor shift,
-                               int64_t act, bool training, bool need_gres) {
+                               int64_t act, bool training, bool need_gres,
+                               c10::optional<at::Tensor> mask) {
   CHECK_GPU(gy);
   check_nhwc(gy, "gy");
   const int C = x.size(1);
@@ -962,11 +1064,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
     auto part = at::empty({rgrid, 2 * C}, fopts);
     const scalar_t* resp =
         has_res ? (const scalar_t*)res->data_ptr() : nullptr;
+    const unsigned char* maskp =
+        mask.has_value() ? (const unsigned char*)mask->data_ptr() : nullptr;
     launch_bwd_reduce<scalar_t>(
         (const scalar_t*)gy.data_ptr(), (const scalar_t*)x.data_ptr(),
         resp, scale.data_ptr<float>(),
         shift.data_ptr<float>(), part.data_ptr<float>(), rows, C, rpb,
-        (int)act, rgrid, stream);
+        (int)act, rgrid, stream, maskp);
     const float inv_cnt = 1.f / (float)rows;
     hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
                        dim3(256), 0, stream, part.data_ptr<float>(), rgrid,
@@ -977,7 +1081,7 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
                        training ? 1 : 0);
     scalar_t* gresp = has_res ? (scalar_t*)gres.data_ptr() : nullptr;
     launch_dx<scalar_t>(gy, x, resp, scale, shift, P1, P2, P3, gx, gresp,
-                        rows, C, rpb, (int)act, rgrid, stream);
+                        rows, C, rpb, (int)act, rgrid, stream, maskp);
   });
   return {gx, gw, gb, gres};
 }

@@ -354,9 +354,17 @@ class _HIPBatchNormAct(torch.autograd.Function):
                 act_id, residual, slot, fork_slot):
         e = ext()
         res = _cl(residual) if residual is not None else None
-        y = e.bn_apply_act(x, scale, shift, act_id, res)
+        mask = None
+        if (res is not None and act_id == 1 and training
+                and torch.is_grad_enabled()):
+            # emit the act' bitmask (1 bit/elem): backward then skips the
+            # res stream and the z recompute on the big residual BNs
+            y, mask = e.bn_apply_act_mask(x, scale, shift, act_id, res)
+        else:
+            y = e.bn_apply_act(x, scale, shift, act_id, res)
         ctx.save_for_backward(x, weight, scale, shift, mean, rstd,
-                              res if res is not None else x.new_empty(0))
+                              res if res is not None else x.new_empty(0),
+                              mask if mask is not None else x.new_empty(0))
         ctx.act_id = act_id
         ctx.training = training
         ctx.has_res = residual is not None
@@ -366,7 +374,10 @@ class _HIPBatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy):
-        x, weight, scale, shift, mean, rstd, res = ctx.saved_tensors
+        (x, weight, scale, shift, mean, rstd, res,
+         mask) = ctx.saved_tensors
+        if mask.numel() == 0:
+            mask = None
         e = ext()
         gy = _cl(gy)
         gamma = weight.float().contiguous()
@@ -386,7 +397,8 @@ class _HIPBatchNormAct(torch.autograd.Function):
         else:
             gx, gw, gb, gres = e.bn_bwd(
                 gy, x, res if ctx.has_res else None, mean, rstd, gamma,
-                scale, shift, ctx.act_id, ctx.training, ctx.has_res)
+                scale, shift, ctx.act_id, ctx.training, ctx.has_res,
+                mask=mask)
         if (ctx.has_res and ctx.fork_slot is not None
                 and "g" not in ctx.fork_slot):
             # the shortcut branch of a residual fork: deposit gres so the
