@@ -601,3 +601,29 @@ def test_hipsgd_table_rebuild_frequency():
         assert builds["n"] <= 1, builds["n"]
     finally:
         O.HIPSGD._build_table = orig
+
+
+@pytest.mark.gpu
+def test_bn_residual_mask_matches_recompute():
+    """The forward-emitted relu bitmask backward must match the z-recompute
+    backward bit-for-bit (same z arithmetic, mask bit == (z > 0))."""
+    e = _ext()
+    cl = torch.channels_last
+    torch.manual_seed(3)
+    x = torch.randn(8, 64, 28, 28, device="cuda",
+                    dtype=torch.bfloat16).contiguous(memory_format=cl)
+    res = torch.randn_like(x).contiguous(memory_format=cl)
+    gy = torch.randn_like(x).contiguous(memory_format=cl)
+    gamma = torch.rand(64, device="cuda") + 0.5
+    beta = torch.randn(64, device="cuda")
+    mean, rstd, scale, shift = e.bn_stats(x, gamma, beta, None, None,
+                                          0.1, 1e-5, True, None)
+    y0 = e.bn_apply_act(x, scale, shift, 1, res)
+    y1, mask = e.bn_apply_act_mask(x, scale, shift, 1, res)
+    assert torch.equal(y0, y1)
+    ref = e.bn_bwd(gy, x, res, mean, rstd, gamma, scale, shift, 1, True,
+                   True)
+    got = e.bn_bwd(gy, x, res, mean, rstd, gamma, scale, shift, 1, True,
+                   True, mask=mask)
+    for a, b in zip(ref, got):
+        assert torch.equal(a, b)
