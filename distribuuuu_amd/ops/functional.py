@@ -356,7 +356,8 @@ class _HIPBatchNormAct(torch.autograd.Function):
         res = _cl(residual) if residual is not None else None
         mask = None
         if (res is not None and act_id == 1 and training
-                and torch.is_grad_enabled()):
+                and torch.is_grad_enabled()
+                and os.environ.get("DISTRIBUUUU_BN_MASK", "1") != "0"):
             # emit the act' bitmask (1 bit/elem): backward then skips the
             # res stream and the z recompute on the big residual BNs
             y, mask = e.bn_apply_act_mask(x, scale, shift, act_id, res)
