@@ -111,6 +111,65 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ gy,
   }
 }
 
+// ---- specialized stem maxpool bwd: K=3 S=2 P=1 ------------------------------
+// Each input element receives from at most 2x2 output windows; all four
+// idx/gy loads are issued up-front (the generic nested runtime loop
+// serialized them: 301 us for ~1 GB of traffic). Row-based grid kills the
+// 64-bit div/mod chain of the flat form.
+template <typename T>
+__global__ void maxpool_bwd_k3s2_kernel(const T* __restrict__ gy,
+                                        const unsigned char* __restrict__ idx,
+                                        T* __restrict__ gx, int N, int H,
+                                        int W, int C, int Ho, int Wo) {
+  constexpr int V = 16 / sizeof(T);
+  using Pk = Pack<T, V>;
+  const int cpacks = C / V;
+  const int rowpacks = W * cpacks;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= rowpacks) return;
+  const int row = blockIdx.y;  // n * H + h
+  const int n = row / H, h = row - n * H;
+  const int w = i / cpacks;
+  const int cp = i - w * cpacks;
+  const int ho0 = max(0, h >> 1), ho1 = min(Ho - 1, (h + 1) >> 1);
+  const int wo0 = max(0, w >> 1), wo1 = min(Wo - 1, (w + 1) >> 1);
+  const bool h2 = ho1 > ho0, w2 = wo1 > wo0;
+  const int64_t base =
+      (((int64_t)n * Ho + ho0) * Wo + wo0) * C + (int64_t)cp * V;
+  const int64_t dW = (int64_t)C * Wo;  // one output row in elements
+  Pk pg[4];
+  __align__(8) unsigned char bi[4][V];
+  auto ld = [&](int k, int64_t o) {
+    pg[k] = *reinterpret_cast<const Pk*>(gy + o);
+    if (V == 8)
+      *reinterpret_cast<uint2*>(&bi[k][0]) =
+          *reinterpret_cast<const uint2*>(idx + o);
+    else
+      *reinterpret_cast<unsigned int*>(&bi[k][0]) =
+          *reinterpret_cast<const unsigned int*>(idx + o);
+  };
+  ld(0, base);
+  if (w2) ld(1, base + C);
+  if (h2) ld(2, base + dW);
+  if (h2 && w2) ld(3, base + dW + C);
+  float acc[V] = {};
+  auto add = [&](int k, int ho, int wo) {
+    const int want = (h - (ho * 2 - 1)) * 3 + (w - (wo * 2 - 1));
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      if (bi[k][j] == want) acc[j] += to_f32(pg[k].v[j]);
+  };
+  add(0, ho0, wo0);
+  if (w2) add(1, ho0, wo1);
+  if (h2) add(2, ho1, wo0);
+  if (h2 && w2) add(3, ho1, wo1);
+  Pk out;
+#pragma unroll
+  for (int j = 0; j < V; ++j) out.v[j] = from_f32<T>(acc[j]);
+  *reinterpret_cast<Pk*>(gx + ((int64_t)row * W + w) * C +
+                         (int64_t)cp * V) = out;
+}
+
 // ---- global average pool: block per (n, 32 c-packs); 8 row lanes ----------
 // (the original thread-per-channel form walked H*W scalar loads serially:
 // 180 us on RegNetY's SE pools)
@@ -281,6 +340,15 @@ at::Tensor maxpool_bwd(at::Tensor gy, at::Tensor idx, int64_t H, int64_t W,
                       gy.options().memory_format(at::MemoryFormat::ChannelsLast));
   DISPATCH_FLOAT_AND_BF16(gy.scalar_type(), "maxpool_bwd", [&] {
     constexpr int V = 16 / sizeof(scalar_t);
+    if (K == 3 && S == 2 && P == 1 && C % V == 0) {
+      const int rowpacks = (int)W * (C / V);
+      hipLaunchKernelGGL((maxpool_bwd_k3s2_kernel<scalar_t>),
+                         dim3((rowpacks + 255) / 256, N * H), dim3(256), 0,
+                         cur_stream(), (const scalar_t*)gy.data_ptr(),
+                         idx.data_ptr<unsigned char>(),
+                         (scalar_t*)gx.data_ptr(), N, H, W, C, Ho, Wo);
+      return;
+    }
     int64_t total = (int64_t)N * H * W * (C / V);
     hipLaunchKernelGGL((maxpool_bwd_kernel<scalar_t>),
                        dim3(grid_1d(total, 256)), dim3(256), 0, cur_stream(),

@@ -245,10 +245,24 @@ class _HIPDepthwiseConv2d(torch.autograd.Function):
         gy = _cl(gy)
         e = ext()
         gx = gw = gb = None
+        side = _wgrad_stream() if ctx.needs_input_grad[1] else False
+        if side is not False:
+            # co-run the depthwise wgrad with the dgrad (see _wgrad_stream)
+            main = torch.cuda.current_stream()
+            side.wait_stream(main)
+            with torch.cuda.stream(side):
+                gw = e.dwconv_wgrad(gy, x, w.shape[2], w.shape[3], stride[0],
+                                    stride[1], padding[0], padding[1])
         if ctx.needs_input_grad[0]:
             gx = e.dwconv_dgrad(gy, w, x.shape[2], x.shape[3], stride[0],
                                 stride[1], padding[0], padding[1])
-        if ctx.needs_input_grad[1]:
+        if side is not False:
+            main.wait_stream(side)
+            if not torch.cuda.is_current_stream_capturing():
+                gw.record_stream(main)
+                gy.record_stream(side)
+                x.record_stream(side)
+        elif ctx.needs_input_grad[1]:
             gw = e.dwconv_wgrad(gy, x, w.shape[2], w.shape[3], stride[0],
                                 stride[1], padding[0], padding[1])
         if has_bias and ctx.needs_input_grad[2]:
