@@ -550,13 +550,43 @@ __global__ __launch_bounds__(256) void mhsa_rel_grad_kernel(
   float acc[16] = {};
   const int64_t r0 = (int64_t)blockIdx.x * p.rows_per_block;
   const int64_t r1 = min(r0 + p.rows_per_block, p.rows);
-  for (int64_t r = r0; r < r1; ++r) {
-    const int bb = (int)(r / p.L), ll = (int)(r - (int64_t)bb * p.L);
+  // division-free row->(batch, pixel) advance + 4-row load batching: the
+  // one-row-at-a-time form paid an int64 div and a serial scalar-gather
+  // latency per row (302 us/call for a ~36 MB reduction)
+  int bb = (int)(r0 / p.L), ll = (int)(r0 - (int64_t)(r0 / p.L) * p.L);
+  int64_t r = r0;
+  for (; r + 3 < r1; r += 4) {
+    float qv[4];
+    const float* drp[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      qv[u] =
+          to_f32(p.q[strided_row(bb, ll, p.L, p.heads, p.qpix, p.D) + d]);
+      drp[u] = p.dr + (r + u) * p.M + m0;
+      if (++ll == p.L) {
+        ll = 0;
+        ++bb;
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      float v[16];
+#pragma unroll
+      for (int m = 0; m < 16; ++m) v[m] = (m < mn) ? drp[u][m] : 0.f;
+#pragma unroll
+      for (int m = 0; m < 16; ++m) acc[m] += v[m] * qv[u];
+    }
+  }
+  for (; r < r1; ++r) {
     const float qv =
         to_f32(p.q[strided_row(bb, ll, p.L, p.heads, p.qpix, p.D) + d]);
     const float* dr = p.dr + r * p.M + m0;
 #pragma unroll 7
     for (int m = 0; m < mn; ++m) acc[m] += dr[m] * qv;
+    if (++ll == p.L) {
+      ll = 0;
+      ++bb;
+    }
   }
   for (int m = 0; m < mn; ++m)
     atomicAdd(&p.out[(int64_t)(m0 + m) * p.D + d], acc[m] * p.scale);
