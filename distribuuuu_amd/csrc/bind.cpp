@@ -56,6 +56,10 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
 std::vector<at::Tensor> topk_acc(at::Tensor logits, at::Tensor target,
                                  int64_t topk);
 // conv.hip
+at::Tensor conv2d_fwd_v1(at::Tensor x, at::Tensor w, at::Tensor y,
+                         int64_t Ho, int64_t Wo, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                         int64_t groups);
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                       int64_t ph, int64_t pw, int64_t dh, int64_t dw,
                       int64_t groups);
@@ -167,6 +171,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_acc", &topk_acc);
   m.def("sgd_step", &sgd_step);
   m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd_v1", &conv2d_fwd_v1);
   m.def("conv2d_fwd_bn", &conv2d_fwd_bn);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_dgrad_bn", &conv2d_dgrad_bn);

@@ -1087,6 +1087,15 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
                          1, 0, 0, part_out, bemit);
 }
 
+// Probe/testing entry: force the v1 predicated kernel (bypasses the
+// small-family routing in conv2d_fwd_impl).
+at::Tensor conv2d_fwd_v1(at::Tensor x, at::Tensor w, at::Tensor y,
+                         int64_t Ho, int64_t Wo, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+                         int64_t groups) {
+  return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups);
+}
+
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, int64_t sh, int64_t sw,
                       int64_t ph, int64_t pw, int64_t dh, int64_t dw,
                       int64_t groups) {
