@@ -505,3 +505,32 @@ def test_fork_grads_match_unforked():
     for a, b in zip(gp_f, gp_p):
         s = b.abs().max().item() + 1e-6
         assert (a - b).abs().max().item() / s < 3e-2
+
+
+@pytest.mark.gpu
+def test_dgrad_pre_transform_matches():
+    """conv2d_dgrad_prep's route gate must stay in sync with
+    conv2d_dgrad_impl's dispatch: feeding the precomputed transform back in
+    (conv2d_dgrad_pre) must reproduce conv2d_dgrad exactly."""
+    e = _ext()
+    cl = torch.channels_last
+    # (C, H, K, R, stride, pad): v2 same-size, small same-size, strided
+    # proj, parity 3x3-s2, deep 3x3 same-size
+    cases = [
+        (512, 14, 512, 3, 1, 1),
+        (256, 56, 64, 1, 1, 0),
+        (512, 28, 1024, 1, 2, 0),
+        (64, 32, 64, 3, 2, 1),
+        (2048, 7, 512, 1, 1, 0),
+    ]
+    for c, h, k, r, s, pad in cases:
+        ho = (h + 2 * pad - r) // s + 1
+        gy = torch.randn(4, k, ho, ho, device="cuda",
+                         dtype=torch.bfloat16).contiguous(memory_format=cl)
+        w = (torch.randn(k, c, r, r, device="cuda", dtype=torch.bfloat16)
+             * 0.05).contiguous(memory_format=cl)
+        pre, kindt = e.conv2d_dgrad_prep(w, k, s, s, pad, pad, 1, 1, 1)
+        ref = e.conv2d_dgrad(gy, w, h, h, s, s, pad, pad, 1, 1, 1)
+        got = e.conv2d_dgrad_pre(gy, w, h, h, s, s, pad, pad, 1, 1, 1,
+                                 pre, int(kindt.item()))
+        assert torch.equal(ref, got), (c, h, k, r, s, pad)
