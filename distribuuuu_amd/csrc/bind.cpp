@@ -39,6 +39,13 @@ std::vector<at::Tensor> bn_bwd(at::Tensor gy, at::Tensor x,
 std::vector<at::Tensor> bn_apply_act_mask(at::Tensor x, at::Tensor scale,
                                           at::Tensor shift, int64_t act,
                                           at::Tensor res);
+at::Tensor bn_apply_act_pad(at::Tensor x, at::Tensor scale, at::Tensor shift,
+                            int64_t act, int64_t ph, int64_t pw);
+std::vector<at::Tensor> bn_bwd_pad(at::Tensor gy, at::Tensor x,
+                                   at::Tensor mean, at::Tensor rstd,
+                                   at::Tensor gamma, at::Tensor scale,
+                                   at::Tensor shift, int64_t act,
+                                   bool training, int64_t ph, int64_t pw);
 // pool.hip
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t K, int64_t S,
                                     int64_t P);
@@ -147,6 +154,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_apply_act", &bn_apply_act, py::arg("x"), py::arg("scale"),
         py::arg("shift"), py::arg("act"), py::arg("res") = py::none());
   m.def("bn_apply_act_mask", &bn_apply_act_mask);
+  m.def("bn_apply_act_pad", &bn_apply_act_pad);
+  m.def("bn_bwd_pad", &bn_bwd_pad);
   m.def("bn_bwd_stats", &bn_bwd_stats, py::arg("gy"), py::arg("x"),
         py::arg("res"), py::arg("scale"), py::arg("shift"),
         py::arg("act"));

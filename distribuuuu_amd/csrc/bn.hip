@@ -916,6 +916,257 @@ at::Tensor bn_apply_act(at::Tensor x, at::Tensor scale, at::Tensor shift,
   return y;
 }
 
+// ---- padded-output BN apply (bn1 -> 3x3 v2 conv fusion) --------------------
+// Writes y into a [N, C, H+2ph, W+2pw] canvas at (h+ph, w+pw) so the
+// consuming v2 conv reads it pad-free and the per-step pad_image pass
+// disappears. No residual (bn1/bn2 only). The interior walk advances
+// (n, h, w) incrementally (requires nrl <= W, host-gated).
+template <typename T>
+__global__ __launch_bounds__(256, 2) void bn_apply_pad_kernel(
+    const T* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, T* __restrict__ y, int N, int H, int W,
+    int C, int ph, int pw, int64_t rows_per_block, int act) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const int64_t rows = (int64_t)N * H * W;
+  const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, rows);
+  const int Hp = H + 2 * ph, Wp = W + 2 * pw;
+  const P* xp = reinterpret_cast<const P*>(x);
+  P* yp = reinterpret_cast<P*>(y);
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sc[V], sh[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      sc[j] = scale[cp * V + j];
+      sh[j] = shift[cp * V + j];
+    }
+    if (!active) continue;
+    int64_t row = row0 + rl;
+    if (row >= row1) continue;
+    const int64_t rstep = (int64_t)nrl * cpacks;
+    const P* xq = xp + row * cpacks + cp;
+    // interior (n, h, w) from the flat row, then incremental advance
+    int n = (int)(row / ((int64_t)H * W));
+    int rem = (int)(row - (int64_t)n * H * W);
+    int h = rem / W, w = rem - (rem / W) * W;
+    for (; row < row1; row += nrl) {
+      P px = xq[0];
+      xq += rstep;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float z = to_f32(px.v[j]) * sc[j] + sh[j];
+        px.v[j] = from_f32<T>(act_apply(z, act));
+      }
+      yp[((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp] = px;
+      w += nrl;
+      if (w >= W) {
+        w -= W;
+        if (++h == H) {
+          h = 0;
+          ++n;
+        }
+      }
+    }
+  }
+}
+
+// zero the pad ring of a [N, C, Hp, Wp] canvas (top/bottom rows full,
+// left/right strips on interior rows)
+template <typename T>
+__global__ void pad_ring_zero_kernel(T* __restrict__ y, int N, int H, int W,
+                                     int C, int ph, int pw) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  const int cpacks = C / V;
+  const int Hp = H + 2 * ph, Wp = W + 2 * pw;
+  const int row = blockIdx.y;  // n * Hp + hp
+  const int hp_ = row % Hp;
+  const bool full = hp_ < ph || hp_ >= H + ph;
+  const int strip = 2 * pw * cpacks;  // packs in the two side strips
+  const int npacks = full ? Wp * cpacks : strip;
+  P z = {};
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < npacks;
+       i += gridDim.x * blockDim.x) {
+    int col;
+    if (full) {
+      col = i;
+    } else {
+      const int half = i / (pw * cpacks);
+      const int off = i - half * (pw * cpacks);
+      col = half == 0 ? off : (W + pw) * cpacks + off;
+    }
+    reinterpret_cast<P*>(y)[(int64_t)row * Wp * cpacks + col] = z;
+  }
+}
+
+// pad-aware gy addressing for BN backward (bn1's gy arrives as the
+// consuming conv's padded-canvas dgrad): gy[(h+ph, w+pw)] of [Hp, Wp],
+// x/gx stay dense. No residual, no mask.
+template <typename T, int ACT>
+__global__ __launch_bounds__(256, 2) void bn_bwd_reduce_pad_kernel(
+    const T* __restrict__ gy, const T* __restrict__ x,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    float* __restrict__ part, int N, int H, int W, int C, int ph, int pw,
+    int64_t rows_per_block) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  __shared__ float red[256 * 2 * (16 / sizeof(T) > 8 ? 16 / sizeof(T) : 8)];
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const int64_t rows = (int64_t)N * H * W;
+  const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, rows);
+  const int Hp = H + 2 * ph, Wp = W + 2 * pw;
+  const P* gp = reinterpret_cast<const P*>(gy);
+  const P* xp = reinterpret_cast<const P*>(x);
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sc[V], sh[V];
+    if (ACT != 0) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        sc[j] = scale[cp * V + j];
+        sh[j] = shift[cp * V + j];
+      }
+    }
+    float accg[V] = {}, accgx[V] = {};
+    if (active && row0 + rl < row1) {
+      int64_t row = row0 + rl;
+      const int64_t rstep = (int64_t)nrl * cpacks;
+      const P* xq = xp + row * cpacks + cp;
+      int n = (int)(row / ((int64_t)H * W));
+      int rem = (int)(row - (int64_t)n * H * W);
+      int h = rem / W, w = rem - (rem / W) * W;
+      for (; row < row1; row += nrl) {
+        P px = xq[0];
+        xq += rstep;
+        P pg = gp[((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp];
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float xv = to_f32(px.v[j]);
+          float g = to_f32(pg.v[j]);
+          if (ACT != 0) g *= act_grad(xv * sc[j] + sh[j], ACT);
+          accg[j] += g;
+          accgx[j] += g * xv;
+        }
+        w += nrl;
+        if (w >= W) {
+          w -= W;
+          if (++h == H) {
+            h = 0;
+            ++n;
+          }
+        }
+      }
+    }
+    float* slot = &red[(rl * ncp + cp0) * 2 * V];
+    if (active) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        slot[j] = accg[j];
+        slot[V + j] = accgx[j];
+      }
+    }
+    __syncthreads();
+    for (int st = nrl >> 1; st > 0; st >>= 1) {
+      if (active && rl < st) {
+        const float* other = &red[((rl + st) * ncp + cp0) * 2 * V];
+#pragma unroll
+        for (int j = 0; j < 2 * V; ++j) slot[j] += other[j];
+      }
+      __syncthreads();
+    }
+    if (active && rl == 0) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        part[(int64_t)blockIdx.x * 2 * C + cp * V + j] = slot[j];
+        part[(int64_t)blockIdx.x * 2 * C + C + cp * V + j] = slot[V + j];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, int ACT>
+__global__ __launch_bounds__(256, 2) void bn_bwd_dx_pad_kernel(
+    const T* __restrict__ gy, const T* __restrict__ x,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    const float* __restrict__ P1c, const float* __restrict__ P2c,
+    const float* __restrict__ P3c, T* __restrict__ gx, int N, int H, int W,
+    int C, int ph, int pw, int64_t rows_per_block) {
+  constexpr int V = 16 / sizeof(T);
+  using P = Pack<T, V>;
+  const int cpacks = C / V;
+  const int ncp = min(cpacks, (int)blockDim.x);
+  const int nrl = p2_floor(blockDim.x / ncp);
+  const int cp0 = threadIdx.x % ncp;
+  const int rl = threadIdx.x / ncp;
+  const bool active = rl < nrl;
+  const int64_t rows = (int64_t)N * H * W;
+  const int64_t row0 = (int64_t)blockIdx.x * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, rows);
+  const int Hp = H + 2 * ph, Wp = W + 2 * pw;
+  const P* gp = reinterpret_cast<const P*>(gy);
+  const P* xp = reinterpret_cast<const P*>(x);
+  P* oxp = reinterpret_cast<P*>(gx);
+  for (int cp = cp0; cp < cpacks; cp += ncp) {
+    float sc[V], sh[V], a1[V], a2[V], a3[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const int cc = cp * V + j;
+      a1[j] = P1c[cc];
+      a2[j] = P2c[cc];
+      a3[j] = P3c[cc];
+      if (ACT != 0) {
+        sc[j] = scale[cc];
+        sh[j] = shift[cc];
+      }
+    }
+    if (!active) continue;
+    int64_t row = row0 + rl;
+    if (row >= row1) continue;
+    const int64_t rstep = (int64_t)nrl * cpacks;
+    const P* xq = xp + row * cpacks + cp;
+    P* oq = oxp + row * cpacks + cp;
+    int n = (int)(row / ((int64_t)H * W));
+    int rem = (int)(row - (int64_t)n * H * W);
+    int h = rem / W, w = rem - (rem / W) * W;
+    for (; row < row1; row += nrl) {
+      P px = xq[0];
+      xq += rstep;
+      P pg = gp[((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp];
+      P ox;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float xv = to_f32(px.v[j]);
+        float g = to_f32(pg.v[j]);
+        if (ACT != 0) g *= act_grad(xv * sc[j] + sh[j], ACT);
+        ox.v[j] = from_f32<T>(a1[j] * g + a3[j] * xv + a2[j]);
+      }
+      oq[0] = ox;
+      oq += rstep;
+      w += nrl;
+      if (w >= W) {
+        w -= W;
+        if (++h == H) {
+          h = 0;
+          ++n;
+        }
+      }
+    }
+  }
+}
+
 // Residual relu BN apply that ALSO emits the act' bitmask (one byte per
 // V-pack) so the backward kernels can drop the res stream and the z
 // recompute. Returns {y, mask}.
@@ -946,6 +1197,137 @@ std::vector<at::Tensor> bn_apply_act_mask(at::Tensor x, at::Tensor scale,
                        (unsigned char*)mask.data_ptr());
   });
   return {y, mask};
+}
+
+// bn1 apply writing into a padded canvas (consumer 3x3 v2 conv reads it
+// pad-free). Returns y [N, C, H+2ph, W+2pw] with a zeroed ring.
+at::Tensor bn_apply_act_pad(at::Tensor x, at::Tensor scale, at::Tensor shift,
+                            int64_t act, int64_t ph, int64_t pw) {
+  CHECK_GPU(x);
+  check_nhwc(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int Hp = H + 2 * (int)ph, Wp = W + 2 * (int)pw;
+  auto y = at::empty({N, C, Hp, Wp},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_apply_act_pad", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    const int cpacks = C / V;
+    int nrl = 1;
+    while (nrl * 2 <= std::max(256 / cpacks, 1)) nrl *= 2;
+    TORCH_CHECK(nrl <= W, "bn pad apply: nrl > W (host gate missed)");
+    const int64_t rows = (int64_t)N * H * W;
+    const int64_t rpb = pick_rows_per_block(rows, std::max(256 / cpacks, 1));
+    const int rgrid = (int)ceil_div(rows, rpb);
+    auto stream = cur_stream();
+    const int ring_max = std::max(Wp * cpacks, 2 * (int)pw * cpacks);
+    hipLaunchKernelGGL((pad_ring_zero_kernel<scalar_t>),
+                       dim3((ring_max + 255) / 256, N * Hp), dim3(256), 0,
+                       stream, (scalar_t*)y.data_ptr(), N, H, W, C, (int)ph,
+                       (int)pw);
+    hipLaunchKernelGGL((bn_apply_pad_kernel<scalar_t>), dim3(rgrid),
+                       dim3(256), 0, stream, (const scalar_t*)x.data_ptr(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       (scalar_t*)y.data_ptr(), N, H, W, C, (int)ph, (int)pw,
+                       rpb, (int)act);
+  });
+  return y;
+}
+
+template <typename scalar_t>
+void launch_bwd_reduce_pad(const at::Tensor& gy, const at::Tensor& x,
+                           const at::Tensor& scale, const at::Tensor& shift,
+                           at::Tensor& part, int N, int H, int W, int C,
+                           int ph, int pw, int64_t rpb, int act, int rgrid,
+                           hipStream_t stream) {
+#define BRP_CASE(A)                                                         \
+  hipLaunchKernelGGL((bn_bwd_reduce_pad_kernel<scalar_t, A>), dim3(rgrid),  \
+                     dim3(256), 0, stream,                                  \
+                     (const scalar_t*)gy.data_ptr(),                        \
+                     (const scalar_t*)x.data_ptr(),                         \
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),      \
+                     part.data_ptr<float>(), N, H, W, C, ph, pw, rpb)
+  switch (act) {
+    case 0: BRP_CASE(0); break;
+    case 1: BRP_CASE(1); break;
+    case 2: BRP_CASE(2); break;
+    default: BRP_CASE(3); break;
+  }
+#undef BRP_CASE
+}
+
+template <typename scalar_t>
+void launch_dx_pad(const at::Tensor& gy, const at::Tensor& x,
+                   const at::Tensor& scale, const at::Tensor& shift,
+                   const at::Tensor& P1, const at::Tensor& P2,
+                   const at::Tensor& P3, at::Tensor& gx, int N, int H, int W,
+                   int C, int ph, int pw, int64_t rpb, int act, int rgrid,
+                   hipStream_t stream) {
+#define BDP_CASE(A)                                                        \
+  hipLaunchKernelGGL((bn_bwd_dx_pad_kernel<scalar_t, A>), dim3(rgrid),     \
+                     dim3(256), 0, stream,                                 \
+                     (const scalar_t*)gy.data_ptr(),                       \
+                     (const scalar_t*)x.data_ptr(),                        \
+                     scale.data_ptr<float>(), shift.data_ptr<float>(),     \
+                     P1.data_ptr<float>(), P2.data_ptr<float>(),           \
+                     P3.data_ptr<float>(), (scalar_t*)gx.data_ptr(), N, H, \
+                     W, C, ph, pw, rpb)
+  switch (act) {
+    case 0: BDP_CASE(0); break;
+    case 1: BDP_CASE(1); break;
+    case 2: BDP_CASE(2); break;
+    default: BDP_CASE(3); break;
+  }
+#undef BDP_CASE
+}
+
+// BN backward when gy arrives in the padded canvas layout (x, gx dense).
+std::vector<at::Tensor> bn_bwd_pad(at::Tensor gy, at::Tensor x,
+                                   at::Tensor mean, at::Tensor rstd,
+                                   at::Tensor gamma, at::Tensor scale,
+                                   at::Tensor shift, int64_t act,
+                                   bool training, int64_t ph, int64_t pw) {
+  CHECK_GPU(gy);
+  check_nhwc(gy, "gy");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(gy.size(2) == H + 2 * ph && gy.size(3) == W + 2 * pw,
+              "bn_bwd_pad: gy canvas shape");
+  const int64_t rows = (int64_t)N * H * W;
+  auto fopts = x.options().dtype(at::kFloat);
+  auto gx = at::empty_like(x);
+  auto gw = at::empty({C}, fopts);
+  auto gb = at::empty({C}, fopts);
+  auto P1 = at::empty({C}, fopts);
+  auto P2 = at::empty({C}, fopts);
+  auto P3 = at::empty({C}, fopts);
+  DISPATCH_FLOAT_AND_BF16(x.scalar_type(), "bn_bwd_pad", [&] {
+    constexpr int V = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % V == 0, "C must be divisible by ", V);
+    auto stream = cur_stream();
+    const int cpacks = C / V;
+    const int nrl = std::max(256 / cpacks, 1);
+    int nrl_p2 = 1;
+    while (nrl_p2 * 2 <= nrl) nrl_p2 *= 2;
+    TORCH_CHECK(nrl_p2 <= W, "bn_bwd_pad: nrl > W");
+    const int64_t rpb = pick_rows_per_block(rows, nrl);
+    const int rgrid = (int)ceil_div(rows, rpb);
+    auto part = at::empty({rgrid, 2 * C}, fopts);
+    launch_bwd_reduce_pad<scalar_t>(gy, x, scale, shift, part, N, H, W, C,
+                                    (int)ph, (int)pw, rpb, (int)act, rgrid,
+                                    stream);
+    const float inv_cnt = 1.f / (float)rows;
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((int)ceil_div(C, 16)),
+                       dim3(256), 0, stream, part.data_ptr<float>(), rgrid,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), gw.data_ptr<float>(),
+                       gb.data_ptr<float>(), P1.data_ptr<float>(),
+                       P2.data_ptr<float>(), P3.data_ptr<float>(), C, inv_cnt,
+                       training ? 1 : 0);
+    launch_dx_pad<scalar_t>(gy, x, scale, shift, P1, P2, P3, gx, N, H, W,
+                            C, (int)ph, (int)pw, rpb, (int)act, rgrid,
+                            stream);
+  });
+  return {gx, gw, gb};
 }
 
 // Local raw grad-stat sums [2C] = [sum(g), sum(g*x)] (stage 1 + stage 2);

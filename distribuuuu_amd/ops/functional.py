@@ -324,6 +324,17 @@ def conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1),
             wp = _pack_narrow_groups(_cl(weight), groups, f)
             return conv2d(x, wp, bias, stride, padding, dilation, groups // f)
     if use_hip(x, "conv2d_fwd") and _hip_conv_ok(x, weight, groups):
+        if (padding != (0, 0) and x.dtype == torch.bfloat16
+                and groups == 1 and stride == (1, 1)):
+            # stride > 1 is excluded: its dgrad would lose the parity
+            # decomposition (gated on ph == 1)
+            if getattr(x, "_padded", None) == padding:
+                # producer BN already wrote the padded canvas
+                padding = (0, 0)
+            elif weight.shape[2] > 1:
+                srcbn = getattr(x, "_bn_weight", None)
+                if srcbn is not None:
+                    srcbn._bn_pad_out = padding
         emit = (bias is None and torch.is_grad_enabled()
                 and getattr(weight, "_emit_bn_partials", False))
         info = (getattr(x, "_bn_bwd_info", None)
@@ -365,8 +376,23 @@ class _HIPBatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, mean, rstd, scale, shift, training,
-                act_id, residual, slot, fork_slot):
+                act_id, residual, slot, fork_slot, pad):
         e = ext()
+        if pad is not None:
+            # padded-canvas output: the consuming 3x3 v2 conv reads it
+            # pad-free (no per-step pad_image pass); backward uses the
+            # pad-aware kernels on the gy canvas
+            y = e.bn_apply_act_pad(x, scale, shift, act_id, pad[0], pad[1])
+            ctx.save_for_backward(x, weight, scale, shift, mean, rstd,
+                                  x.new_empty(0), x.new_empty(0))
+            ctx.act_id = act_id
+            ctx.training = training
+            ctx.has_res = False
+            ctx.slot = None
+            ctx.fork_slot = None
+            ctx.pad = pad
+            return y
+        ctx.pad = None
         res = _cl(residual) if residual is not None else None
         mask = None
         if (res is not None and act_id == 1 and training
@@ -391,6 +417,14 @@ class _HIPBatchNormAct(torch.autograd.Function):
     def backward(ctx, gy):
         (x, weight, scale, shift, mean, rstd, res,
          mask) = ctx.saved_tensors
+        if ctx.pad is not None:
+            e = ext()
+            gamma = weight.float().contiguous()
+            gx, gw, gb = e.bn_bwd_pad(_cl(gy), x, mean, rstd, gamma, scale,
+                                      shift, ctx.act_id, ctx.training,
+                                      ctx.pad[0], ctx.pad[1])
+            return (gx, gw.to(weight.dtype), gb.to(weight.dtype), None,
+                    None, None, None, None, None, None, None, None, None)
         if mask.numel() == 0:
             mask = None
         e = ext()
@@ -421,7 +455,7 @@ class _HIPBatchNormAct(torch.autograd.Function):
             ctx.fork_slot["g"] = gres
         return (gx, gw.to(weight.dtype), gb.to(weight.dtype), None, None,
                 None, None, None, None, gres if ctx.has_res else None, None,
-                None)
+                None, None)
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
@@ -462,9 +496,27 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
         res_fslot = (getattr(residual, "_fork_slot", None)
                      if residual is not None and torch.is_grad_enabled()
                      else None)
+        pad = None
+        if (residual is None and slot is None
+                and x.dtype == torch.bfloat16):
+            pad = getattr(weight, "_bn_pad_out", None)
+            if pad is not None:
+                # host kernels need nrl <= W (row-incremental walk)
+                c = x.shape[1]
+                nrl, lim = 1, max(256 // max(c // 8, 1), 1)
+                while nrl * 2 <= lim:
+                    nrl *= 2
+                if c % 8 != 0 or nrl > x.shape[3]:
+                    pad = None
         y = _HIPBatchNormAct.apply(x, weight, bias, mean, rstd, scale, shift,
                                    training, act_id, residual, slot,
-                                   res_fslot)
+                                   res_fslot, pad)
+        if pad is not None:
+            y._padded = pad
+        elif residual is None:
+            # discovery: a consuming padded 3x3 conv flags this weight and
+            # from the next step on the apply writes the padded canvas
+            y._bn_weight = weight
         if slot is not None:
             # consuming convs pick this up and emit BN-backward stats from
             # their dgrad epilogue (see _HIPConv2d / conv2d_dgrad_bn)

@@ -627,3 +627,70 @@ def test_bn_residual_mask_matches_recompute():
                    True, mask=mask)
     for a, b in zip(ref, got):
         assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_bn_padded_apply_and_backward():
+    """bn_apply_act_pad == pad(bn_apply_act) and bn_bwd_pad == bn_bwd on the
+    interior (pad-aware gy addressing)."""
+    e = _ext()
+    cl = torch.channels_last
+    torch.manual_seed(5)
+    n, c, h, w, ph = 4, 128, 14, 14, 1
+    x = torch.randn(n, c, h, w, device="cuda",
+                    dtype=torch.bfloat16).contiguous(memory_format=cl)
+    gamma = torch.rand(c, device="cuda") + 0.5
+    beta = torch.randn(c, device="cuda")
+    mean, rstd, scale, shift = e.bn_stats(x, gamma, beta, None, None,
+                                          0.1, 1e-5, True, None)
+    y = e.bn_apply_act(x, scale, shift, 1, None)
+    yp = e.bn_apply_act_pad(x, scale, shift, 1, ph, ph)
+    assert yp.shape == (n, c, h + 2, w + 2)
+    ref = F.pad(y.float(), (ph, ph, ph, ph))
+    assert torch.equal(yp.float(), ref)
+    gy = torch.randn(n, c, h, w, device="cuda",
+                     dtype=torch.bfloat16).contiguous(memory_format=cl)
+    gyp = F.pad(gy, (ph, ph, ph, ph)).contiguous(memory_format=cl)
+    # garbage in the ring must be ignored
+    gyp[:, :, 0, :] = 7.0
+    gyp[:, :, -1, :] = -3.0
+    r0 = e.bn_bwd(gy, x, None, mean, rstd, gamma, scale, shift, 1, True,
+                  False, None)
+    r1 = e.bn_bwd_pad(gyp, x, mean, rstd, gamma, scale, shift, 1, True,
+                      ph, ph)
+    for a, b in zip(r0[:3], r1):
+        assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_bn_pad_fusion_end_to_end():
+    """Two-step discovery: step 2 runs bn1 -> padded canvas -> conv2 with
+    padding folded; grads must match an unfused run."""
+    from distribuuuu_amd.ops import BatchNorm2d, Conv2d
+
+    def run(flagged):
+        torch.manual_seed(11)
+        net = torch.nn.Sequential(
+            Conv2d(64, 128, 1, bias=False),
+            BatchNorm2d(128, act="relu"),
+            Conv2d(128, 128, 3, padding=1, bias=False),
+            BatchNorm2d(128, act="relu"),
+        ).cuda().bfloat16().to(memory_format=torch.channels_last)
+        x = torch.randn(4, 64, 28, 28, device="cuda",
+                        dtype=torch.bfloat16).contiguous(
+                            memory_format=torch.channels_last)
+        steps = 2 if flagged else 1
+        for _ in range(steps):
+            for p in net.parameters():
+                p.grad = None
+            y = net(x)
+            y.float().square().mean().backward()
+        if flagged:
+            assert getattr(net[1].weight, "_bn_pad_out", None) == (1, 1)
+        return [p.grad.float().cpu() for p in net.parameters()]
+
+    g_fused = run(True)
+    g_plain = run(False)
+    for a, b in zip(g_fused, g_plain):
+        sc = b.abs().max().item() + 1e-6
+        assert (a - b).abs().max().item() / sc < 3e-2
