@@ -1333,8 +1333,11 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
       const char* e = getenv("DISTRIBUUUU_CONV_V2");
       return e && e[0] == '0';
     }();
-    if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
-        dw * (S - 1) == 2 * pw && Kt0 % (8 * (int)groups) == 0 && !v2off2 &&
+    // any ph <= dh*(R-1) works at stride 1: the as-fwd conv with pad
+    // dh*(R-1)-ph lands on the conv-input size exactly (ph = 0 is the
+    // padded-canvas BN fusion's dgrad)
+    if (sh == 1 && sw == 1 && dh * (R - 1) >= ph &&
+        dw * (S - 1) >= pw && Kt0 % (8 * (int)groups) == 0 && !v2off2 &&
         dKg0 >= v2mink2 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
         gy.scalar_type() == at::kBFloat16) {
       const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
@@ -1389,8 +1392,9 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
     return gx;
   }
   const int Kt_ = gy.size(1);
-  if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
-      dw * (S - 1) == 2 * pw && (groups == 1 || Kt_ % (8 * groups) == 0)) {
+  if (sh == 1 && sw == 1 && dh * (R - 1) >= ph && dw * (S - 1) >= pw &&
+      (R > 1 || (ph == 0 && pw == 0)) &&
+      (groups == 1 || Kt_ % (8 * groups) == 0)) {
     // same-size conv: the plain fwd path applies (the v2-eligible shapes
     // were already taken above with the fused weight transform)
     return conv2d_fwd_impl(gy, wt, 1, 1, dh * (R - 1) - ph,
@@ -1444,8 +1448,8 @@ std::vector<at::Tensor> conv2d_dgrad_prep(at::Tensor w, int64_t Kt0,
   }();
   int64_t kind = 0;
   at::Tensor out;
-  if (sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
-      dw * (S - 1) == 2 * pw && Kt0 % (8 * groups) == 0 && !v2off4 &&
+  if (sh == 1 && sw == 1 && dh * (R - 1) >= ph &&
+      dw * (S - 1) >= pw && Kt0 % (8 * groups) == 0 && !v2off4 &&
       dKg0 >= v2mink4 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
       w.scalar_type() == at::kBFloat16) {
     const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
@@ -1511,8 +1515,9 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
   const int Kt0 = gy.size(1);
   const int dKg0 = Ct / (int)groups;
   const int dCg0 = Kt0 / (int)groups;
-  const bool same_size = sh == 1 && sw == 1 && dh * (R - 1) == 2 * ph &&
-                         dw * (S - 1) == 2 * pw;
+  const bool same_size = sh == 1 && sw == 1 && dh * (R - 1) >= ph &&
+                         dw * (S - 1) >= pw &&
+                         (R > 1 || (ph == 0 && pw == 0));
   static const int v2mink3 = []() {
     const char* e = getenv("DISTRIBUUUU_V2_MINK");
     return e ? atoi(e) : 192;
