@@ -498,7 +498,11 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, training=False,
                      else None)
         pad = None
         if (residual is None and slot is None
-                and x.dtype == torch.bfloat16):
+                and x.dtype == torch.bfloat16
+                and os.environ.get("DISTRIBUUUU_BN_PAD", "0") == "1"):
+            # measured net-negative on ResNet-50 (the pad-aware backward
+            # kernels lack the 4-row unroll and give back more than the
+            # removed pad_image pass saves) — opt-in until they catch up
             pad = getattr(weight, "_bn_pad_out", None)
             if pad is not None:
                 # host kernels need nrl <= W (row-incremental walk)

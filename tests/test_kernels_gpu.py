@@ -636,7 +636,7 @@ def test_bn_padded_apply_and_backward():
     e = _ext()
     cl = torch.channels_last
     torch.manual_seed(5)
-    n, c, h, w, ph = 4, 128, 14, 14, 1
+    n, c, h, w, ph = 4, 128, 28, 28, 1
     x = torch.randn(n, c, h, w, device="cuda",
                     dtype=torch.bfloat16).contiguous(memory_format=cl)
     gamma = torch.rand(c, device="cuda") + 0.5
@@ -669,6 +669,8 @@ def test_bn_pad_fusion_end_to_end():
     from distribuuuu_amd.ops import BatchNorm2d, Conv2d
 
     def run(flagged):
+        import os
+        os.environ["DISTRIBUUUU_BN_PAD"] = "1" if flagged else "0"
         torch.manual_seed(11)
         net = torch.nn.Sequential(
             Conv2d(64, 128, 1, bias=False),
@@ -691,6 +693,8 @@ def test_bn_pad_fusion_end_to_end():
 
     g_fused = run(True)
     g_plain = run(False)
+    import os
+    os.environ.pop("DISTRIBUUUU_BN_PAD", None)
     for a, b in zip(g_fused, g_plain):
         sc = b.abs().max().item() + 1e-6
         assert (a - b).abs().max().item() / sc < 3e-2
