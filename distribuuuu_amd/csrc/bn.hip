@@ -956,6 +956,33 @@ __global__ __launch_bounds__(256, 2) void bn_apply_pad_kernel(
     int n = (int)(row / ((int64_t)H * W));
     int rem = (int)(row - (int64_t)n * H * W);
     int h = rem / W, w = rem - (rem / W) * W;
+    for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+      P px4[4];
+      int64_t yo[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        px4[u] = xq[u * rstep];
+        yo[u] = ((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp;
+        w += nrl;
+        if (w >= W) {
+          w -= W;
+          if (++h == H) {
+            h = 0;
+            ++n;
+          }
+        }
+      }
+      xq += 4 * rstep;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float z = to_f32(px4[u].v[j]) * sc[j] + sh[j];
+          px4[u].v[j] = from_f32<T>(act_apply(z, act));
+        }
+        yp[yo[u]] = px4[u];
+      }
+    }
     for (; row < row1; row += nrl) {
       P px = xq[0];
       xq += rstep;
@@ -1047,6 +1074,34 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_reduce_pad_kernel(
       int n = (int)(row / ((int64_t)H * W));
       int rem = (int)(row - (int64_t)n * H * W);
       int h = rem / W, w = rem - (rem / W) * W;
+      for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+        P px4[4], pg4[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          px4[u] = xq[u * rstep];
+          pg4[u] =
+              gp[((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp];
+          w += nrl;
+          if (w >= W) {
+            w -= W;
+            if (++h == H) {
+              h = 0;
+              ++n;
+            }
+          }
+        }
+        xq += 4 * rstep;
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            float xv = to_f32(px4[u].v[j]);
+            float g = to_f32(pg4[u].v[j]);
+            if (ACT != 0) g *= act_grad(xv * sc[j] + sh[j], ACT);
+            accg[j] += g;
+            accgx[j] += g * xv;
+          }
+      }
       for (; row < row1; row += nrl) {
         P px = xq[0];
         xq += rstep;
@@ -1141,6 +1196,37 @@ __global__ __launch_bounds__(256, 2) void bn_bwd_dx_pad_kernel(
     int n = (int)(row / ((int64_t)H * W));
     int rem = (int)(row - (int64_t)n * H * W);
     int h = rem / W, w = rem - (rem / W) * W;
+    for (; row + 3 * (int64_t)nrl < row1; row += 4 * (int64_t)nrl) {
+      P px4[4], pg4[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        px4[u] = xq[u * rstep];
+        pg4[u] =
+            gp[((((int64_t)n * Hp + h + ph) * Wp) + w + pw) * cpacks + cp];
+        w += nrl;
+        if (w >= W) {
+          w -= W;
+          if (++h == H) {
+            h = 0;
+            ++n;
+          }
+        }
+      }
+      xq += 4 * rstep;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        P ox;
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float xv = to_f32(px4[u].v[j]);
+          float g = to_f32(pg4[u].v[j]);
+          if (ACT != 0) g *= act_grad(xv * sc[j] + sh[j], ACT);
+          ox.v[j] = from_f32<T>(a1[j] * g + a3[j] * xv + a2[j]);
+        }
+        oq[u * rstep] = ox;
+      }
+      oq += 4 * rstep;
+    }
     for (; row < row1; row += nrl) {
       P px = xq[0];
       xq += rstep;
