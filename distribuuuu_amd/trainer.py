@@ -61,8 +61,10 @@ def build_network(device):
 def train_epoch(train_loader, net, criterion, optimizer, epoch, device, dtype):
     """One epoch of the hot loop (reference trainer.py:14-64)."""
     topk = cfg.TRAIN.TOPK
-    batch_time, data_time, losses, top1, topk_m, progress = utils.construct_meters(
-        len(train_loader), f"Epoch[{epoch + 1}/{cfg.OPTIM.MAX_EPOCH}]", topk)
+    (batch_time, data_time, losses, top1, topk_m, ips,
+     progress) = utils.construct_meters(
+        len(train_loader), f"Epoch[{epoch + 1}/{cfg.OPTIM.MAX_EPOCH}]", topk,
+        batch_size=cfg.TRAIN.BATCH_SIZE * utils.get_world_size())
     lr = utils.get_epoch_lr(epoch)
     utils.set_lr(optimizer, lr)
     if utils.get_rank() == 0:
@@ -90,7 +92,10 @@ def train_epoch(train_loader, net, criterion, optimizer, epoch, device, dtype):
             losses.update(metrics[0].item(), inputs.size(0))
             top1.update(metrics[1].item(), inputs.size(0))
             topk_m.update(metrics[2].item(), inputs.size(0))
-        batch_time.update(time.time() - end)
+        step_t = time.time() - end
+        batch_time.update(step_t)
+        ips.update(cfg.TRAIN.BATCH_SIZE * utils.get_world_size()
+                   / max(step_t, 1e-9))
         end = time.time()
         if idx % cfg.TRAIN.PRINT_FREQ == 0 and utils.get_rank() == 0:
             progress.display(idx)

@@ -180,16 +180,23 @@ class ProgressMeter:
         return "[" + f + "/" + f.format(num_batches) + "]"
 
 
-def construct_meters(num_batches, prefix, topk=5):
-    """Time/Data/Loss/Acc@1/Acc@k meter set (reference utils.py:255-262)."""
+def construct_meters(num_batches, prefix, topk=5, batch_size=None):
+    """Time/Data/Loss/Acc@1/Acc@k meter set (reference utils.py:255-262),
+    plus a derived whole-job images/sec meter (the reference
+    never reports throughput directly; img/s = world * batch / batch_time)."""
     batch_time = AverageMeter("Time", ":.3f")
     data_time = AverageMeter("Data", ":.3f")
     losses = AverageMeter("Loss", ":.4e")
     top1 = AverageMeter("Acc@1", ":6.2f")
     topk_m = AverageMeter(f"Acc@{topk}", ":6.2f")
-    progress = ProgressMeter(
-        num_batches, [batch_time, data_time, losses, top1, topk_m], prefix=prefix
-    )
+    meters = [batch_time, data_time, losses, top1, topk_m]
+    ips = None
+    if batch_size:
+        ips = AverageMeter("img/s", ":8.1f")
+        meters.append(ips)
+    progress = ProgressMeter(num_batches, meters, prefix=prefix)
+    if batch_size:
+        return batch_time, data_time, losses, top1, topk_m, ips, progress
     return batch_time, data_time, losses, top1, topk_m, progress
 
 
