@@ -441,9 +441,10 @@ struct MhsaBwdKVParams {
 __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
   const int b = blockIdx.x;
   const int k0 = blockIdx.y * 64;
-  // LDS: q16/do16 [16][D], ds16/p16 [16][64]
-  __shared__ __hip_bfloat16 q16[16][128];
-  __shared__ __hip_bfloat16 do16[16][128];
+  // LDS: q16/do16 staged as FP32 [16][D] (the bf16 form cost 64
+  // v_cvt per j-iteration in the FMA loop); ds16/p16 stay bf16
+  __shared__ float q16[16][128];
+  __shared__ float do16[16][128];
   __shared__ __hip_bfloat16 ds16[16][64];
   __shared__ __hip_bfloat16 p16[16][64];
   const int tid = threadIdx.x;
@@ -495,8 +496,11 @@ __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
   auto stage_write = [&]() {
     {
       const int r = tid >> 4, c8 = (tid & 15) << 3;
-      *reinterpret_cast<bf16x8a*>(&q16[r][c8]) = rq;
-      *reinterpret_cast<bf16x8a*>(&do16[r][c8]) = rdo;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        q16[r][c8 + j] = to_f32((__hip_bfloat16)rq[j]);
+        do16[r][c8 + j] = to_f32((__hip_bfloat16)rdo[j]);
+      }
     }
     if (tid < 128) {
       const int r = tid >> 3, c8 = (tid & 7) << 3;
@@ -513,16 +517,20 @@ __global__ __launch_bounds__(256) void mhsa_bwd_kv_kernel(MhsaBwdKVParams p) {
     for (int j = 0; j < jmax; ++j) {
       const float sv = to_f32(ds16[j][kj - k0]);
       const float pvv = to_f32(p16[j][kj - k0]);
-      // vectorized LDS reads of the 32-wide d slice
+      // vectorized fp32 LDS reads of the 32-wide d slice
 #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        bf16x8a qv = *reinterpret_cast<const bf16x8a*>(&q16[j][d0 + c * 8]);
-        bf16x8a dov = *reinterpret_cast<const bf16x8a*>(&do16[j][d0 + c * 8]);
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          acck[c * 8 + u] += sv * to_f32((__hip_bfloat16)qv[u]);
-          accv[c * 8 + u] += pvv * to_f32((__hip_bfloat16)dov[u]);
-        }
+      for (int c = 0; c < 8; ++c) {
+        const float4 qv = *reinterpret_cast<const float4*>(&q16[j][d0 + c * 4]);
+        const float4 dov =
+            *reinterpret_cast<const float4*>(&do16[j][d0 + c * 4]);
+        acck[c * 4 + 0] += sv * qv.x;
+        acck[c * 4 + 1] += sv * qv.y;
+        acck[c * 4 + 2] += sv * qv.z;
+        acck[c * 4 + 3] += sv * qv.w;
+        accv[c * 4 + 0] += pvv * dov.x;
+        accv[c * 4 + 1] += pvv * dov.y;
+        accv[c * 4 + 2] += pvv * dov.z;
+        accv[c * 4 + 3] += pvv * dov.w;
       }
     }
     __syncthreads();
