@@ -18,6 +18,17 @@
 // barrier).
 #include "common.h"
 
+// v2 minimum reduction depth R*S*Cg (DISTRIBUUUU_V2_MINRSC, default 512):
+// below this the 3-slot glds ring has too few k-steps to spin up.
+static int v2_minrsc() {
+  static const int v = []() {
+    const char* e = getenv("DISTRIBUUUU_V2_MINRSC");
+    return e ? atoi(e) : 512;
+  }();
+  return v;
+}
+
+
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
@@ -882,7 +893,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
     if (!v2i_off && part_out == nullptr && bemit == nullptr && ph == 0 &&
         pw == 0 && sh == 1 && sw == 1 && dh == 1 && dw == 1 &&
         (Si * Cgi) % 64 == 0 && Kgi >= 96 &&
-        (int64_t)Ri * Si * Cgi >= 512 && Cgi % 8 == 0 &&
+        (int64_t)Ri * Si * Cgi >= v2_minrsc() && Cgi % 8 == 0 &&
         Ho + Ri - 1 <= x.size(2) && Wo + Si - 1 <= x.size(3))
       return conv2d_fwd_v2_into(x, w, y, Ho, Wo, groups, osh, osw, oh0, ow0,
                                 acc);
@@ -975,7 +986,8 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
     const char* e = getenv("DISTRIBUUUU_V2_MINK");
     return e ? atoi(e) : 192;
   }();
-  if (!v2_off && v2Kg >= v2_mink && (int64_t)R * S * v2Cg >= 512 &&
+  if (!v2_off && v2Kg >= v2_mink &&
+      (int64_t)R * S * v2Cg >= v2_minrsc() &&
       v2Cg % 8 == 0 && x.scalar_type() == at::kBFloat16)
     return conv2d_fwd_v2p(x, w, sh, sw, ph, pw, dh, dw, groups, part_out,
                           bemit);
@@ -1086,6 +1098,7 @@ static at::Tensor conv2d_fwd_impl(at::Tensor x, at::Tensor w, int64_t sh,
   return conv2d_fwd_into(x, w, y, Ho, Wo, sh, sw, ph, pw, dh, dw, groups, 1,
                          1, 0, 0, part_out, bemit);
 }
+
 
 // Probe/testing entry: force the v1 predicated kernel (bypasses the
 // small-family routing in conv2d_fwd_impl).
@@ -1338,7 +1351,8 @@ static at::Tensor conv2d_dgrad_impl(at::Tensor gy, at::Tensor w, int64_t H,
     // padded-canvas BN fusion's dgrad)
     if (sh == 1 && sw == 1 && dh * (R - 1) >= ph &&
         dw * (S - 1) >= pw && Kt0 % (8 * (int)groups) == 0 && !v2off2 &&
-        dKg0 >= v2mink2 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
+        dKg0 >= v2mink2 && (int64_t)R * S * dCg0 >= v2_minrsc() &&
+        dCg0 % 8 == 0 &&
         gy.scalar_type() == at::kBFloat16) {
       const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
       auto wsp = (pre != nullptr && pre_kind == 1)
@@ -1450,7 +1464,8 @@ std::vector<at::Tensor> conv2d_dgrad_prep(at::Tensor w, int64_t Kt0,
   at::Tensor out;
   if (sh == 1 && sw == 1 && dh * (R - 1) >= ph &&
       dw * (S - 1) >= pw && Kt0 % (8 * groups) == 0 && !v2off4 &&
-      dKg0 >= v2mink4 && (int64_t)R * S * dCg0 >= 512 && dCg0 % 8 == 0 &&
+      dKg0 >= v2mink4 && (int64_t)R * S * dCg0 >= v2_minrsc() &&
+      dCg0 % 8 == 0 &&
       w.scalar_type() == at::kBFloat16) {
     const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
     out = weight_flip_t_span(w, groups, SPAN64);
@@ -1531,7 +1546,7 @@ std::tuple<at::Tensor, int64_t> conv2d_dgrad_acc(
       into.is_contiguous(at::MemoryFormat::ChannelsLast) &&
       into.sizes() == at::IntArrayRef({(int64_t)N, (int64_t)Ct, H, W})) {
     if (same_size && !v2off3 && Kt0 % 8 == 0 && dKg0 >= v2mink3 &&
-        (int64_t)R * S * dCg0 >= 512) {
+        (int64_t)R * S * dCg0 >= v2_minrsc()) {
       const int SPAN64 = ((int)(S * dCg0) + 63) / 64 * 64;
       auto wsp = (pre != nullptr && pre_kind == 1)
                      ? *pre
