@@ -2,7 +2,8 @@
 
 The reference falls back to timm for archs missing from its registry
 (trainer.py:123-128); this zoo implements every shipped baseline arch natively
-(regnet/efficientnet included), so no external model dependency exists.
+(regnet/efficientnet included) and keeps the timm fallback for names outside
+the zoo when timm happens to be installed.
 """
 
 from .botnet import botnet50  # noqa: F401
@@ -23,12 +24,21 @@ from .resnet import (  # noqa: F401
 
 
 def build_model(arch, **kwargs):
-    """Look the arch name up in this module's globals (KeyError when unknown)."""
+    """Look the arch name up in this module's globals; unknown names fall
+    back to ``timm.create_model`` when timm is installed (reference
+    trainer.py:123-128 semantics), else raise KeyError."""
     try:
         factory = globals()[arch]
     except KeyError:
-        raise KeyError(
-            f"Unknown arch '{arch}'. Available: "
-            + ", ".join(sorted(k for k, v in globals().items() if callable(v)))
-        )
+        try:
+            import timm
+        except ImportError:
+            raise KeyError(
+                f"Unknown arch '{arch}' (and timm is not installed). "
+                "Available: "
+                + ", ".join(sorted(k for k, v in globals().items()
+                                   if callable(v)))
+            ) from None
+        nc = kwargs.pop("num_classes", 1000)
+        return timm.create_model(arch, num_classes=nc, **kwargs)
     return factory(**kwargs)
