@@ -83,3 +83,24 @@ def test_densenet_memory_efficient_matches():
     a.eval(), b.eval()
     x = torch.randn(2, 3, 64, 64)
     assert torch.allclose(a(x), b(x), atol=1e-6)
+
+
+def test_build_model_unknown_arch_raises():
+    import pytest as _pytest
+
+    from distribuuuu_amd.models import build_model
+
+    with _pytest.raises(KeyError):
+        build_model("no_such_arch_anywhere")
+
+
+def test_fork_cpu_passthrough():
+    """fork() must be a no-op off-GPU (the accumulate fusion is a GPU
+    epilogue feature); the two handles alias the input."""
+    import torch
+
+    from distribuuuu_amd.ops import functional as DF
+
+    x = torch.randn(2, 4, requires_grad=True)
+    a, b = DF.fork(x)
+    assert a is x and b is x
