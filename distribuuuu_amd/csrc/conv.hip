@@ -28,6 +28,17 @@ static int v2_minrsc() {
   return v;
 }
 
+// separate depth floor for the scatter/accumulate fwd_into route (proj and
+// residual-fork dgrads) so it can be swept independently of the main
+// forward dispatch
+static int v2_into_minrsc() {
+  static const int v = []() {
+    const char* e = getenv("DISTRIBUUUU_V2INTO_MINRSC");
+    return e ? atoi(e) : v2_minrsc();
+  }();
+  return v;
+}
+
 
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -893,7 +904,7 @@ at::Tensor conv2d_fwd_into(at::Tensor x, at::Tensor w, at::Tensor y, int64_t Ho,
     if (!v2i_off && part_out == nullptr && bemit == nullptr && ph == 0 &&
         pw == 0 && sh == 1 && sw == 1 && dh == 1 && dw == 1 &&
         (Si * Cgi) % 64 == 0 && Kgi >= 96 &&
-        (int64_t)Ri * Si * Cgi >= v2_minrsc() && Cgi % 8 == 0 &&
+        (int64_t)Ri * Si * Cgi >= v2_into_minrsc() && Cgi % 8 == 0 &&
         Ho + Ri - 1 <= x.size(2) && Wo + Si - 1 <= x.size(3))
       return conv2d_fwd_v2_into(x, w, y, Ho, Wo, groups, osh, osw, oh0, ow0,
                                 acc);
