@@ -156,8 +156,12 @@ def train_model():
         if rank == 0:
             logger.info(f"Auto-resumed from {ckpt} at epoch {start_epoch}")
     elif cfg.MODEL.WEIGHTS:
-        utils.load_checkpoint(cfg.MODEL.WEIGHTS, net,
-                              optimizer if cfg.TRAIN.LOAD_OPT else None)
+        # reference trainer.py:147-149 also resumes epoch/best from the
+        # weights file; load_checkpoint only advances the epoch when the
+        # optimizer state was actually restored
+        start_epoch, best_acc1 = utils.load_checkpoint(
+            cfg.MODEL.WEIGHTS, net,
+            optimizer if cfg.TRAIN.LOAD_OPT else None)
         if rank == 0:
             logger.info(f"Loaded initial weights from {cfg.MODEL.WEIGHTS}")
 
