@@ -79,3 +79,46 @@ def test_validate_runs(tmp_path):
         assert 0.0 <= top1 <= 100.0
     finally:
         data_mod.DummyDataset = orig
+
+
+def test_resume_equivalence(tmp_path):
+    """train-save-reload-continue must land on the same
+    weights as an uninterrupted run (checkpoint carries model + optimizer
+    momentum; LR is a pure function of the epoch)."""
+    cfg.OUT_DIR = str(tmp_path)
+    torch.manual_seed(3)
+    x = torch.randn(8, 3, 32, 32)
+    y = torch.randint(0, 10, (8,))
+
+    def make():
+        torch.manual_seed(11)
+        net = models.build_model("resnet18", num_classes=10)
+        opt = utils.construct_optimizer(net)
+        utils.set_lr(opt, 0.05)
+        return net, opt
+
+    def step(net, opt):
+        net.train()
+        loss = DF.cross_entropy(net(x), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+
+    # uninterrupted: 4 steps
+    net_a, opt_a = make()
+    for _ in range(4):
+        step(net_a, opt_a)
+
+    # interrupted: 2 steps, checkpoint, fresh objects, resume, 2 steps
+    net_b, opt_b = make()
+    for _ in range(2):
+        step(net_b, opt_b)
+    path = utils.save_checkpoint(net_b, opt_b, epoch=1, best_acc1=0.0)
+    net_c, opt_c = make()
+    start_epoch, _ = utils.load_checkpoint(path, net_c, opt_c)
+    assert start_epoch == 2
+    for _ in range(2):
+        step(net_c, opt_c)
+
+    for pa, pc in zip(net_a.parameters(), net_c.parameters()):
+        assert torch.allclose(pa, pc, atol=1e-6), "resume diverged"
